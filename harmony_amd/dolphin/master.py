@@ -1,0 +1,44 @@
+"""Per-job runner: builds tables/trainer/provider and runs the worker tasklet.
+
+Reference: dolphin/core/master/DolphinMaster.java:177-193 + ETTaskRunner —
+the master submits a ServerTasklet and WorkerTasklet to every executor. In
+the SPMD rebuild the server role has no dedicated thread: all "server work"
+is the owner-side update epilogue of push collectives (exactly as the
+reference's ServerTasklet is a placeholder and real server work happens in
+ET's UPDATE handling, core/server/ServerTasklet.java:28).
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+
+from harmony_amd.config import JobConfig
+from harmony_amd import mlapps
+from harmony_amd.dolphin.metrics import MetricCollector
+from harmony_amd.dolphin.worker import WorkerTasklet
+from harmony_amd.runtime.bootstrap import ExecutorContext
+from harmony_amd.runtime.control import ControlPlane, TaskUnitScheduler
+
+
+def run_job(job: JobConfig, ctx: ExecutorContext,
+            cp: Optional[ControlPlane] = None,
+            tus: Optional[TaskUnitScheduler] = None,
+            stream: Optional[torch.cuda.Stream] = None) -> MetricCollector:
+    """Run one PS job to completion on this rank; returns this rank's metrics.
+
+    Collective: every rank of the job's executor set must call this with the
+    same JobConfig.
+    """
+    cp = cp or ControlPlane(ctx.store, ctx.rank, ctx.world_size)
+    tus = tus or TaskUnitScheduler(cp, {job.job_id}, multi_job=False)
+    app = mlapps.get_app(job.app)
+    tables, trainer, provider = app.build(job, ctx, cp)
+    tasklet = WorkerTasklet(job, trainer, provider, cp, tus,
+                            ctx.rank, ctx.world_size, stream=stream)
+    metrics = tasklet.run()
+    ev = trainer.evaluate_model()
+    for k, v in (ev or {}).items():
+        metrics.add_custom(k, float(v))
+    return metrics

@@ -1,0 +1,88 @@
+"""Model accessor: the worker's view of the PS model table.
+
+Reference: dolphin/core/worker/ModelAccessor.java (pull/push SPI) with
+ETModelAccessor (direct table access, ETModelAccessor.java:43) and
+CachedModelAccessor (worker-side cache with background refresh,
+CachedModelAccessor.java:40-130).
+
+MI355X shape: pull = collective gather (keys or all), push = collective
+scatter of deltas; the cached accessor keeps the last pulled rows resident in
+HBM and refreshes them every `refresh_batches` batches instead of per batch —
+the device-memory analogue of the reference's Guava cache + refresh thread.
+"""
+
+from __future__ import annotations
+
+import time
+from typing import Dict, Optional
+
+import torch
+
+METRIC_PULL_TIME = "total_pull_time_sec"
+METRIC_PUSH_TIME = "total_push_time_sec"
+
+
+class ETModelAccessor:
+    def __init__(self, table):
+        self.table = table
+        self.metrics: Dict[str, float] = {METRIC_PULL_TIME: 0.0, METRIC_PUSH_TIME: 0.0}
+
+    def pull(self, keys: torch.Tensor) -> torch.Tensor:
+        t0 = time.perf_counter()
+        out = self.table.get(keys)
+        self.metrics[METRIC_PULL_TIME] += time.perf_counter() - t0
+        return out
+
+    def pull_all(self) -> torch.Tensor:
+        t0 = time.perf_counter()
+        out = self.table.pull_all()
+        self.metrics[METRIC_PULL_TIME] += time.perf_counter() - t0
+        return out
+
+    def push(self, keys: torch.Tensor, deltas: torch.Tensor) -> None:
+        t0 = time.perf_counter()
+        self.table.update(keys, deltas)
+        self.metrics[METRIC_PUSH_TIME] += time.perf_counter() - t0
+
+    def push_dense(self, grad_full: torch.Tensor) -> None:
+        t0 = time.perf_counter()
+        self.table.push_dense(grad_full)
+        self.metrics[METRIC_PUSH_TIME] += time.perf_counter() - t0
+
+    def get_and_reset_metrics(self) -> Dict[str, float]:
+        m = dict(self.metrics)
+        for k in self.metrics:
+            self.metrics[k] = 0.0
+        return m
+
+
+class CachedModelAccessor(ETModelAccessor):
+    """Keeps the full model resident; re-pulls every `refresh_batches` pulls.
+    Pushes are applied write-through to the cache with the table's update
+    function (reference CachedModelAccessor pushes via UpdateFunction)."""
+
+    def __init__(self, table, refresh_batches: int = 1):
+        super().__init__(table)
+        self.refresh_batches = max(1, refresh_batches)
+        self._cache: Optional[torch.Tensor] = None
+        self._age = 0
+
+    def pull_all(self) -> torch.Tensor:
+        if self._cache is None or self._age >= self.refresh_batches:
+            self._cache = super().pull_all()
+            self._age = 0
+        self._age += 1
+        return self._cache
+
+    def pull(self, keys: torch.Tensor) -> torch.Tensor:
+        return self.pull_all()[keys]
+
+    def push(self, keys: torch.Tensor, deltas: torch.Tensor) -> None:
+        super().push(keys, deltas)
+        if self._cache is not None:
+            from harmony_amd.et import update_functions as uf
+
+            fn = uf.update_fn(self.table.cfg.update_fn)
+            rows = self._cache[keys]
+            self._cache[keys] = fn(rows, deltas.to(rows.dtype),
+                                   **self.table.cfg.update_args)

@@ -1,0 +1,102 @@
+"""The worker tasklet: the per-rank training loop of one PS job.
+
+Reference: dolphin/core/worker/WorkerTasklet.java:96-168 — global barrier,
+per-epoch data preparation, per-batch [SYNC -> PULL -> COMP -> PUSH] with
+task-unit scheduling around each phase, per-batch/epoch metric emission.
+
+MI355X mapping:
+  SYNC  -> SSP clock tick (control store), no collective
+  PULL  -> RCCL collective under a global NET ticket
+  COMP  -> HIP kernels on this job's stream (concurrent across jobs)
+  PUSH  -> RCCL collective under a global NET ticket
+"""
+
+from __future__ import annotations
+
+import contextlib
+import time
+from typing import Optional
+
+import torch
+
+from harmony_amd.config import JobConfig
+from harmony_amd.dolphin.data_provider import TrainingDataProvider
+from harmony_amd.dolphin.metrics import BatchMetrics, EpochMetrics, MetricCollector
+from harmony_amd.dolphin.trainer import Trainer
+from harmony_amd.runtime.control import ControlPlane, SSPClock, TaskUnitScheduler
+
+
+class WorkerTasklet:
+    def __init__(self, job: JobConfig, trainer: Trainer,
+                 provider: TrainingDataProvider, cp: ControlPlane,
+                 tus: TaskUnitScheduler, rank: int, world_size: int,
+                 stream: Optional[torch.cuda.Stream] = None):
+        self.job = job
+        self.trainer = trainer
+        self.provider = provider
+        self.cp = cp
+        self.tus = tus
+        self.rank = rank
+        self.world_size = world_size
+        self.stream = stream
+        self.metrics = MetricCollector(job.job_id, rank)
+        self._phase = 0
+        self.ssp = SSPClock(cp, job.job_id, world_size, job.clock_slack)
+
+    def _next_phase(self) -> int:
+        self._phase += 1
+        return self._phase
+
+    def run(self) -> MetricCollector:
+        jid = self.job.job_id
+        stream_ctx = (torch.cuda.stream(self.stream) if self.stream is not None
+                      else contextlib.nullcontext())
+        with stream_ctx:
+            # initialize() may issue collectives (e.g. LDA's initial count
+            # push) — serialize it like any NET phase.
+            with self.tus.net(jid, self._next_phase()):
+                self.trainer.initialize()
+            # INIT -> RUN global barrier (reference WorkerGlobalBarrier)
+            self.cp.barrier(f"{jid}/run", self.world_size)
+            stopped = False
+            for epoch in range(self.job.max_num_epochs):
+                if stopped:
+                    break
+                ep_t0 = time.perf_counter()
+                ep_examples = 0
+                for bidx, batch in enumerate(self.provider.epoch_iter(epoch)):
+                    # SYNC: SSP clock (reference MiniBatchBarrier -> master)
+                    if not self.ssp.tick_and_wait(self.rank):
+                        stopped = True
+                        break
+                    b_t0 = time.perf_counter()
+                    self.trainer.set_batch_data(batch)
+                    # PULL
+                    t0 = time.perf_counter()
+                    with self.tus.net(jid, self._next_phase()):
+                        self.trainer.pull_model()
+                    pull_t = time.perf_counter() - t0
+                    # COMP
+                    t0 = time.perf_counter()
+                    self.trainer.local_compute()
+                    comp_t = time.perf_counter() - t0
+                    # PUSH
+                    t0 = time.perf_counter()
+                    with self.tus.net(jid, self._next_phase()):
+                        self.trainer.push_update()
+                    push_t = time.perf_counter() - t0
+                    n = self.trainer.num_batch_examples()
+                    ep_examples += n
+                    self.metrics.add_batch(BatchMetrics(
+                        epoch_idx=epoch, batch_idx=bidx, num_examples=n,
+                        batch_time_sec=time.perf_counter() - b_t0,
+                        pull_time_sec=pull_t, comp_time_sec=comp_t,
+                        push_time_sec=push_t))
+                self.trainer.on_epoch_finished(epoch)
+                self.metrics.add_epoch(EpochMetrics(
+                    epoch_idx=epoch, num_examples=ep_examples,
+                    epoch_time_sec=time.perf_counter() - ep_t0))
+            # RUN -> CLEANUP barrier
+            self.cp.barrier(f"{jid}/cleanup", self.world_size)
+            self.trainer.cleanup()
+        return self.metrics

@@ -1,0 +1,138 @@
+"""MLR — multinomial (softmax) logistic regression on the PS.
+
+Reference: dolphin/mlapps/mlr/MLRTrainer.java — the model is partitioned as
+key = classIdx*numPartitionsPerClass + partIdx -> Vector(featuresPerPartition)
+and workers pull ALL partitions each mini-batch (MLRTrainer.java:185-187);
+per sample p = softmax(Wx) with a log-sum-exp guard (:475-489), gradient
+grad_j -= stepSize * p_j * x (+L2) accumulated densely (:374-398); the server
+applies old.addi(delta) (MLRETModelUpdateFunction.java:60-62).
+
+MI355X shape: the whole per-batch math is (batch x F) @ (F x C) GEMM on the
+matrix cores + a fused row-softmax/label-subtract kernel + the outer-product
+gradient GEMM; pull-all = all-gather, dense push = reduce-scatter with the
+add-update epilogue. GEMMs go to rocBLAS (torch.matmul); the softmax+grad
+scaling is the fused HIP kernel `mlr_softmax_grad` (ops/csrc/mlr.hip).
+
+App args: num_classes, num_features, num_parts_per_class, batch_size,
+step_size, lambda (L2), dtype.
+"""
+
+from __future__ import annotations
+
+import torch
+
+from harmony_amd.config import JobConfig, TableConfig
+from harmony_amd.dolphin.data_provider import TrainingDataProvider
+from harmony_amd.dolphin.model_accessor import ETModelAccessor
+from harmony_amd.dolphin.trainer import Trainer, TrainerContext
+from harmony_amd.et.table import Table
+from harmony_amd import ops
+
+MODEL_TABLE = "mlr_model"
+
+
+def defaults(job: JobConfig) -> dict:
+    a = dict(num_classes=10, num_features=1024, num_parts_per_class=8,
+             batch_size=4096, step_size=0.01, lam=1e-4, dtype="float32",
+             decay_rate=0.9, decay_period=5)
+    a.update(job.app_args)
+    return a
+
+
+def model_table_cfg(job: JobConfig, world_size: int) -> TableConfig:
+    a = defaults(job)
+    P = a["num_parts_per_class"]
+    assert a["num_features"] % P == 0
+    return TableConfig(
+        table_id=f"{job.job_id}/{MODEL_TABLE}",
+        num_keys=a["num_classes"] * P,
+        value_dim=a["num_features"] // P,
+        dtype=a["dtype"],
+        num_blocks=max(world_size, min(64, a["num_classes"] * P)),
+        update_fn="add",
+        init_fn="gaussian",
+        init_args={"std": 0.01},
+    )
+
+
+def make_batches(job: JobConfig, rank: int, device: torch.device):
+    """Synthetic classification data: per-class gaussian blobs (deterministic
+    per (job, rank)); one block = one mini-batch (reference
+    ETTrainingDataProvider one-block-one-batch)."""
+    a = defaults(job)
+    C, F, B = a["num_classes"], a["num_features"], a["batch_size"]
+    from harmony_amd.utils import stable_seed
+
+    g = torch.Generator().manual_seed(stable_seed(job.job_id, rank))
+    centers = torch.randn(C, F, generator=g) * 0.5
+    blocks = []
+    n_blocks = job.num_worker_blocks or job.num_mini_batches
+    for _ in range(n_blocks):
+        y = torch.randint(0, C, (B,), generator=g)
+        x = centers[y] + torch.randn(B, F, generator=g)
+        blocks.append((x.to(device), y.to(device)))
+    return blocks
+
+
+class MLRTrainer(Trainer):
+    def __init__(self, ctx: TrainerContext):
+        super().__init__(ctx)
+        self.a = defaults(JobConfig(job_id=ctx.job_id, app="mlr",
+                                    app_args=ctx.app_args))
+        self.accessor = ETModelAccessor(ctx.table(MODEL_TABLE))
+        self.step_size = self.a["step_size"]
+        self.W = None          # [C*P, F/P] pulled model
+        self._loss_sum = 0.0
+        self._loss_n = 0
+        self._correct = 0
+
+    def _w_matrix(self) -> torch.Tensor:
+        C, F = self.a["num_classes"], self.a["num_features"]
+        return self.W.view(C, F)
+
+    def pull_model(self) -> None:
+        self.W = self.accessor.pull_all()
+
+    def local_compute(self) -> None:
+        x, y = self.batch
+        W = self._w_matrix()                       # [C, F]
+        logits = x @ W.t()                         # rocBLAS GEMM [B, C]
+        # fused softmax + label-subtract + CE loss (HIP kernel on GPU)
+        p, loss, correct = ops.softmax_grad_ce(logits, y)
+        grad = p.t() @ x                           # [C, F] GEMM
+        grad = grad / x.shape[0] + self.a["lam"] * W
+        self.grad_delta = (-self.step_size) * grad
+        self._loss_sum += float(loss)
+        self._loss_n += x.shape[0]
+        self._correct += int(correct)
+
+    def push_update(self) -> None:
+        C, P = self.a["num_classes"], self.a["num_parts_per_class"]
+        self.accessor.push_dense(self.grad_delta.view(C * P, -1))
+
+    def on_epoch_finished(self, epoch: int) -> None:
+        if (epoch + 1) % self.a["decay_period"] == 0:
+            self.step_size *= self.a["decay_rate"]
+
+    def evaluate_model(self):
+        if not self._loss_n:
+            return {}
+        out = {"cross_entropy": self._loss_sum / self._loss_n,
+               "accuracy": self._correct / self._loss_n}
+        self._loss_sum, self._loss_n, self._correct = 0.0, 0, 0
+        return out
+
+    def num_batch_examples(self) -> int:
+        return self.batch[0].shape[0]
+
+
+def build(job: JobConfig, ctx, cp):
+    cfg = model_table_cfg(job, ctx.world_size)
+    comm = ctx.new_data_plane()
+    table = Table(cfg, ctx.rank, ctx.world_size, ctx.device, comm=comm)
+    tctx = TrainerContext(job_id=job.job_id, rank=ctx.rank,
+                          world_size=ctx.world_size, device=ctx.device,
+                          tables={MODEL_TABLE: table}, app_args=job.app_args)
+    trainer = MLRTrainer(tctx)
+    provider = TrainingDataProvider(make_batches(job, ctx.rank, ctx.device))
+    return {MODEL_TABLE: table}, trainer, provider

@@ -1,0 +1,145 @@
+"""NMF — async-SGD non-negative matrix factorization on the PS.
+
+Reference: dolphin/mlapps/nmf/NMFTrainer.java — model table holds R columns
+(colIdx -> Vector(rank)), a worker-local table holds L rows; per nonzero
+(i,j,v): e = L_i.R_j - v, lGrad += 2e R_j (+L2), rGrad_j += 2e L_i (+L2)
+(updateGradient:328-367); per-thread gradient maps merged then pushed
+(aggregateGradient:375-406); server applies new = clamp(old - step*delta)
+(NMFETModelUpdateFunction.java:48-52); loss = squared error (:414-456).
+
+MI355X shape: the batch's nonzeros are device-resident CSR; K1 (nmf_grad,
+ops/csrc/nmf.hip) computes e/lgrad/rgrad in one kernel with rank-wide lanes;
+the per-thread hashmap merge of the reference (K2) becomes an on-device
+segment-sum over sorted column keys; pull = all-to-all-v key gather; push =
+all-to-all-v of column deltas with the clamp update fused on the owner.
+
+App args: num_rows (global), num_cols, rank, nnz_per_row, batch_size (rows),
+step_size, lam, max_val, decay_rate, decay_period.
+"""
+
+from __future__ import annotations
+
+import torch
+
+from harmony_amd.config import JobConfig, TableConfig
+from harmony_amd.dolphin.data_provider import TrainingDataProvider
+from harmony_amd.dolphin.model_accessor import ETModelAccessor
+from harmony_amd.dolphin.trainer import Trainer, TrainerContext
+from harmony_amd.et.table import Table
+from harmony_amd.utils import stable_seed
+from harmony_amd import ops
+
+MODEL_TABLE = "nmf_model"
+
+
+def defaults(job: JobConfig) -> dict:
+    a = dict(num_rows=16384, num_cols=16384, rank=100, nnz_per_row=64,
+             rows_per_batch=2048, step_size=0.01, lam=0.0, max_val=1e6,
+             decay_rate=0.9, decay_period=5, dtype="float32")
+    a.update(job.app_args)
+    return a
+
+
+def model_table_cfg(job: JobConfig, world_size: int) -> TableConfig:
+    a = defaults(job)
+    return TableConfig(
+        table_id=f"{job.job_id}/{MODEL_TABLE}",
+        num_keys=a["num_cols"],
+        value_dim=a["rank"],
+        dtype=a["dtype"],
+        num_blocks=max(world_size, min(256, a["num_cols"])),
+        update_fn="nmf_sgd",
+        update_args={"step_size": a["step_size"], "max_val": a["max_val"]},
+        init_fn="uniform_clamped",
+        init_args={"max_val": 1.0},
+    )
+
+
+class NMFBatch:
+    """One mini-batch: CSR nonzeros of a contiguous range of local L rows,
+    with the batch's unique columns precomputed (static per block)."""
+
+    def __init__(self, l_rows: torch.Tensor, row_idx: torch.Tensor,
+                 col_idx: torch.Tensor, vals: torch.Tensor):
+        self.l_rows = l_rows            # [n_rows] indices into local L
+        self.row_idx = row_idx          # [nnz] index into l_rows (0..n_rows)
+        self.col_idx = col_idx          # [nnz] global column key
+        self.vals = vals                # [nnz]
+        self.uniq_cols, self.col_local = torch.unique(col_idx, return_inverse=True)
+        self.num_examples = l_rows.shape[0]
+
+
+def make_batches(job: JobConfig, rank: int, device: torch.device):
+    a = defaults(job)
+    n_blocks = job.num_worker_blocks or job.num_mini_batches
+    rows_local = a["rows_per_batch"] * n_blocks
+    g = torch.Generator().manual_seed(stable_seed(job.job_id, "data", rank))
+    blocks = []
+    for b in range(n_blocks):
+        lo = b * a["rows_per_batch"]
+        l_rows = torch.arange(lo, lo + a["rows_per_batch"])
+        nnz = a["rows_per_batch"] * a["nnz_per_row"]
+        row_idx = torch.arange(a["rows_per_batch"]).repeat_interleave(a["nnz_per_row"])
+        col_idx = torch.randint(0, a["num_cols"], (nnz,), generator=g)
+        vals = torch.rand(nnz, generator=g)
+        blocks.append(NMFBatch(l_rows.to(device), row_idx.to(device),
+                               col_idx.to(device), vals.to(device)))
+    return blocks, rows_local
+
+
+class NMFTrainer(Trainer):
+    def __init__(self, ctx: TrainerContext, num_local_rows: int):
+        super().__init__(ctx)
+        self.a = defaults(JobConfig(job_id=ctx.job_id, app="nmf",
+                                    app_args=ctx.app_args))
+        self.accessor = ETModelAccessor(ctx.table(MODEL_TABLE))
+        # Worker-local model table (reference: local table rowKey -> L row,
+        # DolphinJobEntity.java:100-110): device-resident, never leaves HBM.
+        g = torch.Generator().manual_seed(stable_seed(ctx.job_id, "L", ctx.rank))
+        self.L = (torch.rand(num_local_rows, self.a["rank"], generator=g)
+                  .to(ctx.device))
+        self.step_size = self.a["step_size"]
+        self._sq_err = 0.0
+        self.R_batch = None
+
+    def pull_model(self) -> None:
+        self.R_batch = self.accessor.pull(self.batch.uniq_cols)
+
+    def local_compute(self) -> None:
+        b = self.batch
+        L_batch = self.L[b.l_rows]
+        lgrad, rgrad, sq = ops.nmf_grad(L_batch, self.R_batch, b.row_idx,
+                                        b.col_local, b.vals, self.a["lam"])
+        # local L update (worker-side SGD apply, same rule as the server's)
+        self.L[b.l_rows] = (L_batch - self.step_size * lgrad).clamp_(
+            0.0, self.a["max_val"])
+        self.rgrad = rgrad
+        self._sq_err += float(sq)
+
+    def push_update(self) -> None:
+        self.accessor.push(self.batch.uniq_cols, self.rgrad)
+
+    def on_epoch_finished(self, epoch: int) -> None:
+        if (epoch + 1) % self.a["decay_period"] == 0:
+            self.step_size *= self.a["decay_rate"]
+
+    def evaluate_model(self):
+        out = {"sq_err": self._sq_err}
+        self._sq_err = 0.0
+        return out
+
+    def num_batch_examples(self) -> int:
+        return self.batch.num_examples
+
+
+def build(job: JobConfig, ctx, cp):
+    cfg = model_table_cfg(job, ctx.world_size)
+    comm = ctx.new_data_plane()
+    table = Table(cfg, ctx.rank, ctx.world_size, ctx.device, comm=comm)
+    blocks, rows_local = make_batches(job, ctx.rank, ctx.device)
+    tctx = TrainerContext(job_id=job.job_id, rank=ctx.rank,
+                          world_size=ctx.world_size, device=ctx.device,
+                          tables={MODEL_TABLE: table}, app_args=job.app_args)
+    trainer = NMFTrainer(tctx, rows_local)
+    provider = TrainingDataProvider(blocks)
+    return {MODEL_TABLE: table}, trainer, provider

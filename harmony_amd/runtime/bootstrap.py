@@ -1,0 +1,151 @@
+"""Executor bootstrap: one process per GPU, RCCL data plane + control store.
+
+Replaces the reference's REEF driver/evaluator lifecycle (ExecutorManager,
+evalmanager — reference et/driver/impl/ExecutorManager.java:62,
+evalmanager/api/EvaluatorManager.java:39). Launch model:
+
+  torchrun --nnodes=1 --nproc-per-node N --master-addr 127.0.0.1 <prog>
+
+Each rank pins one MI355X, joins the default process group (backend "nccl" ==
+RCCL on ROCm; "gloo" for CPU tests), and connects to the control TCPStore
+hosted by rank 0.
+"""
+
+from __future__ import annotations
+
+import datetime
+import os
+from dataclasses import dataclass
+from typing import Optional
+
+import torch
+import torch.distributed as dist
+
+from harmony_amd.config import RuntimeConfig
+
+
+@dataclass
+class ExecutorContext:
+    rank: int
+    world_size: int
+    device: torch.device
+    backend: str               # "nccl" | "gloo" | "none"
+    store: object              # control-plane store (TCPStore or LocalStore)
+
+    @property
+    def is_master(self) -> bool:
+        return self.rank == 0
+
+    def new_data_plane(self, group=None):
+        from harmony_amd.et.comm import DataPlane
+
+        if self.backend == "none":
+            return None
+        return DataPlane(group, self.rank, self.world_size, self.device)
+
+    def barrier(self):
+        if dist.is_initialized():
+            dist.barrier()
+
+
+def _pick_device(cfg_device: str, local_rank: int) -> torch.device:
+    if cfg_device == "cpu":
+        return torch.device("cpu")
+    if cfg_device == "cuda" or (cfg_device == "auto" and torch.cuda.is_available()):
+        if not torch.cuda.is_available():
+            raise RuntimeError("device=cuda requested but no GPU is visible")
+        torch.cuda.set_device(local_rank % torch.cuda.device_count())
+        return torch.device("cuda", local_rank % torch.cuda.device_count())
+    return torch.device("cpu")
+
+
+def init_executor(cfg: Optional[RuntimeConfig] = None) -> ExecutorContext:
+    cfg = cfg or RuntimeConfig()
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", str(cfg.world_size)))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    master_addr = os.environ.get("MASTER_ADDR", cfg.master_addr)
+    master_port = int(os.environ.get("MASTER_PORT", str(cfg.master_port)))
+
+    device = _pick_device(cfg.device, local_rank)
+
+    if world > 1 or "RANK" in os.environ:
+        backend = cfg.backend
+        if backend == "auto":
+            backend = "nccl" if device.type == "cuda" else "gloo"
+        if not dist.is_initialized():
+            dist.init_process_group(
+                backend=backend, rank=rank, world_size=world,
+                timeout=datetime.timedelta(seconds=300))
+        from torch.distributed import TCPStore
+
+        store = TCPStore(master_addr, master_port + 7, world,
+                         is_master=(rank == 0),
+                         timeout=datetime.timedelta(seconds=300))
+    else:
+        backend = "none"
+        store = LocalStore()
+
+    return ExecutorContext(rank=rank, world_size=world, device=device,
+                           backend=backend, store=store)
+
+
+class LocalStore:
+    """In-process stand-in for TCPStore in single-process mode (tests, local
+    runtime). Implements the subset of the Store API the control plane uses."""
+
+    def __init__(self):
+        import threading
+
+        self._d = {}
+        self._lock = threading.Lock()
+        self._cv = threading.Condition(self._lock)
+
+    def set(self, key: str, value) -> None:
+        with self._cv:
+            self._d[key] = _to_bytes(value)
+            self._cv.notify_all()
+
+    def get(self, key: str) -> bytes:
+        import time
+
+        deadline = time.monotonic() + 300
+        with self._cv:
+            while key not in self._d:
+                if not self._cv.wait(timeout=1.0) and time.monotonic() > deadline:
+                    raise KeyError(key)
+            return self._d[key]
+
+    def add(self, key: str, amount: int) -> int:
+        with self._cv:
+            cur = int(self._d.get(key, b"0")) + amount
+            self._d[key] = str(cur).encode()
+            self._cv.notify_all()
+            return cur
+
+    def compare_set(self, key: str, expected: str, desired: str) -> bytes:
+        with self._cv:
+            cur = self._d.get(key)
+            if (cur is None and expected == "") or cur == _to_bytes(expected):
+                self._d[key] = _to_bytes(desired)
+                self._cv.notify_all()
+                return _to_bytes(desired)
+            return cur if cur is not None else _to_bytes(expected)
+
+    def wait(self, keys) -> None:
+        for k in keys:
+            self.get(k)
+
+    def check(self, keys) -> bool:
+        with self._lock:
+            return all(k in self._d for k in keys)
+
+    def delete_key(self, key: str) -> bool:
+        with self._cv:
+            return self._d.pop(key, None) is not None
+
+
+def _to_bytes(v) -> bytes:
+    if isinstance(v, bytes):
+        return v
+    return str(v).encode()

@@ -1,0 +1,219 @@
+"""Control plane: barriers, SSP clock, and global task-unit ordering.
+
+Covers three reference subsystems with one TCPStore-backed service:
+
+* WorkerGlobalBarrier / MiniBatchBarrier (reference
+  dolphin/core/worker/WorkerGlobalBarrier.java:32, MiniBatchBarrier.java:30)
+  -> named epoch-counted barriers.
+* MiniBatchController's SSP bounded-async clock (reference
+  dolphin/core/master/MiniBatchController.java:34-117) -> per-worker clock
+  counters with a decentralized slack check and a total-batch stop decision.
+* LocalTaskUnitScheduler + GlobalTaskUnitScheduler (reference
+  et/evaluator/impl/LocalTaskUnitScheduler.java:33,
+  et/driver/impl/GlobalTaskUnitScheduler.java:36) -> a global NET-phase
+  sequencer. Concurrent jobs on the same GPUs issue RCCL collectives; RCCL
+  deadlocks if two ranks enqueue different jobs' collectives in different
+  orders, so every NET phase draws a globally-unique ticket at first request
+  and each rank issues NET phases in ticket order (filtered to the jobs it
+  participates in). COMP phases run concurrently on per-job HIP streams —
+  GPU streams replace the reference's CPU semaphore.
+"""
+
+from __future__ import annotations
+
+import threading
+import time
+from typing import Dict, Optional, Set
+
+
+class ControlPlane:
+    def __init__(self, store, rank: int, world_size: int):
+        self.store = store
+        self.rank = rank
+        self.world_size = world_size
+
+    # ------------------------------------------------------------- barriers
+
+    def barrier(self, name: str, n: Optional[int] = None) -> None:
+        """Named reusable barrier over n participants (default: world)."""
+        n = n or self.world_size
+        if n <= 1:
+            return
+        arrived = self.store.add(f"bar/{name}/a", 1)
+        epoch = (arrived - 1) // n + 1
+        target = epoch * n
+        while int(self.store.add(f"bar/{name}/a", 0)) < target:
+            time.sleep(0.0005)
+
+    # -------------------------------------------------------------- counters
+
+    def incr(self, key: str, amount: int = 1) -> int:
+        return int(self.store.add(key, amount))
+
+    def read(self, key: str) -> int:
+        return int(self.store.add(key, 0))
+
+    def set_flag(self, key: str) -> None:
+        self.store.set(key, "1")
+
+    def flag_set(self, key: str) -> bool:
+        try:
+            return self.store.check([key])
+        except Exception:
+            return False
+
+
+class SSPClock:
+    """Bounded-asynchrony clock for one job (reference MiniBatchController).
+
+    Each worker ticks once per mini-batch. A worker blocks while it is more
+    than `slack` batches ahead of the slowest worker. Early stop (used by the
+    elasticity orchestrator's optimization window) is expressed as a
+    *per-worker batch index* (`stop_at`), never a racy flag: every rank stops
+    after exactly the same number of batches, so all ranks issue the same
+    number of collectives — a flag-based stop could split the job's ranks
+    across a collective and deadlock RCCL.
+    """
+
+    def __init__(self, cp: ControlPlane, job_id: str, num_workers: int,
+                 slack: int):
+        self.cp = cp
+        self.job_id = job_id
+        self.num_workers = num_workers
+        self.slack = slack
+
+    def _ckey(self, r: int) -> str:
+        return f"ssp/{self.job_id}/clock/{r}"
+
+    def request_stop_at(self, batch_idx: int) -> None:
+        """Ask all workers to stop after `batch_idx` batches (master/orchestrator)."""
+        self.cp.store.set(f"ssp/{self.job_id}/stop_at", str(batch_idx))
+
+    def _stop_at(self) -> int:
+        if self.cp.flag_set(f"ssp/{self.job_id}/stop_at"):
+            return int(self.cp.store.get(f"ssp/{self.job_id}/stop_at"))
+        return 1 << 60
+
+    def tick_and_wait(self, rank: int) -> bool:
+        """Advance my clock; block per SSP slack. Returns False if this worker
+        passed the stop point."""
+        mine = self.cp.incr(self._ckey(rank), 1)
+        if mine > self._stop_at():
+            return False
+        if self.slack >= 0 and self.num_workers > 1:
+            while True:
+                slowest = min(self.cp.read(self._ckey(r))
+                              for r in range(self.num_workers))
+                if mine - slowest <= self.slack:
+                    break
+                time.sleep(0.0005)
+        return True
+
+
+class TaskUnitScheduler:
+    """Global NET-phase sequencer (see module docstring).
+
+    Usage per rank:
+        tus = TaskUnitScheduler(cp, my_jobs={"jobA", "jobB"})
+        with tus.net(job_id, phase_idx):
+            ... issue collectives ...
+
+    Ticket allocation: the first rank to reach phase (job, idx) wins a
+    compare_set race, draws seq = incr("tu/seq"), and publishes
+    tu/seq_of/<job>/<idx> = seq and tu/job_of/<seq> = job. Tickets are
+    therefore gap-free and per-job monotone (a job's phase N+1 is only
+    requested after phase N's collective completed somewhere, which is after
+    phase N drew its ticket).
+
+    Each rank runs phases in ticket order filtered to `my_jobs`: before
+    entering seq S it waits until every seq < S that belongs to one of its
+    jobs has locally completed. Single-job mode short-circuits (no store
+    traffic) — ordering is only needed when jobs co-locate.
+    """
+
+    def __init__(self, cp: ControlPlane, my_jobs: Optional[Set[str]] = None,
+                 multi_job: bool = False):
+        self.cp = cp
+        self.my_jobs = my_jobs or set()
+        self.multi_job = multi_job
+        self._done: Set[int] = set()
+        self._watermark = 1            # smallest seq not yet locally complete
+        self._job_cache: Dict[int, str] = {}
+        self._lock = threading.Lock()
+        self._cv = threading.Condition(self._lock)
+
+    def set_jobs(self, jobs: Set[str], multi_job: Optional[bool] = None) -> None:
+        with self._lock:
+            self.my_jobs = set(jobs)
+            if multi_job is not None:
+                self.multi_job = multi_job
+
+    def _ticket(self, job_id: str, phase_idx: int) -> int:
+        key = f"tu/seq_of/{job_id}/{phase_idx}"
+        won = self.cp.store.compare_set(key, "", "PENDING")
+        if won == b"PENDING":
+            seq = self.cp.incr("tu/seq", 1)
+            self.cp.store.set(f"tu/job_of/{seq}", job_id)
+            self.cp.store.set(key + "/v", str(seq))
+            return seq
+        self.cp.store.wait([key + "/v"])
+        return int(self.cp.store.get(key + "/v"))
+
+    def _job_of(self, seq: int) -> str:
+        job = self._job_cache.get(seq)
+        if job is None:
+            self.cp.store.wait([f"tu/job_of/{seq}"])
+            job = self.cp.store.get(f"tu/job_of/{seq}").decode()
+            self._job_cache[seq] = job
+        return job
+
+    def net(self, job_id: str, phase_idx: int):
+        return _NetPhase(self, job_id, phase_idx)
+
+    def _enter(self, job_id: str, phase_idx: int) -> int:
+        if not self.multi_job:
+            return -1
+        seq = self._ticket(job_id, phase_idx)
+        # Wait until all earlier tickets of my jobs completed locally.
+        # Seqs of jobs this rank does not run are marked done immediately.
+        while True:
+            with self._cv:
+                while self._watermark in self._done:
+                    self._done.discard(self._watermark)
+                    self._watermark += 1
+                pending = [s for s in range(self._watermark, seq)
+                           if s not in self._done]
+            blockers = []
+            for s in pending:
+                if self._job_of(s) in self.my_jobs:
+                    blockers.append(s)
+                else:
+                    with self._cv:
+                        self._done.add(s)
+            if not blockers:
+                return seq
+            with self._cv:
+                self._cv.wait(timeout=0.001)
+
+    def _exit(self, seq: int) -> None:
+        if seq < 0:
+            return
+        with self._cv:
+            self._done.add(seq)
+            self._cv.notify_all()
+
+
+class _NetPhase:
+    def __init__(self, tus: TaskUnitScheduler, job_id: str, phase_idx: int):
+        self.tus = tus
+        self.job_id = job_id
+        self.phase_idx = phase_idx
+        self.seq = -1
+
+    def __enter__(self):
+        self.seq = self.tus._enter(self.job_id, self.phase_idx)
+        return self
+
+    def __exit__(self, *exc):
+        self.tus._exit(self.seq)
+        return False
