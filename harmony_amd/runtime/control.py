@@ -150,8 +150,12 @@ class TaskUnitScheduler:
 
     def _ticket(self, job_id: str, phase_idx: int) -> int:
         key = f"tu/seq_of/{job_id}/{phase_idx}"
-        won = self.cp.store.compare_set(key, "", "PENDING")
-        if won == b"PENDING":
+        # compare_set returns the value CURRENTLY stored — a loser of the race
+        # sees the winner's token, so the token must be contender-unique
+        # (one contender per rank: a phase belongs to exactly one tasklet).
+        token = f"P{self.cp.rank}"
+        cur = self.cp.store.compare_set(key, "", token)
+        if cur == token.encode():
             seq = self.cp.incr("tu/seq", 1)
             self.cp.store.set(f"tu/job_of/{seq}", job_id)
             self.cp.store.set(key + "/v", str(seq))

@@ -1,0 +1,57 @@
+"""Spawn N-rank test workers (gloo on CPU here; same code path as RCCL on GPU)."""
+
+from __future__ import annotations
+
+import multiprocessing as mp
+import os
+import pickle
+import socket
+import traceback
+
+
+def free_port() -> int:
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
+
+
+def _entry(fn, rank, world, port, q, args):
+    try:
+        os.environ["RANK"] = str(rank)
+        os.environ["LOCAL_RANK"] = str(rank)
+        os.environ["WORLD_SIZE"] = str(world)
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        out = fn(rank, world, *args)
+        q.put((rank, "ok", pickle.dumps(out)))
+    except Exception:  # noqa: BLE001
+        q.put((rank, "err", traceback.format_exc()))
+    finally:
+        import torch.distributed as dist
+
+        if dist.is_initialized():
+            dist.destroy_process_group()
+
+
+def run_dist(fn, world: int = 2, args=(), timeout: int = 120):
+    """Run fn(rank, world, *args) in `world` processes; returns [out_rank0..]."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = free_port()
+    procs = [ctx.Process(target=_entry, args=(fn, r, world, port, q, args))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    results = {}
+    try:
+        for _ in range(world):
+            rank, status, payload = q.get(timeout=timeout)
+            if status == "err":
+                raise RuntimeError(f"rank {rank} failed:\n{payload}")
+            results[rank] = pickle.loads(payload)
+    finally:
+        for p in procs:
+            p.join(timeout=10)
+            if p.is_alive():
+                p.terminate()
+    return [results[r] for r in range(world)]

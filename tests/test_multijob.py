@@ -1,0 +1,57 @@
+"""Concurrent PS jobs sharing executors — the multi-tenancy core.
+
+Two jobs (MLR + NMF) run in two tasklet threads on each of 2 ranks; the
+global task-unit scheduler must order their NET phases identically on every
+rank (else collectives interleave and deadlock). Mirrors the reference's
+LocalTaskUnitScheduler/GlobalTaskUnitScheduler behavior
+(SURVEY.md §2.1/§2.2)."""
+
+import threading
+
+from tests.dist_helper import run_dist
+
+
+def _two_jobs_worker(rank, world):
+    from harmony_amd.config import JobConfig, RuntimeConfig
+    from harmony_amd.dolphin.master import run_job
+    from harmony_amd.runtime.bootstrap import init_executor
+    from harmony_amd.runtime.control import ControlPlane, TaskUnitScheduler
+
+    ctx = init_executor(RuntimeConfig(device="cpu"))
+    cp = ControlPlane(ctx.store, ctx.rank, ctx.world_size)
+    jobs = [
+        JobConfig(job_id="mj_mlr", app="mlr", max_num_epochs=2,
+                  num_mini_batches=3,
+                  app_args={"num_classes": 4, "num_features": 32,
+                            "num_parts_per_class": 2, "batch_size": 64}),
+        JobConfig(job_id="mj_nmf", app="nmf", max_num_epochs=2,
+                  num_mini_batches=2,
+                  app_args={"num_cols": 128, "rank": 8, "nnz_per_row": 4,
+                            "rows_per_batch": 64}),
+    ]
+    tus = TaskUnitScheduler(cp, {j.job_id for j in jobs}, multi_job=True)
+    results = {}
+    errs = []
+
+    def run_one(job):
+        try:
+            results[job.job_id] = run_job(job, ctx, cp=cp, tus=tus).summary()
+        except Exception as e:  # noqa: BLE001
+            import traceback
+
+            errs.append(traceback.format_exc())
+
+    threads = [threading.Thread(target=run_one, args=(j,)) for j in jobs]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join(timeout=100)
+    assert not errs, errs[0]
+    return {jid: s["num_batches"] for jid, s in results.items()}
+
+
+def test_two_concurrent_jobs_two_ranks():
+    res = run_dist(_two_jobs_worker, world=2, timeout=180)
+    for r in res:
+        assert r["mj_mlr"] == 6
+        assert r["mj_nmf"] == 4
