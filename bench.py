@@ -1,0 +1,229 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: 3 concurrent PS jobs (NMF + MLR + LDA) sharing N MI355X.
+
+This measures the BASELINE.json headline metric — aggregate examples/sec
+(the reference's dataProcessingRate, WorkerTasklet.java:203) for three
+concurrent parameter-server jobs co-scheduled on the same GPUs, weak scaling
+(per-GPU batch work fixed as N grows).
+
+Launch (driver contract):
+  python bench.py --gpus N --steps K --warmup W
+  # N>1: python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+  #        --master-addr 127.0.0.1 --master-port P bench.py --gpus N ...
+
+One "step" = one mini-batch of EACH of the three jobs on every rank (the
+jobs run concurrently in tasklet threads on per-job HIP streams; NET phases
+are globally ticket-ordered). Data is synthetic, weights random-init.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import threading
+import time
+
+import torch
+import torch.distributed as dist
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=20)
+    p.add_argument("--warmup", type=int, default=5)
+    p.add_argument("--apps", type=str, default="nmf,mlr,lda")
+    p.add_argument("--device", type=str, default="auto")
+    # per-GPU shapes (weak scaling: these are PER RANK)
+    p.add_argument("--nmf-rank", type=int, default=100)
+    p.add_argument("--nmf-cols", type=int, default=65536)
+    p.add_argument("--nmf-rows-per-batch", type=int, default=16384)
+    p.add_argument("--nmf-nnz-per-row", type=int, default=128)
+    p.add_argument("--mlr-classes", type=int, default=10)
+    p.add_argument("--mlr-features", type=int, default=16384)
+    p.add_argument("--mlr-batch", type=int, default=16384)
+    p.add_argument("--lda-vocab", type=int, default=100000)
+    p.add_argument("--lda-topics", type=int, default=256)
+    p.add_argument("--lda-docs-per-batch", type=int, default=8192)
+    p.add_argument("--lda-tokens-per-doc", type=int, default=128)
+    return p.parse_args()
+
+
+def make_jobs(args, world: int):
+    from harmony_amd.config import JobConfig
+
+    n_blocks = 4  # resident synthetic blocks per rank, cycled
+    jobs = {}
+    if "nmf" in args.apps:
+        jobs["nmf"] = JobConfig(
+            job_id="bench_nmf", app="nmf", num_mini_batches=n_blocks,
+            num_worker_blocks=n_blocks,
+            app_args={"num_cols": args.nmf_cols, "rank": args.nmf_rank,
+                      "nnz_per_row": args.nmf_nnz_per_row,
+                      "rows_per_batch": args.nmf_rows_per_batch,
+                      "step_size": 0.01})
+    if "mlr" in args.apps:
+        jobs["mlr"] = JobConfig(
+            job_id="bench_mlr", app="mlr", num_mini_batches=n_blocks,
+            num_worker_blocks=n_blocks,
+            app_args={"num_classes": args.mlr_classes,
+                      "num_features": args.mlr_features,
+                      "num_parts_per_class": 8,
+                      "batch_size": args.mlr_batch, "step_size": 0.01})
+    if "lda" in args.apps:
+        jobs["lda"] = JobConfig(
+            job_id="bench_lda", app="lda", num_mini_batches=n_blocks,
+            num_worker_blocks=n_blocks,
+            app_args={"num_vocabs": args.lda_vocab,
+                      "num_topics": args.lda_topics,
+                      "tokens_per_doc": args.lda_tokens_per_doc,
+                      "docs_per_batch": args.lda_docs_per_batch})
+    return jobs
+
+
+class JobBench:
+    """One job's tasklet: builds tables/trainer/data, steps on command."""
+
+    def __init__(self, job, ctx, cp, tus, use_stream: bool):
+        from harmony_amd import mlapps
+
+        self.job = job
+        self.tus = tus
+        app = mlapps.get_app(job.app)
+        self.tables, self.trainer, self.provider = app.build(job, ctx, cp)
+        self.stream = (torch.cuda.Stream() if use_stream else None)
+        self.blocks = self.provider.blocks
+        self._phase = 0
+        self._i = 0
+        self.examples_per_batch = 0
+
+    def _next_phase(self):
+        self._phase += 1
+        return self._phase
+
+    def initialize(self):
+        import contextlib
+
+        sctx = (torch.cuda.stream(self.stream) if self.stream is not None
+                else contextlib.nullcontext())
+        with sctx:
+            with self.tus.net(self.job.job_id, self._next_phase()):
+                self.trainer.initialize()
+
+    def step(self):
+        import contextlib
+
+        sctx = (torch.cuda.stream(self.stream) if self.stream is not None
+                else contextlib.nullcontext())
+        with sctx:
+            batch = self.blocks[self._i % len(self.blocks)]
+            self._i += 1
+            self.trainer.set_batch_data(batch)
+            with self.tus.net(self.job.job_id, self._next_phase()):
+                self.trainer.pull_model()
+            self.trainer.local_compute()
+            with self.tus.net(self.job.job_id, self._next_phase()):
+                self.trainer.push_update()
+            self.examples_per_batch = self.trainer.num_batch_examples()
+
+
+def main():
+    args = parse_args()
+    from harmony_amd.config import RuntimeConfig
+    from harmony_amd.runtime.bootstrap import init_executor
+    from harmony_amd.runtime.control import ControlPlane, TaskUnitScheduler
+
+    ctx = init_executor(RuntimeConfig(device=args.device))
+    rank, world = ctx.rank, ctx.world_size
+    dev_cuda = ctx.device.type == "cuda"
+    cp = ControlPlane(ctx.store, rank, world)
+    jobs = make_jobs(args, world)
+    multi = len(jobs) > 1
+    tus = TaskUnitScheduler(cp, {j.job_id for j in jobs.values()},
+                            multi_job=multi)
+    benches = [JobBench(j, ctx, cp, tus, use_stream=dev_cuda)
+               for j in jobs.values()]
+    for b in benches:
+        b.initialize()
+
+    def run_steps(n: int):
+        errs = []
+
+        def worker(b):
+            try:
+                for _ in range(n):
+                    b.step()
+            except Exception:  # noqa: BLE001
+                import traceback
+
+                errs.append(traceback.format_exc())
+
+        threads = [threading.Thread(target=worker, args=(b,)) for b in benches]
+        for t in threads:
+            t.start()
+        for t in threads:
+            t.join()
+        if errs:
+            raise RuntimeError(errs[0])
+
+    def sync_all():
+        if dev_cuda:
+            torch.cuda.synchronize()
+        if dist.is_initialized():
+            dist.barrier()
+        if dev_cuda:
+            torch.cuda.synchronize()
+
+    run_steps(args.warmup)
+    sync_all()
+    t0 = time.perf_counter()
+    run_steps(args.steps)
+    sync_all()
+    elapsed = time.perf_counter() - t0
+
+    # MAX elapsed over ranks
+    if dist.is_initialized():
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device=ctx.device if dev_cuda else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.cpu())
+
+    examples_per_step_rank = sum(b.examples_per_batch for b in benches)
+    total_examples = examples_per_step_rank * args.steps * world
+    value = total_examples / elapsed
+    if rank == 0:
+        out = {
+            "metric": "aggregate_examples_per_sec_3job",
+            "value": value,
+            "unit": "examples/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1000.0,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": "+".join(jobs.keys()) + " concurrent PS jobs",
+                "global_batch": examples_per_step_rank * world,
+                "seq_len": None,
+                "parallelism": f"ps-dp{world}",
+                "nmf": {"rank": args.nmf_rank, "cols": args.nmf_cols,
+                        "rows_per_batch": args.nmf_rows_per_batch,
+                        "nnz_per_row": args.nmf_nnz_per_row},
+                "mlr": {"classes": args.mlr_classes,
+                        "features": args.mlr_features,
+                        "batch": args.mlr_batch},
+                "lda": {"vocab": args.lda_vocab, "topics": args.lda_topics,
+                        "docs_per_batch": args.lda_docs_per_batch,
+                        "tokens_per_doc": args.lda_tokens_per_doc},
+            },
+        }
+        print(json.dumps(out))
+
+
+if __name__ == "__main__":
+    main()

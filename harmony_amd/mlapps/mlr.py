@@ -48,7 +48,10 @@ def model_table_cfg(job: JobConfig, world_size: int) -> TableConfig:
         num_keys=a["num_classes"] * P,
         value_dim=a["num_features"] // P,
         dtype=a["dtype"],
-        num_blocks=max(world_size, min(64, a["num_classes"] * P)),
+        # one key per block (a partition row IS the unit of placement, as in
+        # the reference's per-partition table entries) -> no keyspace padding,
+        # clean [C*P, F/P] <-> [C, F] views.
+        num_blocks=a["num_classes"] * P,
         update_fn="add",
         init_fn="gaussian",
         init_args={"std": 0.01},
@@ -91,7 +94,8 @@ class MLRTrainer(Trainer):
         return self.W.view(C, F)
 
     def pull_model(self) -> None:
-        self.W = self.accessor.pull_all()
+        C, P = self.a["num_classes"], self.a["num_parts_per_class"]
+        self.W = self.accessor.pull_all()[:C * P]
 
     def local_compute(self) -> None:
         x, y = self.batch
