@@ -27,6 +27,7 @@ from typing import Any, Callable, Dict, List, Optional
 import torch
 
 from harmony_amd.config import TableConfig, dtype_of
+from harmony_amd import ops
 from harmony_amd.et import update_functions as uf
 from harmony_amd.et.ownership import Ownership
 from harmony_amd.et.partitioner import HashBasedPartitioner, OrderingBasedPartitioner
@@ -122,6 +123,13 @@ class Table:
         after multiUpdate merging (TableImpl.java:460, BlockImpl.update:71).
         """
         rows = self.local_rows_of(keys)
+        if self.device.type == "cuda" and ops.fused_apply_supported(self.cfg.update_fn):
+            # fused gather-apply-scatter kernel (K3/K9)
+            ops.scatter_apply(self.shard, rows, deltas.to(self.dtype),
+                              self.cfg.update_fn,
+                              self.cfg.update_args.get("step_size", 0.0),
+                              self.cfg.update_args.get("max_val", 0.0))
+            return
         fn = uf.update_fn(self.cfg.update_fn)
         vals = self.shard[rows]
         self.shard[rows] = fn(vals, deltas.to(vals.dtype), **self.cfg.update_args)
@@ -129,6 +137,12 @@ class Table:
     def apply_update_dense_local(self, agg_delta: torch.Tensor) -> None:
         """Update the whole local shard with an aggregated dense delta
         (the fused epilogue of a reduce-scatter push)."""
+        if self.device.type == "cuda" and ops.fused_apply_supported(self.cfg.update_fn):
+            ops.dense_apply(self.shard, agg_delta.to(self.dtype),
+                            self.cfg.update_fn,
+                            self.cfg.update_args.get("step_size", 0.0),
+                            self.cfg.update_args.get("max_val", 0.0))
+            return
         fn = uf.update_fn(self.cfg.update_fn)
         fn(self.shard, agg_delta.to(self.shard.dtype), **self.cfg.update_args)
 

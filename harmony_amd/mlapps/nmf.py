@@ -59,10 +59,10 @@ class NMFBatch:
     """One mini-batch: CSR nonzeros of a contiguous range of local L rows,
     with the batch's unique columns precomputed (static per block)."""
 
-    def __init__(self, l_rows: torch.Tensor, row_idx: torch.Tensor,
+    def __init__(self, l_rows: torch.Tensor, row_ptr: torch.Tensor,
                  col_idx: torch.Tensor, vals: torch.Tensor):
         self.l_rows = l_rows            # [n_rows] indices into local L
-        self.row_idx = row_idx          # [nnz] index into l_rows (0..n_rows)
+        self.row_ptr = row_ptr          # [n_rows+1] CSR offsets over nonzeros
         self.col_idx = col_idx          # [nnz] global column key
         self.vals = vals                # [nnz]
         self.uniq_cols, self.col_local = torch.unique(col_idx, return_inverse=True)
@@ -79,10 +79,10 @@ def make_batches(job: JobConfig, rank: int, device: torch.device):
         lo = b * a["rows_per_batch"]
         l_rows = torch.arange(lo, lo + a["rows_per_batch"])
         nnz = a["rows_per_batch"] * a["nnz_per_row"]
-        row_idx = torch.arange(a["rows_per_batch"]).repeat_interleave(a["nnz_per_row"])
+        row_ptr = torch.arange(0, nnz + 1, a["nnz_per_row"])
         col_idx = torch.randint(0, a["num_cols"], (nnz,), generator=g)
         vals = torch.rand(nnz, generator=g)
-        blocks.append(NMFBatch(l_rows.to(device), row_idx.to(device),
+        blocks.append(NMFBatch(l_rows.to(device), row_ptr.to(device),
                                col_idx.to(device), vals.to(device)))
     return blocks, rows_local
 
@@ -108,7 +108,7 @@ class NMFTrainer(Trainer):
     def local_compute(self) -> None:
         b = self.batch
         L_batch = self.L[b.l_rows]
-        lgrad, rgrad, sq = ops.nmf_grad(L_batch, self.R_batch, b.row_idx,
+        lgrad, rgrad, sq = ops.nmf_grad(L_batch, self.R_batch, b.row_ptr,
                                         b.col_local, b.vals, self.a["lam"])
         # local L update (worker-side SGD apply, same rule as the server's)
         self.L[b.l_rows] = (L_batch - self.step_size * lgrad).clamp_(

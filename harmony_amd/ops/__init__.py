@@ -100,18 +100,21 @@ def softmax_grad_ce(logits: torch.Tensor, labels: torch.Tensor
 # K1: NMF gradient over a sparse batch (CSR by row)
 # ---------------------------------------------------------------------------
 
-def nmf_grad(L: torch.Tensor, R: torch.Tensor, row_idx: torch.Tensor,
+def nmf_grad(L: torch.Tensor, R: torch.Tensor, row_ptr: torch.Tensor,
              col_idx: torch.Tensor, vals: torch.Tensor, lam: float
              ) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
     """For each nonzero (i, j, v): e = L_i . R_j - v;
     lgrad_i += 2 e R_j, rgrad_j += 2 e L_i; plus L2 terms 2*lam*{L_i,R_j}
     per nonzero (reference NMFTrainer.updateGradient:328-367).
-    L: [n_rows, k] (rows indexed by row_idx), R: [n_cols, k] (indexed by
-    col_idx). Returns (lgrad [n_rows,k], rgrad [n_cols,k], sq_err_sum)."""
+    L: [n_rows, k], CSR over rows: row_ptr [n_rows+1], col_idx/vals [nnz]
+    (col_idx indexes R's rows). Returns (lgrad, rgrad, sq_err_sum)."""
     if _use_hip(L):
-        return _hip.nmf_grad(L.contiguous(), R.contiguous(),
-                             row_idx.contiguous(), col_idx.contiguous(),
-                             vals.contiguous(), float(lam))
+        return tuple(_hip.nmf_grad(L.contiguous(), R.contiguous(),
+                                   row_ptr.contiguous(), col_idx.contiguous(),
+                                   vals.contiguous(), float(lam)))
+    row_idx = torch.repeat_interleave(
+        torch.arange(L.shape[0], device=L.device),
+        row_ptr[1:] - row_ptr[:-1])
     Lr = L[row_idx]                       # [nnz, k]
     Rr = R[col_idx]
     e = (Lr * Rr).sum(dim=1) - vals       # [nnz]
@@ -145,15 +148,20 @@ def lda_gibbs(doc_topic: torch.Tensor, word_topic: torch.Tensor,
     p(k) ∝ (n_dk + α) (n_wk + β) / (n_k + V β)
     (reference SparseLDASampler.java:141-274's s/r/q bucket decomposition is a
     CPU sparsity optimization; on CDNA4 the dense K-way distribution is
-    computed by a wave per document — see ops/csrc/lda.hip.)"""
+    computed by a wave per document — see ops/csrc/lda.hip.)
+
+    The torch path mirrors the device kernel: same counter-based RNG keyed by
+    (seed, token index), same float32 probability terms, same "first k with
+    cumsum > u" draw — CPU and GPU agree except at float summation-order
+    tie-breaks (tests require >=99% identical samples + exact invariants)."""
     if _use_hip(word_topic):
         return _hip.lda_gibbs(doc_topic, word_topic, topic_sum, doc_offsets,
                               word_ids, assignments, float(alpha), float(beta),
                               int(num_vocabs), int(seed))
-    K = word_topic.shape[1]
+    from harmony_amd.ops.rng import rng_uniform
+
     n_docs = doc_topic.shape[0]
-    g = torch.Generator(device="cpu").manual_seed(seed & 0x7FFFFFFF)
-    denom_base = topic_sum.float() + num_vocabs * beta      # [K]
+    inv_den = (1.0 / (topic_sum.float() + num_vocabs * beta)).float()   # [K]
     # Lockstep over token positions: docs advance one token per step so the
     # per-doc sequential dependency (n_dk) is honored while steps stay
     # vectorized over docs.
@@ -164,16 +172,52 @@ def lda_gibbs(doc_topic: torch.Tensor, word_topic: torch.Tensor,
         tok = doc_offsets[active] + pos
         w = word_ids[tok].long()
         old = assignments[tok].long()
-        # remove token from its current topic (doc side only; word side is
-        # batch-stale by design)
+        ar = torch.arange(active.shape[0])
         dt = doc_topic[active]
-        dt[torch.arange(active.shape[0]), old] -= 1
-        probs = (dt.float() + alpha) * (word_topic[w].float() + beta) / denom_base
-        new = torch.multinomial(probs.clamp_min(1e-30), 1, generator=g).squeeze(1)
-        dt[torch.arange(active.shape[0]), new] += 1
+        dt[ar, old] -= 1
+        probs = ((dt.float() + alpha) * (word_topic[w].float() + beta)
+                 * inv_den)                                  # [n_active, K]
+        tot = probs.sum(dim=1)
+        u = rng_uniform(seed & 0xFFFFFFFF, tok.long()) * tot
+        cdf = probs.cumsum(dim=1)
+        new = torch.searchsorted(cdf, u.unsqueeze(1).to(cdf.dtype),
+                                 right=True).squeeze(1)
+        bad = new >= probs.shape[1]
+        new = torch.where(bad, old, new)                     # numeric edge
+        dt[ar, new] += 1
         doc_topic[active] = dt
         assignments[tok] = new.to(assignments.dtype)
     return assignments
+
+
+# ---------------------------------------------------------------------------
+# K3/K9: fused owner-side update application ("server" compute)
+# ---------------------------------------------------------------------------
+
+# update-function name (et.update_functions) -> device kernel mode
+_APPLY_MODES = {"add": 0, "assign": 1, "nmf_sgd": 2, "lda_counts": 3}
+
+
+def fused_apply_supported(update_fn_name: str) -> bool:
+    return update_fn_name in _APPLY_MODES
+
+
+def scatter_apply(shard: torch.Tensor, rows: torch.Tensor,
+                  deltas: torch.Tensor, update_fn_name: str,
+                  step_size: float = 0.0, max_val: float = 0.0) -> None:
+    """shard[rows] = f(shard[rows], deltas) in one kernel (GPU only)."""
+    assert _use_hip(shard), "scatter_apply is the GPU fused path"
+    _hip.scatter_apply(shard, rows.contiguous(), deltas.contiguous(),
+                       _APPLY_MODES[update_fn_name], float(step_size),
+                       float(max_val))
+
+
+def dense_apply(shard: torch.Tensor, delta: torch.Tensor,
+                update_fn_name: str, step_size: float = 0.0,
+                max_val: float = 0.0) -> None:
+    assert _use_hip(shard), "dense_apply is the GPU fused path"
+    _hip.dense_apply(shard, delta.contiguous(), _APPLY_MODES[update_fn_name],
+                     float(step_size), float(max_val))
 
 
 # ---------------------------------------------------------------------------

@@ -1,0 +1,30 @@
+// Python bindings for the CDNA4 kernel set (built in-tree as
+// harmony_amd/ops/_hip_ops by setup_ops.py; PYTORCH_ROCM_ARCH=gfx950).
+
+#include <torch/extension.h>
+
+std::vector<torch::Tensor> mlr_softmax_grad(torch::Tensor logits,
+                                            torch::Tensor labels);
+std::vector<torch::Tensor> nmf_grad(torch::Tensor L, torch::Tensor R,
+                                    torch::Tensor row_ptr,
+                                    torch::Tensor col_idx, torch::Tensor vals,
+                                    double lam);
+torch::Tensor lda_gibbs(torch::Tensor doc_topic, torch::Tensor word_topic,
+                        torch::Tensor topic_sum, torch::Tensor doc_offsets,
+                        torch::Tensor word_ids, torch::Tensor assignments,
+                        double alpha, double beta, int64_t num_vocabs,
+                        int64_t seed);
+void scatter_apply(torch::Tensor shard, torch::Tensor rows,
+                   torch::Tensor deltas, int64_t mode, double step,
+                   double maxval);
+void dense_apply(torch::Tensor shard, torch::Tensor delta, int64_t mode,
+                 double step, double maxval);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("mlr_softmax_grad", &mlr_softmax_grad,
+        "fused softmax + label-subtract + CE/accuracy (K4)");
+  m.def("nmf_grad", &nmf_grad, "NMF sparse-batch gradient (K1+K2)");
+  m.def("lda_gibbs", &lda_gibbs, "LDA collapsed Gibbs sweep (K7)");
+  m.def("scatter_apply", &scatter_apply, "owner-side sparse update (K9)");
+  m.def("dense_apply", &dense_apply, "owner-side dense update (K3)");
+}
