@@ -23,6 +23,8 @@ import torch.distributed as dist
 
 from harmony_amd.config import RuntimeConfig
 
+_STORE_KEEPALIVE: list = []
+
 
 @dataclass
 class ExecutorContext:
@@ -82,6 +84,10 @@ def init_executor(cfg: Optional[RuntimeConfig] = None) -> ExecutorContext:
         store = TCPStore(master_addr, master_port + 7, world,
                          is_master=(rank == 0),
                          timeout=datetime.timedelta(seconds=300))
+        # Pin the store for the process lifetime: if rank 0's TCPStore object
+        # is GC'd when the caller's frame dies, the server socket closes and
+        # every peer still polling it crashes.
+        _STORE_KEEPALIVE.append(store)
     else:
         backend = "none"
         store = LocalStore()

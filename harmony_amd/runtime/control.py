@@ -42,6 +42,8 @@ class ControlPlane:
         arrived = self.store.add(f"bar/{name}/a", 1)
         epoch = (arrived - 1) // n + 1
         target = epoch * n
+        if arrived == target:       # last arriver: no poll (the store master
+            return                  # may exit right after its own barrier)
         while int(self.store.add(f"bar/{name}/a", 0)) < target:
             time.sleep(0.0005)
 

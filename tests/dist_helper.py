@@ -30,6 +30,11 @@ def _entry(fn, rank, world, port, q, args):
         import torch.distributed as dist
 
         if dist.is_initialized():
+            try:
+                # keep the rank-0 control store alive until everyone is done
+                dist.barrier()
+            except Exception:  # noqa: BLE001
+                pass
             dist.destroy_process_group()
 
 
@@ -43,9 +48,23 @@ def run_dist(fn, world: int = 2, args=(), timeout: int = 120):
     for p in procs:
         p.start()
     results = {}
+    import queue as queue_mod
+    import time
+
     try:
-        for _ in range(world):
-            rank, status, payload = q.get(timeout=timeout)
+        deadline = time.monotonic() + timeout
+        while len(results) < world:
+            try:
+                rank, status, payload = q.get(timeout=2)
+            except queue_mod.Empty:
+                dead = [(i, p.exitcode) for i, p in enumerate(procs)
+                        if not p.is_alive() and p.exitcode != 0]
+                if dead:
+                    raise RuntimeError(f"worker died: (rank, exitcode)={dead}")
+                if time.monotonic() > deadline:
+                    raise TimeoutError(f"ranks {sorted(set(range(world)) - set(results))} "
+                                       f"did not report within {timeout}s")
+                continue
             if status == "err":
                 raise RuntimeError(f"rank {rank} failed:\n{payload}")
             results[rank] = pickle.loads(payload)
