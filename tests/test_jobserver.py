@@ -1,0 +1,115 @@
+"""Job server integration: submit over the socket, run, wait, shutdown
+(the analogue of the reference's REEF local-runtime jobserver tests)."""
+
+import json
+import socket
+import threading
+import time
+
+from tests.dist_helper import free_port, run_dist
+
+
+def test_jobserver_single_process():
+    from harmony_amd.config import JobConfig, RuntimeConfig
+    from harmony_amd.jobserver import client
+    from harmony_amd.jobserver.server import JobServerDriver
+    from harmony_amd.runtime.bootstrap import init_executor
+
+    port = free_port()
+    ctx = init_executor(RuntimeConfig(device="cpu"))
+    driver = JobServerDriver(ctx, scheduler="default", port=port)
+    t = threading.Thread(target=driver.run, daemon=True)
+    t.start()
+    time.sleep(0.3)
+
+    job = JobConfig(job_id="js1", app="addvector", max_num_epochs=2,
+                    num_mini_batches=3, app_args={"num_keys": 16,
+                                                  "vector_dim": 4})
+    res = client.submit(job, port=port, wait=True, timeout=60)
+    assert res["status"] == "done", res
+    assert res["per_rank"][0]["num_batches"] == 6
+
+    # second concurrent pair of jobs
+    j2 = JobConfig(job_id="js2", app="mlr", max_num_epochs=1,
+                   num_mini_batches=2,
+                   app_args={"num_classes": 3, "num_features": 16,
+                             "num_parts_per_class": 2, "batch_size": 32})
+    j3 = JobConfig(job_id="js3", app="nmf", max_num_epochs=1,
+                   num_mini_batches=2,
+                   app_args={"num_cols": 64, "rank": 8, "nnz_per_row": 4,
+                             "rows_per_batch": 32})
+    client.submit(j2, port=port)
+    client.submit(j3, port=port)
+    r2 = client._send({"cmd": "WAIT", "job_id": "js2", "timeout": 60}, port)
+    r3 = client._send({"cmd": "WAIT", "job_id": "js3", "timeout": 60}, port)
+    assert r2["status"] == "done" and r3["status"] == "done"
+
+    out = client.shutdown(port=port)
+    assert out["status"] == "ok"
+    t.join(timeout=30)
+    assert not t.is_alive()
+
+
+def _jobserver_2rank_worker(rank, world):
+    from harmony_amd.config import JobConfig, RuntimeConfig
+    from harmony_amd.jobserver import client
+    from harmony_amd.jobserver.server import JobServerDriver
+    from harmony_amd.runtime.bootstrap import init_executor
+    import os
+
+    port = 7100 + int(os.environ["MASTER_PORT"]) % 500
+    ctx = init_executor(RuntimeConfig(device="cpu"))
+    driver = JobServerDriver(ctx, scheduler="default", port=port)
+
+    if rank == 0:
+        def submit_then_stop():
+            time.sleep(0.5)
+            job = JobConfig(job_id="js2r", app="addvector", max_num_epochs=2,
+                            num_mini_batches=2,
+                            app_args={"num_keys": 16, "vector_dim": 4})
+            res = client.submit(job, port=port, wait=True, timeout=90)
+            assert res["status"] == "done", res
+            # both ranks ran 4 batches; every worker pushed to every key
+            assert all(s["num_batches"] == 4 for s in res["per_rank"])
+            client.shutdown(port=port)
+
+        threading.Thread(target=submit_then_stop, daemon=True).start()
+    driver.run()
+    return True
+
+
+def test_jobserver_two_ranks():
+    assert all(run_dist(_jobserver_2rank_worker, world=2, timeout=150))
+
+
+def _jobserver_subset_worker(rank, world):
+    """least_loaded scheduler places a 1-executor job on one rank only."""
+    from harmony_amd.config import JobConfig, RuntimeConfig
+    from harmony_amd.jobserver import client
+    from harmony_amd.jobserver.server import JobServerDriver
+    from harmony_amd.runtime.bootstrap import init_executor
+    import os
+
+    port = 7600 + int(os.environ["MASTER_PORT"]) % 300
+    ctx = init_executor(RuntimeConfig(device="cpu"))
+    driver = JobServerDriver(ctx, scheduler="least_loaded", port=port)
+
+    if rank == 0:
+        def submit_then_stop():
+            time.sleep(0.5)
+            job = JobConfig(job_id="sub1", app="addvector", max_num_epochs=1,
+                            num_mini_batches=2,
+                            app_args={"num_keys": 8, "vector_dim": 2,
+                                      "num_executors": 1})
+            res = client.submit(job, port=port, wait=True, timeout=90)
+            assert res["status"] == "done", res
+            assert len(res["per_rank"]) == 1   # ran on exactly one executor
+            client.shutdown(port=port)
+
+        threading.Thread(target=submit_then_stop, daemon=True).start()
+    driver.run()
+    return True
+
+
+def test_jobserver_subset_scheduling():
+    assert all(run_dist(_jobserver_subset_worker, world=2, timeout=150))
