@@ -92,6 +92,9 @@ class JobConfig:
     num_trainer_threads: int = 1
     app_args: Dict[str, Any] = field(default_factory=dict)
     model_is_local: bool = False       # also create a local-model table
+    optimizer: Optional[str] = None    # elasticity optimizer ("homogeneous"
+                                       # or "module:Class"); None = off
+    optimizer_period: int = 8          # batches between optimization windows
 
     def to_json(self) -> str:
         return json.dumps(dataclasses.asdict(self))

@@ -22,10 +22,24 @@ from harmony_amd.runtime.bootstrap import ExecutorContext
 from harmony_amd.runtime.control import ControlPlane, TaskUnitScheduler
 
 
+def _make_optimizer(name: str):
+    from harmony_amd.optimizer.optimizers import HomogeneousCostOptimizer
+
+    if name == "homogeneous":
+        return HomogeneousCostOptimizer()
+    if ":" in name:
+        import importlib
+
+        mod, cls = name.split(":", 1)
+        return getattr(importlib.import_module(mod), cls)()
+    raise KeyError(f"unknown optimizer '{name}'")
+
+
 def run_job(job: JobConfig, ctx: ExecutorContext,
             cp: Optional[ControlPlane] = None,
             tus: Optional[TaskUnitScheduler] = None,
-            stream: Optional[torch.cuda.Stream] = None) -> MetricCollector:
+            stream: Optional[torch.cuda.Stream] = None,
+            optimizer=None) -> MetricCollector:
     """Run one PS job to completion on this rank; returns this rank's metrics.
 
     Collective: every rank of the job's executor set must call this with the
@@ -35,8 +49,19 @@ def run_job(job: JobConfig, ctx: ExecutorContext,
     tus = tus or TaskUnitScheduler(cp, {job.job_id}, multi_job=False)
     app = mlapps.get_app(job.app)
     tables, trainer, provider = app.build(job, ctx, cp)
+    orch = None
+    if optimizer is None and job.optimizer:
+        optimizer = _make_optimizer(job.optimizer)
+    if optimizer is not None:
+        from harmony_amd.optimizer.orchestrator import OptimizationOrchestrator
+
+        orch = OptimizationOrchestrator(
+            cp, job.job_id, ctx.rank, ctx.world_size, tables,
+            optimizer=optimizer, check_period=job.optimizer_period,
+            group=getattr(ctx, "group", None))
     tasklet = WorkerTasklet(job, trainer, provider, cp, tus,
-                            ctx.rank, ctx.world_size, stream=stream)
+                            ctx.rank, ctx.world_size, stream=stream,
+                            orchestrator=orch)
     metrics = tasklet.run()
     ev = trainer.evaluate_model()
     for k, v in (ev or {}).items():

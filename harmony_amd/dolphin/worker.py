@@ -30,7 +30,8 @@ class WorkerTasklet:
     def __init__(self, job: JobConfig, trainer: Trainer,
                  provider: TrainingDataProvider, cp: ControlPlane,
                  tus: TaskUnitScheduler, rank: int, world_size: int,
-                 stream: Optional[torch.cuda.Stream] = None):
+                 stream: Optional[torch.cuda.Stream] = None,
+                 orchestrator=None):
         self.job = job
         self.trainer = trainer
         self.provider = provider
@@ -39,6 +40,7 @@ class WorkerTasklet:
         self.rank = rank
         self.world_size = world_size
         self.stream = stream
+        self.orch = orchestrator   # optimizer.OptimizationOrchestrator
         self.metrics = MetricCollector(job.job_id, rank)
         self._phase = 0
         self.ssp = SSPClock(cp, job.job_id, world_size, job.clock_slack)
@@ -87,11 +89,21 @@ class WorkerTasklet:
                     push_t = time.perf_counter() - t0
                     n = self.trainer.num_batch_examples()
                     ep_examples += n
+                    bt = time.perf_counter() - b_t0
                     self.metrics.add_batch(BatchMetrics(
                         epoch_idx=epoch, batch_idx=bidx, num_examples=n,
-                        batch_time_sec=time.perf_counter() - b_t0,
+                        batch_time_sec=bt,
                         pull_time_sec=pull_t, comp_time_sec=comp_t,
                         push_time_sec=push_t))
+                    # optimization window (reference RUN<->OPTIMIZE state):
+                    # a quiesced gap between batches where an elasticity
+                    # plan (block migration) applies collectively
+                    if self.orch is not None:
+                        self.orch.report_batch(bt, comp_t, pull_t, push_t, n)
+                        plan = self.orch.boundary_plan()
+                        if plan is not None:
+                            with self.tus.net(jid, self._next_phase()):
+                                self.orch.apply(plan)
                 self.trainer.on_epoch_finished(epoch)
                 self.metrics.add_epoch(EpochMetrics(
                     epoch_idx=epoch, num_examples=ep_examples,

@@ -1,0 +1,118 @@
+"""Optimization orchestrator: collect metrics -> optimize -> publish plan ->
+all ranks apply it collectively at the same batch boundary.
+
+Reference: dolphin/optimizer/impl/ETOptimizationOrchestrator.java:36 —
+background loop gathering EMA'd worker/server metrics, running the
+Optimizer, compiling and executing the plan inside the
+WorkerStateManager's RUN<->OPTIMIZE window.
+
+MI355X shape: decisions are made on the job's rank 0 from per-rank metric
+summaries published through the control store; the chosen plan is published
+with an `apply_at` batch index (a multiple of the check period). Every rank
+checks at those boundaries and executes the plan inside a NET-phase ticket —
+the optimization window is the quiesced gap between two mini-batches, so no
+update can straddle a migration (ownership-first correctness for free).
+"""
+
+from __future__ import annotations
+
+import json
+from typing import Dict, List, Optional
+
+from harmony_amd.optimizer.optimizers import Optimizer, RankMetrics
+from harmony_amd.optimizer.plan import Plan, PlanExecutor
+from harmony_amd.runtime.control import ControlPlane
+
+
+class OptimizationOrchestrator:
+    """One per (job, rank). Worker loop calls on_batch_boundary() between
+    batches; rank 0 additionally decides and publishes plans."""
+
+    def __init__(self, cp: ControlPlane, job_id: str, rank: int,
+                 world_size: int, tables: Dict[str, object],
+                 optimizer: Optional[Optimizer] = None,
+                 check_period: int = 8, min_metrics: int = 1,
+                 ema: float = 0.5, group=None):
+        self.cp = cp
+        self.job_id = job_id
+        self.rank = rank
+        self.world_size = world_size
+        self.tables = tables
+        self.optimizer = optimizer
+        self.period = max(1, check_period)
+        self.min_metrics = min_metrics
+        self.ema_w = ema
+        self.executor = PlanExecutor(tables, rank, world_size, group=group)
+        self._ema: Dict[int, RankMetrics] = {}
+        self._batches = 0
+        self._windows = 0
+        self.applied_plans = 0
+
+    # ---------------------------------------------------------- metric flow
+
+    def report_batch(self, batch_time: float, comp: float, pull: float,
+                     push: float, n_examples: int) -> None:
+        self._batches += 1
+        key = f"opt/{self.job_id}/m/{self.rank}/{self._batches}"
+        if self._batches % self.period == 0:
+            self.cp.store.set(key, json.dumps(
+                [batch_time, comp, pull, push, n_examples]))
+
+    def _gather_metrics(self, at_batch: int) -> Optional[List[RankMetrics]]:
+        out = []
+        for r in range(self.world_size):
+            key = f"opt/{self.job_id}/m/{r}/{at_batch}"
+            if not self.cp.flag_set(key):
+                return None   # skip round if any executor's window missing
+            b, c, pl, ps, n = json.loads(self.cp.store.get(key))
+            prev = self._ema.get(r)
+            if prev is None:
+                m = RankMetrics(r, b, c, pl, ps, n)
+            else:  # EMA (reference MetricProcessor MetricWeightFactor)
+                w = self.ema_w
+                m = RankMetrics(r, w * b + (1 - w) * prev.batch_time_sec,
+                                w * c + (1 - w) * prev.comp_time_sec,
+                                w * pl + (1 - w) * prev.pull_time_sec,
+                                w * ps + (1 - w) * prev.push_time_sec, n)
+            self._ema[r] = m
+            out.append(m)
+        return out
+
+    # ------------------------------------------------------------- decision
+
+    def boundary_plan(self) -> Optional[Plan]:
+        """Call between batches on EVERY rank (after report_batch). Rank 0
+        runs the optimizer; returns the plan scheduled for THIS boundary if
+        one exists (identical answer on every rank — the key is derived from
+        the deterministic batch counter). The caller applies it with
+        apply() inside a NET-phase ticket."""
+        if self._batches % self.period != 0:
+            return None
+        at = self._batches
+        if self.rank == 0 and self.optimizer is not None:
+            self._decide(at, plan_key_base=f"opt/{self.job_id}/plan")
+        my_key = f"opt/{self.job_id}/plan/{at}"
+        if self.cp.flag_set(my_key):
+            plan = Plan.from_json(self.cp.store.get(my_key).decode())
+            if not plan.empty():
+                return plan
+        return None
+
+    def apply(self, plan: Plan) -> None:
+        """Collective: execute the plan (inside the caller's NET ticket)."""
+        self.executor.execute(plan)
+        self.applied_plans += 1
+
+    def _decide(self, at: int, plan_key_base: str) -> None:
+        self._windows += 1
+        if self._windows < self.min_metrics:
+            return
+        metrics = self._gather_metrics(at)
+        if metrics is None:
+            return
+        owners = {t.cfg.table_id: t.ownership.owner.tolist()
+                  for t in self.tables.values() if hasattr(t, "cfg")}
+        plan = self.optimizer.optimize(metrics, owners, self.world_size)
+        if plan is not None and not plan.empty():
+            self.cp.store.set(f"{plan_key_base}/{at + self.period}",
+                              plan.to_json())
