@@ -173,15 +173,15 @@ class DataPlane:
     # -------------------------------------------------------------- push keys
 
     def push_keys(self, table, keys: torch.Tensor, deltas: torch.Tensor) -> None:
+        from harmony_amd.et.update_functions import merge_key_deltas
+
         keys = keys.to(self.device, torch.int64)
         deltas = deltas.to(self.device)
         # Aggregate locally first (reference CommManager serializes per-block
-        # writes; summing before the wire preserves update semantics because
-        # every registered update function is delta-merge associative).
-        uniq, inv = torch.unique(keys, return_inverse=True)
-        agg = torch.zeros((uniq.shape[0], deltas.shape[1]), dtype=deltas.dtype,
-                          device=self.device)
-        agg.index_add_(0, inv, deltas)
+        # writes; merging before the wire preserves update semantics because
+        # every registered update function is delta-merge associative — the
+        # merge algebra (sum/min/last) comes from the update fn's MERGE_MODE).
+        uniq, agg = merge_key_deltas(keys, deltas, table.cfg.update_fn)
         sorted_keys, order, send_counts = self._route(table, uniq)
         sorted_deltas = agg[order]
         recv_counts = self._exchange_counts(send_counts.to("cpu"))
@@ -191,10 +191,8 @@ class DataPlane:
         if recv_keys.numel() == 0:
             return
         # Aggregate across source ranks, then one update-fn apply per key.
-        u2, inv2 = torch.unique(recv_keys, return_inverse=True)
-        agg2 = torch.zeros((u2.shape[0], recv_deltas.shape[1]),
-                           dtype=recv_deltas.dtype, device=self.device)
-        agg2.index_add_(0, inv2, recv_deltas)
+        u2, agg2 = merge_key_deltas(recv_keys, recv_deltas,
+                                    table.cfg.update_fn)
         table.update_local(u2, agg2)
 
     # ---------------------------------------------------------- object tables

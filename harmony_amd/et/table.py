@@ -164,7 +164,7 @@ class Table:
         """multiUpdate (reference TableImpl.java:460): route deltas to owner
         ranks; the owner applies the table's update function."""
         if self._local_only():
-            keys, deltas = _aggregate(keys, deltas)
+            keys, deltas = uf.merge_key_deltas(keys, deltas, self.cfg.update_fn)
             self.update_local(keys, deltas)
             return
         self.comm.push_keys(self, keys, deltas)
@@ -286,15 +286,6 @@ class ObjectTable:
 
     def adopt_blocks(self, blocks: Dict[int, Dict[int, Any]]) -> None:
         self.blocks.update(blocks)
-
-
-def _aggregate(keys: torch.Tensor, deltas: torch.Tensor):
-    """Sum deltas of duplicate keys -> (unique_keys, summed_deltas)."""
-    uniq, inv = torch.unique(keys, return_inverse=True)
-    out = torch.zeros((uniq.shape[0], deltas.shape[1]), dtype=deltas.dtype,
-                      device=deltas.device)
-    out.index_add_(0, inv, deltas)
-    return uniq, out
 
 
 def _block_seed(table_id: str, block_id: int) -> int:

@@ -119,5 +119,50 @@ def _update_lda(values: torch.Tensor, delta: torch.Tensor, **_) -> torch.Tensor:
     return values
 
 
+@register_update("min")
+def _update_min(values: torch.Tensor, delta: torch.Tensor, **_) -> torch.Tensor:
+    # Min-combiner (Pregel shortest-path message semantics).
+    torch.minimum(values, delta.to(values.dtype), out=values)
+    return values
+
+
+# How duplicate-key deltas are merged BEFORE the update function applies
+# (push aggregation in et/comm.py). Must match the update function's algebra:
+# f(f(v,a),b) == f(v, merge(a,b)).
+MERGE_MODE = {
+    "add": "sum",
+    "assign": "last",
+    "nmf_sgd": "sum",
+    "lda_counts": "sum",
+    "min": "min",
+}
+
+
+def merge_key_deltas(keys: torch.Tensor, deltas: torch.Tensor,
+                     update_fn_name: str):
+    """Aggregate duplicate keys per the update function's merge algebra ->
+    (unique_keys, merged_deltas)."""
+    mode = MERGE_MODE.get(update_fn_name, "sum")
+    uniq, inv = torch.unique(keys, return_inverse=True)
+    if mode == "last":
+        out = torch.empty((uniq.shape[0], deltas.shape[1]), dtype=deltas.dtype,
+                          device=deltas.device)
+        out[inv] = deltas          # later duplicates win (any is valid)
+        return uniq, out
+    if mode == "min":
+        out = torch.full((uniq.shape[0], deltas.shape[1]),
+                         torch.finfo(deltas.dtype).max
+                         if deltas.is_floating_point() else
+                         torch.iinfo(deltas.dtype).max,
+                         dtype=deltas.dtype, device=deltas.device)
+        out.scatter_reduce_(0, inv.unsqueeze(1).expand_as(deltas), deltas,
+                            reduce="amin")
+        return uniq, out
+    out = torch.zeros((uniq.shape[0], deltas.shape[1]), dtype=deltas.dtype,
+                      device=deltas.device)
+    out.index_add_(0, inv, deltas)
+    return uniq, out
+
+
 ALL_INIT = _INIT_FNS
 ALL_UPDATE = _UPDATE_FNS

@@ -164,11 +164,17 @@ class JobServerDriver:
 
     def _run_job_thread(self, job: JobConfig, view: JobView) -> None:
         from harmony_amd.dolphin.master import run_job
+        from harmony_amd.pregel.runner import PREGEL_APPS, run_pregel_job
 
         try:
-            stream = (torch.cuda.Stream() if view.device.type == "cuda" else None)
-            metrics = run_job(job, view, cp=self.cp, tus=self.tus, stream=stream)
-            summary = metrics.summary()
+            if job.app in PREGEL_APPS:
+                summary = run_pregel_job(job, view, cp=self.cp, tus=self.tus)
+            else:
+                stream = (torch.cuda.Stream()
+                          if view.device.type == "cuda" else None)
+                metrics = run_job(job, view, cp=self.cp, tus=self.tus,
+                                  stream=stream)
+                summary = metrics.summary()
         except Exception as e:  # noqa: BLE001
             traceback.print_exc()
             summary = {"status": "failed", "error": str(e), "rank": view.rank}

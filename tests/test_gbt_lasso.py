@@ -1,0 +1,70 @@
+import torch
+
+from harmony_amd.config import JobConfig, RuntimeConfig
+from harmony_amd.dolphin.master import run_job
+from harmony_amd.runtime.bootstrap import init_executor
+from tests.dist_helper import run_dist
+
+
+def test_lasso_recovers_sparse_signal():
+    job = JobConfig(job_id="t_lasso", app="lasso", max_num_epochs=4,
+                    num_mini_batches=2,
+                    app_args={"num_features": 64, "num_parts": 8,
+                              "batch_size": 512, "lam": 0.02,
+                              "density": 0.2, "noise": 0.05})
+    ctx = init_executor(RuntimeConfig(device="cpu"))
+    m = run_job(job, ctx)
+    s = m.summary()
+    assert s["num_batches"] == 8
+    # converged to small residual: mse near noise level
+    assert s["mse"] < 0.2, s
+
+
+def test_gbt_fits_nonlinear_signal():
+    job = JobConfig(job_id="t_gbt", app="gbt", max_num_epochs=8,
+                    num_mini_batches=2,
+                    app_args={"num_features": 8, "batch_size": 1024,
+                              "num_bins": 32, "max_depth": 5,
+                              "step_size": 0.3, "noise": 0.05})
+    ctx = init_executor(RuntimeConfig(device="cpu"))
+    m = run_job(job, ctx)
+    s = m.summary()
+    assert s["num_batches"] == 16
+    assert s["num_trees"] == 15          # forest pulled before last push
+    # boosting reduced mse well below the signal variance (~9)
+    assert s["mse"] < 1.5, s
+
+
+def test_gbt_tree_predict_routing():
+    from harmony_amd.mlapps.gbt import GBTree
+
+    t = GBTree(depth=2, feature=[0, 1, 1], threshold=[2, 1, -1],
+               leaf_value=[10.0, 20.0, 30.0, 40.0])
+    bins = torch.tensor([[0, 0], [0, 5], [5, 0], [5, 9]])
+    out = t.predict_bins(bins)
+    # f0<=2 -> left subtree: f1<=1 ? leaf0 : leaf1
+    # f0>2  -> right subtree: threshold -1 -> always left -> leaf2
+    assert out.tolist() == [10.0, 20.0, 30.0, 30.0]
+
+
+def _gbt_2rank_worker(rank, world):
+    from harmony_amd.config import JobConfig, RuntimeConfig
+    from harmony_amd.dolphin.master import run_job
+    from harmony_amd.runtime.bootstrap import init_executor
+
+    ctx = init_executor(RuntimeConfig(device="cpu"))
+    job = JobConfig(job_id="t_gbt2", app="gbt", max_num_epochs=2,
+                    num_mini_batches=2,
+                    app_args={"num_features": 8, "batch_size": 256,
+                              "num_bins": 16, "max_depth": 3})
+    m = run_job(job, ctx)
+    s = m.summary()
+    # each batch, BOTH ranks push one tree -> forest grows by 2 per batch
+    return (s["num_batches"], s["num_trees"])
+
+
+def test_gbt_two_ranks_forest_shared():
+    res = run_dist(_gbt_2rank_worker, world=2, timeout=120)
+    for nb, nt in res:
+        assert nb == 4
+        assert nt == 6   # before the 4th batch's push: 3 batches * 2 ranks
