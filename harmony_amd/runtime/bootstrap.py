@@ -79,11 +79,8 @@ def init_executor(cfg: Optional[RuntimeConfig] = None) -> ExecutorContext:
             dist.init_process_group(
                 backend=backend, rank=rank, world_size=world,
                 timeout=datetime.timedelta(seconds=300))
-        from torch.distributed import TCPStore
-
-        store = TCPStore(master_addr, master_port + 7, world,
-                         is_master=(rank == 0),
-                         timeout=datetime.timedelta(seconds=300))
+        store = ThreadLocalTCPStore(master_addr, master_port + 7, world,
+                                    is_master=(rank == 0))
         # Pin the store for the process lifetime: if rank 0's TCPStore object
         # is GC'd when the caller's frame dies, the server socket closes and
         # every peer still polling it crashes.
@@ -94,6 +91,72 @@ def init_executor(cfg: Optional[RuntimeConfig] = None) -> ExecutorContext:
 
     return ExecutorContext(rank=rank, world_size=world, device=device,
                            backend=backend, store=store)
+
+
+class ThreadLocalTCPStore:
+    """One TCPStore CLIENT per thread.
+
+    A single TCPStore client multiplexes one socket; a thread blocked in
+    wait() holds the connection while another thread's compare_set on the
+    same client wedges behind it — with concurrent job tasklets this
+    deadlocks (observed: one tasklet in wait(tu/job_of/..), the other stuck
+    inside compare_set forever). Each thread therefore gets its own client
+    connection; the rank-0 master listener lives on the first instance.
+    """
+
+    def __init__(self, host: str, port: int, world: int, is_master: bool):
+        import threading
+
+        self._host = host
+        self._port = port
+        self._world = world
+        self._tls = threading.local()
+        self._master_store = None
+        if is_master:
+            self._master_store = self._new_client(is_master=True)
+
+    def _new_client(self, is_master: bool = False):
+        from torch.distributed import TCPStore
+
+        # wait_for_workers=False: clients are created lazily per thread, so
+        # nobody can count on a fixed connection census.
+        return TCPStore(self._host, self._port, self._world,
+                        is_master=is_master,
+                        timeout=datetime.timedelta(seconds=300),
+                        wait_for_workers=False)
+
+    def _client(self):
+        c = getattr(self._tls, "c", None)
+        if c is None:
+            if self._master_store is not None and not hasattr(self._tls, "used_master"):
+                # the creating thread reuses the master instance
+                self._tls.used_master = True
+                c = self._master_store
+            else:
+                c = self._new_client()
+            self._tls.c = c
+        return c
+
+    def set(self, key, value):
+        return self._client().set(key, value)
+
+    def get(self, key):
+        return self._client().get(key)
+
+    def add(self, key, amount):
+        return self._client().add(key, amount)
+
+    def compare_set(self, key, expected, desired):
+        return self._client().compare_set(key, expected, desired)
+
+    def wait(self, keys):
+        return self._client().wait(keys)
+
+    def check(self, keys):
+        return self._client().check(keys)
+
+    def delete_key(self, key):
+        return self._client().delete_key(key)
 
 
 class LocalStore:
