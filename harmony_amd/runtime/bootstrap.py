@@ -111,7 +111,9 @@ class ThreadLocalTCPStore:
         self._port = port
         self._world = world
         self._tls = threading.local()
+        self._lock = threading.Lock()
         self._master_store = None
+        self._master_claimed = False
         if is_master:
             self._master_store = self._new_client(is_master=True)
 
@@ -128,11 +130,13 @@ class ThreadLocalTCPStore:
     def _client(self):
         c = getattr(self._tls, "c", None)
         if c is None:
-            if self._master_store is not None and not hasattr(self._tls, "used_master"):
-                # the creating thread reuses the master instance
-                self._tls.used_master = True
-                c = self._master_store
-            else:
+            with self._lock:
+                if self._master_store is not None and not self._master_claimed:
+                    # exactly ONE thread may reuse the master instance — the
+                    # claim flag must be process-wide, not thread-local
+                    self._master_claimed = True
+                    c = self._master_store
+            if c is None:
                 c = self._new_client()
             self._tls.c = c
         return c
