@@ -39,9 +39,10 @@ class ETModelAccessor:
         self.metrics[METRIC_PULL_TIME] += time.perf_counter() - t0
         return out
 
-    def push(self, keys: torch.Tensor, deltas: torch.Tensor) -> None:
+    def push(self, keys: torch.Tensor, deltas: torch.Tensor,
+             assume_unique: bool = False) -> None:
         t0 = time.perf_counter()
-        self.table.update(keys, deltas)
+        self.table.update(keys, deltas, assume_unique=assume_unique)
         self.metrics[METRIC_PUSH_TIME] += time.perf_counter() - t0
 
     def push_dense(self, grad_full: torch.Tensor) -> None:
@@ -77,8 +78,9 @@ class CachedModelAccessor(ETModelAccessor):
     def pull(self, keys: torch.Tensor) -> torch.Tensor:
         return self.pull_all()[keys]
 
-    def push(self, keys: torch.Tensor, deltas: torch.Tensor) -> None:
-        super().push(keys, deltas)
+    def push(self, keys: torch.Tensor, deltas: torch.Tensor,
+             assume_unique: bool = False) -> None:
+        super().push(keys, deltas, assume_unique=assume_unique)
         if self._cache is not None:
             from harmony_amd.et import update_functions as uf
 

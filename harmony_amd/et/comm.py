@@ -34,6 +34,10 @@ class DataPlane:
         self.device = device
         self.backend = dist.get_backend(group) if dist.is_initialized() else "none"
         self._perm_cache: Dict[Tuple[str, int], torch.Tensor] = {}
+        # routing cache for STATIC key sets (a mini-batch block's keys never
+        # change, so the per-pull argsort/bincount is paid once per
+        # (block, ownership version) instead of every batch)
+        self._route_cache: Dict[tuple, tuple] = {}
 
     # ------------------------------------------------------------------ utils
 
@@ -150,13 +154,29 @@ class DataPlane:
     # -------------------------------------------------------------- pull keys
 
     def _route(self, table, keys: torch.Tensor):
-        """Sort keys by owner rank; returns (sorted_keys, order, send_splits)."""
+        """Sort keys by owner rank; returns (sorted_keys, order, send_splits).
+
+        Routing is cached only for tensors explicitly marked long-lived
+        (`keys._harmony_static = True`, set by batch constructors): caching by
+        data_ptr alone would alias freed/reallocated tensors."""
+        cacheable = getattr(keys, "_harmony_static", False)
+        ck = (table.cfg.table_id, keys.data_ptr(), keys.numel(),
+              table.ownership.version) if cacheable else None
+        if ck is not None:
+            hit = self._route_cache.get(ck)
+            if hit is not None:
+                return hit
         owners = table.ownership.owner.to(keys.device)[table.part.block_of(keys)]
         order = torch.argsort(owners.to(torch.int64), stable=True)
         sorted_keys = keys[order]
         splits = torch.bincount(owners.to(torch.int64),
                                 minlength=self.world_size)
-        return sorted_keys, order, splits
+        out = (sorted_keys, order, splits)
+        if ck is not None:
+            if len(self._route_cache) > 512:
+                self._route_cache.clear()
+            self._route_cache[ck] = out
+        return out
 
     def pull_keys(self, table, keys: torch.Tensor) -> torch.Tensor:
         keys = keys.to(self.device, torch.int64)
@@ -172,7 +192,8 @@ class DataPlane:
 
     # -------------------------------------------------------------- push keys
 
-    def push_keys(self, table, keys: torch.Tensor, deltas: torch.Tensor) -> None:
+    def push_keys(self, table, keys: torch.Tensor, deltas: torch.Tensor,
+                  assume_unique: bool = False) -> None:
         from harmony_amd.et.update_functions import merge_key_deltas
 
         keys = keys.to(self.device, torch.int64)
@@ -181,7 +202,10 @@ class DataPlane:
         # writes; merging before the wire preserves update semantics because
         # every registered update function is delta-merge associative — the
         # merge algebra (sum/min/last) comes from the update fn's MERGE_MODE).
-        uniq, agg = merge_key_deltas(keys, deltas, table.cfg.update_fn)
+        if assume_unique:
+            uniq, agg = keys, deltas     # caller already aggregated per key
+        else:
+            uniq, agg = merge_key_deltas(keys, deltas, table.cfg.update_fn)
         sorted_keys, order, send_counts = self._route(table, uniq)
         sorted_deltas = agg[order]
         recv_counts = self._exchange_counts(send_counts.to("cpu"))
@@ -194,6 +218,24 @@ class DataPlane:
         u2, agg2 = merge_key_deltas(recv_keys, recv_deltas,
                                     table.cfg.update_fn)
         table.update_local(u2, agg2)
+
+    # ------------------------------------------------------------ pair push
+
+    def push_pairs(self, table, keys: torch.Tensor,
+                   payload: torch.Tensor, apply_fn) -> None:
+        """Route (key, payload-row) pairs to owners; the owner applies them
+        with apply_fn(table, local_keys, local_payload). Used for compressed
+        delta formats (LDA TopicChanges: payload = (old_topic, new_topic))
+        where a dense per-key row would waste xGMI bandwidth."""
+        keys = keys.to(self.device, torch.int64)
+        sorted_keys, order, send_counts = self._route(table, keys)
+        sorted_payload = payload[order]
+        recv_counts = self._exchange_counts(send_counts.to("cpu"))
+        ssp, rsp = send_counts.tolist(), recv_counts.tolist()
+        recv_keys = self._all_to_all_v(sorted_keys, ssp, rsp)
+        recv_payload = self._all_to_all_v(sorted_payload, ssp, rsp)
+        if recv_keys.numel():
+            apply_fn(table, recv_keys, recv_payload)
 
     # ---------------------------------------------------------- object tables
 

@@ -160,14 +160,19 @@ class Table:
     # get_or_init is get: the shard is always fully initialized.
     get_or_init = get
 
-    def update(self, keys: torch.Tensor, deltas: torch.Tensor) -> None:
+    def update(self, keys: torch.Tensor, deltas: torch.Tensor,
+               assume_unique: bool = False) -> None:
         """multiUpdate (reference TableImpl.java:460): route deltas to owner
-        ranks; the owner applies the table's update function."""
+        ranks; the owner applies the table's update function.
+        assume_unique: keys are already unique + per-key aggregated (skips a
+        device sort/unique on the hot path)."""
         if self._local_only():
-            keys, deltas = uf.merge_key_deltas(keys, deltas, self.cfg.update_fn)
+            if not assume_unique:
+                keys, deltas = uf.merge_key_deltas(keys, deltas,
+                                                   self.cfg.update_fn)
             self.update_local(keys, deltas)
             return
-        self.comm.push_keys(self, keys, deltas)
+        self.comm.push_keys(self, keys, deltas, assume_unique=assume_unique)
 
     def pull_all(self) -> torch.Tensor:
         """Gather the whole table (dense apps pull every partition each batch,

@@ -135,10 +135,13 @@ class JobServerDriver:
             conn.close()
 
     def _reap_finished(self) -> None:
-        done = [jid for jid in list(self.pool.running)
-                if self.cp.flag_set(f"js/result/{jid}")]
+        # called from the listener loop AND connection-handler threads
+        with self._lock:
+            done = [jid for jid in list(self.pool.running)
+                    if self.cp.flag_set(f"js/result/{jid}")]
+            for jid in done:
+                self.pool.running.pop(jid, None)
         for jid in done:
-            del self.pool.running[jid]
             self.scheduler.on_job_finish(jid, self.pool)
         if done:
             self._try_schedule()

@@ -66,6 +66,7 @@ class NMFBatch:
         self.col_idx = col_idx          # [nnz] global column key
         self.vals = vals                # [nnz]
         self.uniq_cols, self.col_local = torch.unique(col_idx, return_inverse=True)
+        self.uniq_cols._harmony_static = True   # routing cached (et/comm.py)
         self.num_examples = l_rows.shape[0]
 
 
@@ -117,7 +118,9 @@ class NMFTrainer(Trainer):
         self._sq_err += float(sq)
 
     def push_update(self) -> None:
-        self.accessor.push(self.batch.uniq_cols, self.rgrad)
+        # uniq_cols are unique and rgrad is already per-key aggregated by K1
+        self.accessor.push(self.batch.uniq_cols, self.rgrad,
+                           assume_unique=True)
 
     def on_epoch_finished(self, epoch: int) -> None:
         if (epoch + 1) % self.a["decay_period"] == 0:

@@ -190,6 +190,22 @@ def lda_gibbs(doc_topic: torch.Tensor, word_topic: torch.Tensor,
     return assignments
 
 
+def lda_apply_pairs(shard: torch.Tensor, rows: torch.Tensor,
+                    old_t: torch.Tensor, new_t: torch.Tensor) -> None:
+    """Apply (row, old_topic, new_topic) ±1 count pairs to the word-topic
+    shard (the reference's TopicChanges wire format — 12 B per changed token
+    instead of a dense K-int row per touched word)."""
+    if _use_hip(shard):
+        _hip.lda_apply_pairs(shard, rows.contiguous(),
+                             old_t.contiguous(), new_t.contiguous())
+        return
+    K = shard.shape[1]
+    flat = shard.view(-1)
+    ones = torch.ones(rows.shape[0], dtype=shard.dtype, device=shard.device)
+    flat.scatter_add_(0, rows * K + old_t.long(), -ones)
+    flat.scatter_add_(0, rows * K + new_t.long(), ones)
+
+
 # ---------------------------------------------------------------------------
 # K3/K9: fused owner-side update application ("server" compute)
 # ---------------------------------------------------------------------------

@@ -98,6 +98,39 @@ __global__ void lda_gibbs_kernel(int* __restrict__ doc_topic,
 
 }  // namespace
 
+namespace {
+
+// K9 (sparse form): apply (row, old_topic, new_topic) +/-1 pairs to the
+// owner's word-topic shard — the wire format is the reference's TopicChanges
+// delta pairs (lda/TopicChanges, LDAETModelUpdateFunction.java:43-64), 12 B
+// per changed token instead of a dense K-int row per touched word.
+__global__ void lda_apply_pairs_kernel(int* __restrict__ shard,
+                                       const int64_t* __restrict__ rows,
+                                       const int* __restrict__ old_t,
+                                       const int* __restrict__ new_t,
+                                       int64_t n, int K) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  int64_t base = rows[i] * K;
+  atomicSub(&shard[base + old_t[i]], 1);
+  atomicAdd(&shard[base + new_t[i]], 1);
+}
+
+}  // namespace
+
+void lda_apply_pairs(torch::Tensor shard, torch::Tensor rows,
+                     torch::Tensor old_t, torch::Tensor new_t) {
+  CHECK_IN(shard); CHECK_IN(rows); CHECK_IN(old_t); CHECK_IN(new_t);
+  TORCH_CHECK(shard.dtype() == torch::kInt32);
+  const int64_t n = rows.size(0);
+  if (n == 0) return;
+  const int K = shard.size(1);
+  dim3 blk(256), grid((unsigned)((n + 255) / 256));
+  hipLaunchKernelGGL(lda_apply_pairs_kernel, grid, blk, 0, current_stream(),
+                     shard.data_ptr<int>(), rows.data_ptr<int64_t>(),
+                     old_t.data_ptr<int>(), new_t.data_ptr<int>(), n, K);
+}
+
 torch::Tensor lda_gibbs(torch::Tensor doc_topic, torch::Tensor word_topic,
                         torch::Tensor topic_sum, torch::Tensor doc_offsets,
                         torch::Tensor word_ids, torch::Tensor assignments,

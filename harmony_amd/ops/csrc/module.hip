@@ -14,6 +14,8 @@ torch::Tensor lda_gibbs(torch::Tensor doc_topic, torch::Tensor word_topic,
                         torch::Tensor word_ids, torch::Tensor assignments,
                         double alpha, double beta, int64_t num_vocabs,
                         int64_t seed);
+void lda_apply_pairs(torch::Tensor shard, torch::Tensor rows,
+                     torch::Tensor old_t, torch::Tensor new_t);
 void scatter_apply(torch::Tensor shard, torch::Tensor rows,
                    torch::Tensor deltas, int64_t mode, double step,
                    double maxval);
@@ -25,6 +27,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused softmax + label-subtract + CE/accuracy (K4)");
   m.def("nmf_grad", &nmf_grad, "NMF sparse-batch gradient (K1+K2)");
   m.def("lda_gibbs", &lda_gibbs, "LDA collapsed Gibbs sweep (K7)");
+  m.def("lda_apply_pairs", &lda_apply_pairs,
+        "apply TopicChanges +/-1 pairs to the word-topic shard (K9 sparse)");
   m.def("scatter_apply", &scatter_apply, "owner-side sparse update (K9)");
   m.def("dense_apply", &dense_apply, "owner-side dense update (K3)");
 }

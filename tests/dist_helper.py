@@ -16,6 +16,11 @@ def free_port() -> int:
 
 
 def _entry(fn, rank, world, port, q, args):
+    import sys
+
+    # tee stderr to a per-rank file so run_dist can show it on failure
+    err_path = f"/tmp/dist_helper_r{rank}_p{port}.err"
+    sys.stderr = open(err_path, "w", buffering=1)
     try:
         os.environ["RANK"] = str(rank)
         os.environ["LOCAL_RANK"] = str(rank)
@@ -36,6 +41,18 @@ def _entry(fn, rank, world, port, q, args):
             except Exception:  # noqa: BLE001
                 pass
             dist.destroy_process_group()
+
+
+def _stderr_tails(world: int, port: int, n: int = 40) -> str:
+    out = []
+    for r in range(world):
+        p = f"/tmp/dist_helper_r{r}_p{port}.err"
+        try:
+            lines = open(p).readlines()[-n:]
+            out.append(f"--- rank {r} stderr tail ---\n" + "".join(lines))
+        except OSError:
+            pass
+    return "\n".join(out)
 
 
 def run_dist(fn, world: int = 2, args=(), timeout: int = 120):
@@ -60,10 +77,13 @@ def run_dist(fn, world: int = 2, args=(), timeout: int = 120):
                 dead = [(i, p.exitcode) for i, p in enumerate(procs)
                         if not p.is_alive() and p.exitcode != 0]
                 if dead:
-                    raise RuntimeError(f"worker died: (rank, exitcode)={dead}")
+                    raise RuntimeError(f"worker died: (rank, exitcode)={dead}"
+                                       f"\n{_stderr_tails(world, port)}")
                 if time.monotonic() > deadline:
-                    raise TimeoutError(f"ranks {sorted(set(range(world)) - set(results))} "
-                                       f"did not report within {timeout}s")
+                    raise TimeoutError(
+                        f"ranks {sorted(set(range(world)) - set(results))} "
+                        f"did not report within {timeout}s\n"
+                        f"{_stderr_tails(world, port)}")
                 continue
             if status == "err":
                 raise RuntimeError(f"rank {rank} failed:\n{payload}")
