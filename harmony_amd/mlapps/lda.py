@@ -86,6 +86,10 @@ def make_batches(job: JobConfig, rank: int, device: torch.device):
         # Zipf-ish word draw: square a uniform to skew mass to low ids
         u = torch.rand(D * T, generator=g)
         word_ids = (u * u * a["num_vocabs"]).long().clamp_(0, a["num_vocabs"] - 1)
+        # sort tokens by word WITHIN each doc: Gibbs is valid under any
+        # within-doc token order, and consecutive same-word tokens let the
+        # sampler reuse the word-topic row from cache (big win on Zipf data)
+        word_ids = word_ids.view(D, T).sort(dim=1).values.reshape(-1)
         batch = LDABatch(doc_ids.to(device), offsets.to(device),
                          word_ids.to(device), a["num_vocabs"])
         batch.block_idx = b

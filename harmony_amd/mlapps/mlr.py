@@ -100,9 +100,8 @@ class MLRTrainer(Trainer):
     def local_compute(self) -> None:
         x, y = self.batch
         W = self._w_matrix()                       # [C, F]
-        logits = x @ W.t()                         # rocBLAS GEMM [B, C]
-        # fused softmax + label-subtract + CE loss (HIP kernel on GPU)
-        p, loss, correct = ops.softmax_grad_ce(logits, y)
+        # fused forward: GEMM + softmax + label-subtract + CE (one X read)
+        p, loss, correct = ops.mlr_forward(x, W, y)
         grad = p.t() @ x                           # [C, F] GEMM
         grad = grad / x.shape[0] + self.a["lam"] * W
         self.grad_delta = (-self.step_size) * grad

@@ -68,6 +68,19 @@ class NMFBatch:
         self.uniq_cols, self.col_local = torch.unique(col_idx, return_inverse=True)
         self.uniq_cols._harmony_static = True   # routing cached (et/comm.py)
         self.num_examples = l_rows.shape[0]
+        # static column-sorted view for the atomic-free two-pass gradient
+        # (ops.nmf_grad col_sorted): perm sorts nonzeros by LOCAL column,
+        # seg_ptr covers every pulled R row, row_sorted = L row per nonzero
+        perm = torch.argsort(self.col_local, stable=True)
+        counts = torch.bincount(self.col_local,
+                                minlength=self.uniq_cols.shape[0])
+        seg_ptr = torch.zeros(self.uniq_cols.shape[0] + 1, dtype=torch.int64,
+                              device=col_idx.device)
+        seg_ptr[1:] = counts.cumsum(0)
+        row_of = torch.repeat_interleave(
+            torch.arange(row_ptr.shape[0] - 1, device=col_idx.device),
+            row_ptr[1:] - row_ptr[:-1])
+        self.col_sorted = (perm, seg_ptr, row_of[perm])
 
 
 def make_batches(job: JobConfig, rank: int, device: torch.device):
@@ -110,7 +123,8 @@ class NMFTrainer(Trainer):
         b = self.batch
         L_batch = self.L[b.l_rows]
         lgrad, rgrad, sq = ops.nmf_grad(L_batch, self.R_batch, b.row_ptr,
-                                        b.col_local, b.vals, self.a["lam"])
+                                        b.col_local, b.vals, self.a["lam"],
+                                        col_sorted=b.col_sorted)
         # local L update (worker-side SGD apply, same rule as the server's)
         self.L[b.l_rows] = (L_batch - self.step_size * lgrad).clamp_(
             0.0, self.a["max_val"])

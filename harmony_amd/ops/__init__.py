@@ -75,6 +75,20 @@ def _use_hip(t: torch.Tensor) -> bool:
 # K4: MLR fused softmax + label-subtract + CE/accuracy
 # ---------------------------------------------------------------------------
 
+def mlr_forward(X: torch.Tensor, W: torch.Tensor, labels: torch.Tensor
+                ) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
+    """Fused MLR forward: logits = X @ W^T, row softmax (LSE-guarded),
+    label subtract, CE/accuracy — X is read once (K4; replaces the skinny-N
+    rocBLAS GEMM + separate softmax kernel). Returns (p - onehot [B,C],
+    loss_sum, n_correct). Falls back to GEMM + softmax_grad_ce when C > 16
+    or on CPU."""
+    C = W.shape[0]
+    if C <= 16 and _use_hip(X):
+        return tuple(_hip.mlr_fwd(X.contiguous(), W.contiguous(),
+                                  labels.contiguous()))
+    return softmax_grad_ce(X @ W.t(), labels)
+
+
 def softmax_grad_ce(logits: torch.Tensor, labels: torch.Tensor
                     ) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
     """Row softmax with log-sum-exp guard; returns (p - onehot(label),
@@ -101,14 +115,24 @@ def softmax_grad_ce(logits: torch.Tensor, labels: torch.Tensor
 # ---------------------------------------------------------------------------
 
 def nmf_grad(L: torch.Tensor, R: torch.Tensor, row_ptr: torch.Tensor,
-             col_idx: torch.Tensor, vals: torch.Tensor, lam: float
-             ) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
+             col_idx: torch.Tensor, vals: torch.Tensor, lam: float,
+             col_sorted=None) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
     """For each nonzero (i, j, v): e = L_i . R_j - v;
     lgrad_i += 2 e R_j, rgrad_j += 2 e L_i; plus L2 terms 2*lam*{L_i,R_j}
     per nonzero (reference NMFTrainer.updateGradient:328-367).
     L: [n_rows, k], CSR over rows: row_ptr [n_rows+1], col_idx/vals [nnz]
-    (col_idx indexes R's rows). Returns (lgrad, rgrad, sq_err_sum)."""
+    (col_idx indexes R's rows). Returns (lgrad, rgrad, sq_err_sum).
+
+    col_sorted: optional static precompute (perm, seg_ptr, row_sorted) — the
+    column-sorted view of the nonzeros (seg_ptr covers ALL R rows) — enables
+    the atomic-free two-pass kernel on GPU."""
     if _use_hip(L):
+        if col_sorted is not None:
+            perm, seg_ptr, row_sorted = col_sorted
+            return tuple(_hip.nmf_grad_twopass(
+                L.contiguous(), R.contiguous(), row_ptr.contiguous(),
+                col_idx.contiguous(), vals.contiguous(), perm.contiguous(),
+                seg_ptr.contiguous(), row_sorted.contiguous(), float(lam)))
         return tuple(_hip.nmf_grad(L.contiguous(), R.contiguous(),
                                    row_ptr.contiguous(), col_idx.contiguous(),
                                    vals.contiguous(), float(lam)))

@@ -9,6 +9,12 @@ std::vector<torch::Tensor> nmf_grad(torch::Tensor L, torch::Tensor R,
                                     torch::Tensor row_ptr,
                                     torch::Tensor col_idx, torch::Tensor vals,
                                     double lam);
+std::vector<torch::Tensor> nmf_grad_twopass(
+    torch::Tensor L, torch::Tensor R, torch::Tensor row_ptr,
+    torch::Tensor col_idx, torch::Tensor vals, torch::Tensor perm,
+    torch::Tensor seg_ptr, torch::Tensor row_sorted, double lam);
+std::vector<torch::Tensor> mlr_fwd(torch::Tensor X, torch::Tensor W,
+                                   torch::Tensor labels);
 torch::Tensor lda_gibbs(torch::Tensor doc_topic, torch::Tensor word_topic,
                         torch::Tensor topic_sum, torch::Tensor doc_offsets,
                         torch::Tensor word_ids, torch::Tensor assignments,
@@ -26,6 +32,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mlr_softmax_grad", &mlr_softmax_grad,
         "fused softmax + label-subtract + CE/accuracy (K4)");
   m.def("nmf_grad", &nmf_grad, "NMF sparse-batch gradient (K1+K2)");
+  m.def("nmf_grad_twopass", &nmf_grad_twopass,
+        "NMF gradient, segmented-reduce rgrad, no atomics (K1+K2)");
+  m.def("mlr_fwd", &mlr_fwd,
+        "fused MLR forward: X@W^T + softmax + grad + CE/acc (K4)");
   m.def("lda_gibbs", &lda_gibbs, "LDA collapsed Gibbs sweep (K7)");
   m.def("lda_apply_pairs", &lda_apply_pairs,
         "apply TopicChanges +/-1 pairs to the word-topic shard (K9 sparse)");

@@ -165,3 +165,72 @@ def test_apps_end_to_end_gpu():
                         num_mini_batches=2, app_args=args)
         m = run_job(job, ctx)
         assert m.summary()["num_batches"] == 4, app
+
+
+def test_mlr_fwd_fused_vs_ref():
+    torch.manual_seed(7)
+    B, F, C = 2048, 512, 10
+    X = torch.randn(B, F, device="cuda")
+    W = torch.randn(C, F, device="cuda") * 0.1
+    y = torch.randint(0, C, (B,), device="cuda")
+    g, loss, cor = ops.mlr_forward(X, W, y)
+    logits = (X.cpu().double() @ W.cpu().double().t()).float()
+    g_r, loss_r, cor_r = ops.softmax_grad_ce(logits, y.cpu())
+    assert torch.allclose(g.cpu(), g_r, atol=2e-4)
+    assert abs(float(loss) - float(loss_r)) / max(1.0, float(loss_r)) < 1e-3
+    assert int(cor) == int(cor_r)
+
+
+def test_mlr_fwd_fused_odd_batch():
+    torch.manual_seed(8)
+    B, F, C = 1021, 130, 7   # non-multiples of tile sizes
+    X = torch.randn(B, F, device="cuda")
+    W = torch.randn(C, F, device="cuda") * 0.1
+    y = torch.randint(0, C, (B,), device="cuda")
+    g, loss, cor = ops.mlr_forward(X, W, y)
+    g_r, loss_r, cor_r = ops.softmax_grad_ce(
+        (X.cpu().double() @ W.cpu().double().t()).float(), y.cpu())
+    assert torch.allclose(g.cpu(), g_r, atol=2e-4)
+    assert int(cor) == int(cor_r)
+
+
+def test_nmf_twopass_matches_onepass():
+    torch.manual_seed(9)
+    n, m, k = 256, 100, 100
+    L = torch.rand(n, k, device="cuda")
+    R = torch.rand(m, k, device="cuda")
+    nnz_per_row = 9
+    nnz = n * nnz_per_row
+    row_ptr = torch.arange(0, nnz + 1, nnz_per_row, device="cuda")
+    col = torch.randint(0, m, (nnz,), device="cuda")
+    vals = torch.rand(nnz, device="cuda")
+    # precompute column-sorted view (as NMFBatch does)
+    perm = torch.argsort(col, stable=True)
+    counts = torch.bincount(col, minlength=m)
+    seg_ptr = torch.zeros(m + 1, dtype=torch.int64, device="cuda")
+    seg_ptr[1:] = counts.cumsum(0)
+    row_of = torch.repeat_interleave(torch.arange(n, device="cuda"),
+                                     row_ptr[1:] - row_ptr[:-1])
+    lg2, rg2, sq2 = ops.nmf_grad(L, R, row_ptr, col, vals, 0.01,
+                                 col_sorted=(perm, seg_ptr, row_of[perm]))
+    lg1, rg1, sq1 = ops.nmf_grad(L, R, row_ptr, col, vals, 0.01)
+    assert torch.allclose(lg1.cpu(), lg2.cpu(), atol=1e-4)
+    assert torch.allclose(rg1.cpu(), rg2.cpu(), atol=1e-3)
+    assert abs(float(sq1) - float(sq2)) < 1e-3
+    # and against the CPU reference
+    lg_r, rg_r, sq_r = _cpu_ref(ops.nmf_grad, L, R, row_ptr, col, vals, 0.01)
+    assert torch.allclose(rg2.cpu(), rg_r, atol=1e-3)
+
+
+def test_lda_apply_pairs_gpu():
+    torch.manual_seed(10)
+    W, K = 100, 32
+    shard = torch.randint(1, 50, (W, K), dtype=torch.int32, device="cuda")
+    n = 500
+    rows = torch.randint(0, W, (n,), device="cuda")
+    old = torch.randint(0, K, (n,), dtype=torch.int32, device="cuda")
+    new = torch.randint(0, K, (n,), dtype=torch.int32, device="cuda")
+    ref = shard.cpu().clone()
+    ops.lda_apply_pairs(ref, rows.cpu(), old.cpu(), new.cpu())
+    ops.lda_apply_pairs(shard, rows, old, new)
+    assert torch.equal(shard.cpu(), ref)
