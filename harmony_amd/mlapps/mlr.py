@@ -102,7 +102,7 @@ class MLRTrainer(Trainer):
         W = self._w_matrix()                       # [C, F]
         # fused forward: GEMM + softmax + label-subtract + CE (one X read)
         p, loss, correct = ops.mlr_forward(x, W, y)
-        grad = p.t() @ x                           # [C, F] GEMM
+        grad = ops.mlr_grad_gemm(p, x)             # [C, F]
         grad = grad / x.shape[0] + self.a["lam"] * W
         self.grad_delta = (-self.step_size) * grad
         self._loss_sum += float(loss)

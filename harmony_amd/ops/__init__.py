@@ -89,6 +89,14 @@ def mlr_forward(X: torch.Tensor, W: torch.Tensor, labels: torch.Tensor
     return softmax_grad_ce(X @ W.t(), labels)
 
 
+def mlr_grad_gemm(P: torch.Tensor, X: torch.Tensor) -> torch.Tensor:
+    """grad = P^T @ X (K5: skinny-C B-tile kernel on GPU; Tensile's generic
+    GEMM measured ~3x the HBM floor on C ~ 10 shapes)."""
+    if P.shape[1] <= 16 and _use_hip(X):
+        return _hip.mlr_grad(P.contiguous(), X.contiguous())
+    return P.t() @ X
+
+
 def softmax_grad_ce(logits: torch.Tensor, labels: torch.Tensor
                     ) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
     """Row softmax with log-sum-exp guard; returns (p - onehot(label),

@@ -15,6 +15,7 @@ std::vector<torch::Tensor> nmf_grad_twopass(
     torch::Tensor seg_ptr, torch::Tensor row_sorted, double lam);
 std::vector<torch::Tensor> mlr_fwd(torch::Tensor X, torch::Tensor W,
                                    torch::Tensor labels);
+torch::Tensor mlr_grad(torch::Tensor P, torch::Tensor X);
 torch::Tensor lda_gibbs(torch::Tensor doc_topic, torch::Tensor word_topic,
                         torch::Tensor topic_sum, torch::Tensor doc_offsets,
                         torch::Tensor word_ids, torch::Tensor assignments,
@@ -36,6 +37,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "NMF gradient, segmented-reduce rgrad, no atomics (K1+K2)");
   m.def("mlr_fwd", &mlr_fwd,
         "fused MLR forward: X@W^T + softmax + grad + CE/acc (K4)");
+  m.def("mlr_grad", &mlr_grad, "skinny-C gradient GEMM P^T @ X (K5)");
   m.def("lda_gibbs", &lda_gibbs, "LDA collapsed Gibbs sweep (K7)");
   m.def("lda_apply_pairs", &lda_apply_pairs,
         "apply TopicChanges +/-1 pairs to the word-topic shard (K9 sparse)");
