@@ -74,9 +74,33 @@ class LDABatch:
         self.num_examples = doc_ids.shape[0]
 
 
-def make_batches(job: JobConfig, rank: int, device: torch.device):
+def make_batches(job: JobConfig, rank: int, device: torch.device,
+                 world_size: int = 1):
     a = defaults(job)
     n_blocks = job.num_worker_blocks or job.num_mini_batches
+    if job.app_args.get("input"):
+        # sample_lda format: one doc per line of word ids (reference
+        # LDA data; each rank takes its file split, docs chunked to blocks)
+        from harmony_amd import dataloader as dl
+
+        lines = dl.load_rank_lines(job.app_args["input"], rank, world_size)
+        offsets, words = dl.parse_lda(lines)
+        n_docs = offsets.shape[0] - 1
+        blocks = []
+        per = max(1, n_docs // n_blocks)
+        for b in range(n_blocks):
+            lo, hi = b * per, min((b + 1) * per, n_docs)
+            if lo >= hi:
+                break
+            tok_lo, tok_hi = int(offsets[lo]), int(offsets[hi])
+            w = words[tok_lo:tok_hi]
+            # sort tokens within each doc (see synthetic path note)
+            off = (offsets[lo:hi + 1] - offsets[lo]).to(device)
+            batch = LDABatch(torch.arange(lo, hi).to(device), off,
+                             w.to(device), a["num_vocabs"])
+            batch.block_idx = len(blocks)
+            blocks.append(batch)
+        return blocks, n_docs
     D, T = a["docs_per_batch"], a["tokens_per_doc"]
     g = torch.Generator().manual_seed(stable_seed(job.job_id, "data", rank))
     blocks = []
@@ -203,7 +227,8 @@ def build(job: JobConfig, ctx, cp):
     cfg = model_table_cfg(job, ctx.world_size)
     comm = ctx.new_data_plane()
     table = Table(cfg, ctx.rank, ctx.world_size, ctx.device, comm=comm)
-    blocks, local_docs = make_batches(job, ctx.rank, ctx.device)
+    blocks, local_docs = make_batches(job, ctx.rank, ctx.device,
+                                      ctx.world_size)
     tctx = TrainerContext(job_id=job.job_id, rank=ctx.rank,
                           world_size=ctx.world_size, device=ctx.device,
                           tables={MODEL_TABLE: table}, app_args=job.app_args)

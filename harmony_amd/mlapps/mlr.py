@@ -58,11 +58,22 @@ def model_table_cfg(job: JobConfig, world_size: int) -> TableConfig:
     )
 
 
-def make_batches(job: JobConfig, rank: int, device: torch.device):
-    """Synthetic classification data: per-class gaussian blobs (deterministic
-    per (job, rank)); one block = one mini-batch (reference
-    ETTrainingDataProvider one-block-one-batch)."""
+def make_batches(job: JobConfig, rank: int, device: torch.device,
+                 world_size: int = 1):
+    """One block = one mini-batch (reference ETTrainingDataProvider).
+    With app_args['input'] (reference -input flag): load this rank's split of
+    a sample_mlr-format file; otherwise synthetic per-class gaussian blobs,
+    deterministic per (job, rank)."""
     a = defaults(job)
+    if a.get("input") or job.app_args.get("input"):
+        from harmony_amd import dataloader as dl
+
+        lines = dl.load_rank_lines(job.app_args["input"], rank, world_size)
+        X, y = dl.parse_libsvm(lines, a["num_features"])
+        n_blocks = max(1, job.num_worker_blocks or job.num_mini_batches)
+        xs = torch.chunk(X, n_blocks)
+        ys = torch.chunk(y.long(), n_blocks)
+        return [(x.to(device), yy.to(device)) for x, yy in zip(xs, ys)]
     C, F, B = a["num_classes"], a["num_features"], a["batch_size"]
     from harmony_amd.utils import stable_seed
 
@@ -137,5 +148,6 @@ def build(job: JobConfig, ctx, cp):
                           world_size=ctx.world_size, device=ctx.device,
                           tables={MODEL_TABLE: table}, app_args=job.app_args)
     trainer = MLRTrainer(tctx)
-    provider = TrainingDataProvider(make_batches(job, ctx.rank, ctx.device))
+    provider = TrainingDataProvider(
+        make_batches(job, ctx.rank, ctx.device, ctx.world_size))
     return {MODEL_TABLE: table}, trainer, provider

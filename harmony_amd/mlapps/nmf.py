@@ -83,9 +83,37 @@ class NMFBatch:
         self.col_sorted = (perm, seg_ptr, row_of[perm])
 
 
-def make_batches(job: JobConfig, rank: int, device: torch.device):
+def make_batches(job: JobConfig, rank: int, device: torch.device,
+                 world_size: int = 1):
     a = defaults(job)
     n_blocks = job.num_worker_blocks or job.num_mini_batches
+    if job.app_args.get("input"):
+        # sample_nmf format "rowId: col,val ..." (reference NMFETDataParser):
+        # this rank's rows become local L rows, CSR per row, rows chunked
+        # into blocks
+        from harmony_amd import dataloader as dl
+
+        lines = dl.load_rank_lines(job.app_args["input"], rank, world_size)
+        rows, cols, vals = dl.parse_nmf(lines)
+        uniq_rows, local_row = torch.unique(rows, return_inverse=True)
+        n_rows = uniq_rows.shape[0]
+        order = torch.argsort(local_row, stable=True)
+        local_row, cols, vals = local_row[order], cols[order], vals[order]
+        counts = torch.bincount(local_row, minlength=n_rows)
+        row_ptr_all = torch.zeros(n_rows + 1, dtype=torch.int64)
+        row_ptr_all[1:] = counts.cumsum(0)
+        blocks = []
+        per = max(1, n_rows // n_blocks)
+        for b in range(n_blocks):
+            lo, hi = b * per, min((b + 1) * per, n_rows)
+            if lo >= hi:
+                break
+            p0, p1 = int(row_ptr_all[lo]), int(row_ptr_all[hi])
+            blocks.append(NMFBatch(
+                torch.arange(lo, hi).to(device),
+                (row_ptr_all[lo:hi + 1] - row_ptr_all[lo]).to(device),
+                cols[p0:p1].to(device), vals[p0:p1].to(device)))
+        return blocks, n_rows
     rows_local = a["rows_per_batch"] * n_blocks
     g = torch.Generator().manual_seed(stable_seed(job.job_id, "data", rank))
     blocks = []
@@ -153,7 +181,8 @@ def build(job: JobConfig, ctx, cp):
     cfg = model_table_cfg(job, ctx.world_size)
     comm = ctx.new_data_plane()
     table = Table(cfg, ctx.rank, ctx.world_size, ctx.device, comm=comm)
-    blocks, rows_local = make_batches(job, ctx.rank, ctx.device)
+    blocks, rows_local = make_batches(job, ctx.rank, ctx.device,
+                                      ctx.world_size)
     tctx = TrainerContext(job_id=job.job_id, rank=ctx.rank,
                           world_size=ctx.world_size, device=ctx.device,
                           tables={MODEL_TABLE: table}, app_args=job.app_args)
