@@ -92,9 +92,11 @@ class JobConfig:
     num_trainer_threads: int = 1
     app_args: Dict[str, Any] = field(default_factory=dict)
     model_is_local: bool = False       # also create a local-model table
-    optimizer: Optional[str] = None    # elasticity optimizer ("homogeneous"
-                                       # or "module:Class"); None = off
+    optimizer: Optional[str] = None    # elasticity optimizer ("homogeneous",
+                                       # "hetero_ilp" or "module:Class")
     optimizer_period: int = 8          # batches between optimization windows
+    dashboard_url: Optional[str] = None  # POST epoch metrics here
+    trace_path: Optional[str] = None   # JSONL span output (rocTX on GPU)
 
     def to_json(self) -> str:
         return json.dumps(dataclasses.asdict(self))

@@ -27,6 +27,10 @@ def _make_optimizer(name: str):
 
     if name == "homogeneous":
         return HomogeneousCostOptimizer()
+    if name == "hetero_ilp":
+        from harmony_amd.optimizer.hetero import HeterogeneousOptimizer
+
+        return HeterogeneousOptimizer()
     if ":" in name:
         import importlib
 
@@ -59,10 +63,24 @@ def run_job(job: JobConfig, ctx: ExecutorContext,
             cp, job.job_id, ctx.rank, ctx.world_size, tables,
             optimizer=optimizer, check_period=job.optimizer_period,
             group=getattr(ctx, "group", None))
+    tracer = None
+    if job.trace_path:
+        from harmony_amd.utils.tracing import Tracer
+
+        tracer = Tracer(rank=ctx.rank, job=job.job_id,
+                        out_path=f"{job.trace_path}.r{ctx.rank}.jsonl")
+    dashboard = None
+    if job.dashboard_url:
+        from harmony_amd.dashboard import DashboardConnector
+
+        dashboard = DashboardConnector(job.dashboard_url)
     tasklet = WorkerTasklet(job, trainer, provider, cp, tus,
                             ctx.rank, ctx.world_size, stream=stream,
-                            orchestrator=orch)
+                            orchestrator=orch, tracer=tracer,
+                            dashboard=dashboard)
     metrics = tasklet.run()
+    if tracer is not None:
+        tracer.flush()
     ev = trainer.evaluate_model()
     for k, v in (ev or {}).items():
         metrics.add_custom(k, float(v))

@@ -31,7 +31,9 @@ class WorkerTasklet:
                  provider: TrainingDataProvider, cp: ControlPlane,
                  tus: TaskUnitScheduler, rank: int, world_size: int,
                  stream: Optional[torch.cuda.Stream] = None,
-                 orchestrator=None):
+                 orchestrator=None, tracer=None, dashboard=None):
+        from harmony_amd.utils.tracing import null_tracer
+
         self.job = job
         self.trainer = trainer
         self.provider = provider
@@ -41,6 +43,8 @@ class WorkerTasklet:
         self.world_size = world_size
         self.stream = stream
         self.orch = orchestrator   # optimizer.OptimizationOrchestrator
+        self.tracer = tracer or null_tracer()
+        self.dashboard = dashboard  # dashboard.DashboardConnector or None
         self.metrics = MetricCollector(job.job_id, rank)
         self._phase = 0
         self.ssp = SSPClock(cp, job.job_id, world_size, job.clock_slack)
@@ -75,17 +79,20 @@ class WorkerTasklet:
                     self.trainer.set_batch_data(batch)
                     # PULL
                     t0 = time.perf_counter()
-                    with self.tus.net(jid, self._next_phase()):
-                        self.trainer.pull_model()
+                    with self.tracer.span("pull"):
+                        with self.tus.net(jid, self._next_phase()):
+                            self.trainer.pull_model()
                     pull_t = time.perf_counter() - t0
                     # COMP
                     t0 = time.perf_counter()
-                    self.trainer.local_compute()
+                    with self.tracer.span("comp"):
+                        self.trainer.local_compute()
                     comp_t = time.perf_counter() - t0
                     # PUSH
                     t0 = time.perf_counter()
-                    with self.tus.net(jid, self._next_phase()):
-                        self.trainer.push_update()
+                    with self.tracer.span("push"):
+                        with self.tus.net(jid, self._next_phase()):
+                            self.trainer.push_update()
                     push_t = time.perf_counter() - t0
                     n = self.trainer.num_batch_examples()
                     ep_examples += n
@@ -105,9 +112,16 @@ class WorkerTasklet:
                             with self.tus.net(jid, self._next_phase()):
                                 self.orch.apply(plan)
                 self.trainer.on_epoch_finished(epoch)
+                ep_dt = time.perf_counter() - ep_t0
                 self.metrics.add_epoch(EpochMetrics(
                     epoch_idx=epoch, num_examples=ep_examples,
-                    epoch_time_sec=time.perf_counter() - ep_t0))
+                    epoch_time_sec=ep_dt))
+                if self.dashboard is not None:
+                    self.dashboard.send(jid, self.rank, {
+                        "epoch": epoch,
+                        "data_processing_rate":
+                            ep_examples / ep_dt if ep_dt else 0.0,
+                        "epoch_time_sec": ep_dt}, t=time.time())
             # RUN -> CLEANUP barrier
             self.cp.barrier(f"{jid}/cleanup", self.world_size)
             self.trainer.cleanup()

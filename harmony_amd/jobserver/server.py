@@ -230,10 +230,38 @@ class JobServerDriver:
         for t in self._threads:
             t.join(timeout=600)
 
+    # ------------------------------------------------- failure detection
+
+    def _heartbeat_loop(self) -> None:
+        """Every rank heartbeats; rank 0 watches. Recovery is fail-fast, as
+        in the reference (JobServerDriver failed-evaluator handlers throw —
+        TODO #677 'no recovery'): a dead executor marks the server failed
+        and shuts it down; restart + checkpoint-restore is the story."""
+        period = 2.0
+        while not self.cp.flag_set("js/shutdown"):
+            self.cp.store.set(f"js/hb/{self.ctx.rank}", str(time.time()))
+            if self.ctx.is_master:
+                now = time.time()
+                for r in range(self.ctx.world_size):
+                    try:
+                        last = float(self.ctx.store.get(f"js/hb/{r}"))
+                    except Exception:  # noqa: BLE001
+                        continue
+                    if now - last > 10 * period:
+                        print(f"[jobserver] executor {r} heartbeat lost "
+                              f"({now - last:.0f}s) — failing fast",
+                              flush=True)
+                        self.cp.set_flag("js/failed")
+                        self.cp.set_flag("js/shutdown")
+                        return
+            time.sleep(period)
+
     def run(self) -> None:
         if self.ctx.is_master:
             lt = threading.Thread(target=self._listen_loop, daemon=True)
             lt.start()
+        hb = threading.Thread(target=self._heartbeat_loop, daemon=True)
+        hb.start()
         self.dispatch_loop()
         if dist.is_initialized():
             dist.barrier()
