@@ -58,6 +58,9 @@ void mlr_fwd_kernel(const float* __restrict__ X,
                     float* __restrict__ loss,
                     int* __restrict__ correct,
                     int B, int F, int C) {
+  // Each thread owns 4 consecutive f (float4 loads: the scalar-load version
+  // was VMEM-ISSUE-bound at 3x the HBM floor); threads stride F in steps of
+  // 4*FWD_THREADS; ROWS_PER_WG rows share every W load.
   __shared__ float part[ROWS_PER_WG][FWD_THREADS / WAVE][MLR_MAXC];
   const int row0 = blockIdx.x * ROWS_PER_WG;
   const int tid = threadIdx.x;
@@ -70,18 +73,33 @@ void mlr_fwd_kernel(const float* __restrict__ X,
 #pragma unroll
     for (int c = 0; c < MLR_MAXC; ++c) acc[r][c] = 0.f;
 
-  for (int f = tid; f < F; f += FWD_THREADS) {
-    float w[MLR_MAXC];
+  const int f4 = F >> 2;                  // F % 4 handled by the tail loop
+  for (int i = tid; i < f4; i += FWD_THREADS) {
+    const int f = i * 4;
+    float4 w4[MLR_MAXC];
 #pragma unroll
-    for (int c = 0; c < MLR_MAXC; ++c)        // static unroll: runtime-bound
-      w[c] = (c < C) ? W[(int64_t)c * F + f] : 0.f;  // loops spill (rule 20)
+    for (int c = 0; c < MLR_MAXC; ++c)
+      w4[c] = (c < C) ? *(const float4*)&W[(int64_t)c * F + f]
+                      : make_float4(0.f, 0.f, 0.f, 0.f);
 #pragma unroll
     for (int r = 0; r < ROWS_PER_WG; ++r) {
-      // guard, never break: break blocks unrolling -> scratch spill (rule 20)
+      const float4 x = (row0 + r < B)
+          ? *(const float4*)&X[(int64_t)(row0 + r) * F + f]
+          : make_float4(0.f, 0.f, 0.f, 0.f);
+#pragma unroll
+      for (int c = 0; c < MLR_MAXC; ++c)
+        if (c < C)
+          acc[r][c] += x.x * w4[c].x + x.y * w4[c].y + x.z * w4[c].z +
+                       x.w * w4[c].w;
+    }
+  }
+  for (int f = f4 * 4 + tid; f < F; f += FWD_THREADS) {   // tail (F % 4)
+#pragma unroll
+    for (int r = 0; r < ROWS_PER_WG; ++r) {
       const float x = (row0 + r < B) ? X[(int64_t)(row0 + r) * F + f] : 0.f;
 #pragma unroll
       for (int c = 0; c < MLR_MAXC; ++c)
-        if (c < C) acc[r][c] += x * w[c];
+        if (c < C) acc[r][c] += x * W[(int64_t)c * F + f];
     }
   }
 #pragma unroll
@@ -130,32 +148,68 @@ void mlr_fwd_kernel(const float* __restrict__ X,
 // partials per thread in registers, one atomicAdd per (c, f) per tile into
 // the L2-resident 640 KB grad buffer. Tensile's generic GEMM measured ~3x
 // the HBM floor on this shape.
+constexpr int GRAD_BTILE = 512;
+constexpr int GRAD_FTILE = 4 * FWD_THREADS;   // 4 consecutive f per thread
+
 __global__ __launch_bounds__(FWD_THREADS)
 void mlr_grad_kernel(const float* __restrict__ P,
                      const float* __restrict__ X,
                      float* __restrict__ grad,
                      int B, int F, int C) {
-  constexpr int BTILE = 256;
-  __shared__ float pl[BTILE * MLR_MAXC];
-  const int b0 = blockIdx.x * BTILE;
+  // 2D grid (f-tile, b-tile): the single-axis version launched only B/256
+  // workgroups (1 wave/CU on a 16k batch - measured 12.7 ms). P tile in
+  // LDS (broadcast), X float4-streamed once, one atomicAdd per (c, f) per
+  // b-tile into the L2-resident grad buffer.
+  __shared__ float pl[GRAD_BTILE * MLR_MAXC];
+  const int f0 = blockIdx.x * GRAD_FTILE;
+  const int b0 = blockIdx.y * GRAD_BTILE;
   const int tid = threadIdx.x;
-  const int bmax = min(BTILE, B - b0);
+  const int bmax = min(GRAD_BTILE, B - b0);
   for (int idx = tid; idx < bmax * C; idx += FWD_THREADS)
-    pl[(idx / C) * MLR_MAXC + (idx % C)] = P[(int64_t)(b0 + idx / C) * C + idx % C];
+    pl[(idx / C) * MLR_MAXC + (idx % C)] =
+        P[(int64_t)(b0 + idx / C) * C + idx % C];
   __syncthreads();
-  for (int f = tid; f < F; f += FWD_THREADS) {
-    float facc[MLR_MAXC];
+  const int f = f0 + tid * 4;
+  if (f + 3 < F) {
+    float4 facc[MLR_MAXC];
 #pragma unroll
-    for (int c = 0; c < MLR_MAXC; ++c) facc[c] = 0.f;
+    for (int c = 0; c < MLR_MAXC; ++c)
+      facc[c] = make_float4(0.f, 0.f, 0.f, 0.f);
     for (int b = 0; b < bmax; ++b) {
-      const float x = X[(int64_t)(b0 + b) * F + f];
+      const float4 x = *(const float4*)&X[(int64_t)(b0 + b) * F + f];
 #pragma unroll
       for (int c = 0; c < MLR_MAXC; ++c)
-        if (c < C) facc[c] += pl[b * MLR_MAXC + c] * x;
+        if (c < C) {
+          const float p = pl[b * MLR_MAXC + c];
+          facc[c].x += p * x.x;
+          facc[c].y += p * x.y;
+          facc[c].z += p * x.z;
+          facc[c].w += p * x.w;
+        }
     }
 #pragma unroll
     for (int c = 0; c < MLR_MAXC; ++c)
-      if (c < C) atomicAdd(&grad[(int64_t)c * F + f], facc[c]);
+      if (c < C) {
+        atomicAdd(&grad[(int64_t)c * F + f + 0], facc[c].x);
+        atomicAdd(&grad[(int64_t)c * F + f + 1], facc[c].y);
+        atomicAdd(&grad[(int64_t)c * F + f + 2], facc[c].z);
+        atomicAdd(&grad[(int64_t)c * F + f + 3], facc[c].w);
+      }
+  } else {
+    for (int ff = f; ff < min(f + 4, F); ++ff) {
+      float facc[MLR_MAXC];
+#pragma unroll
+      for (int c = 0; c < MLR_MAXC; ++c) facc[c] = 0.f;
+      for (int b = 0; b < bmax; ++b) {
+        const float x = X[(int64_t)(b0 + b) * F + ff];
+#pragma unroll
+        for (int c = 0; c < MLR_MAXC; ++c)
+          if (c < C) facc[c] += pl[b * MLR_MAXC + c] * x;
+      }
+#pragma unroll
+      for (int c = 0; c < MLR_MAXC; ++c)
+        if (c < C) atomicAdd(&grad[(int64_t)c * F + ff], facc[c]);
+    }
   }
 }
 
@@ -167,7 +221,9 @@ torch::Tensor mlr_grad(torch::Tensor P, torch::Tensor X) {
   TORCH_CHECK(P.size(0) == B && C <= MLR_MAXC);
   auto grad = torch::zeros({C, F}, X.options());
   if (B > 0) {
-    dim3 blk(FWD_THREADS), grid((B + 255) / 256);
+    dim3 blk(FWD_THREADS);
+    dim3 grid((F + GRAD_FTILE - 1) / GRAD_FTILE,
+              (B + GRAD_BTILE - 1) / GRAD_BTILE);
     hipLaunchKernelGGL(mlr_grad_kernel, grid, blk, 0, current_stream(),
                        P.data_ptr<float>(), X.data_ptr<float>(),
                        grad.data_ptr<float>(), B, F, C);

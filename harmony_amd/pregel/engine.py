@@ -150,7 +150,8 @@ class PregelEngine:
                     nxt.update(g.edge_dst, edge_msgs)
                 sent = int(g.edge_dst.numel())
             # SYNC: halt vote (allVerticesHalt AND noOngoingMsgs)
-            votes = torch.tensor([float(active.sum()), float(sent)])
+            votes = torch.tensor([float(active.sum()), float(sent)],
+                                 device=dev)
             if dist.is_initialized():
                 with self.tus.net(jid, self._next_phase()):
                     dist.all_reduce(votes, group=getattr(self.ctx, "group", None))
