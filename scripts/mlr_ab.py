@@ -9,6 +9,9 @@ import torch
 sys.path.insert(0, ".")
 from harmony_amd import ops  # noqa: E402
 
+hip = ops._load_hip()
+assert hip is not None
+
 
 def bench(fn, iters=30, warmup=5):
     for _ in range(warmup):
@@ -35,11 +38,11 @@ def main():
     variants = {
         "fwd_gemm_BN": lambda: X @ W.t(),                       # M=B,N=C
         "fwd_gemm_CT": lambda: (W @ X.t()).t().contiguous(),    # M=C,N=B
-        "fwd_fused_kernel": lambda: ops._hip.mlr_fwd(X, W, y),
-        "fwd_gemmCT_softmax": lambda: ops._hip.mlr_softmax_grad(
+        "fwd_fused_kernel": lambda: hip.mlr_fwd(X, W, y),
+        "fwd_gemmCT_softmax": lambda: hip.mlr_softmax_grad(
             (W @ X.t()).t().contiguous(), y),
         "grad_gemm": lambda: P.t() @ X,
-        "grad_kernel": lambda: ops._hip.mlr_grad(P, X),
+        "grad_kernel": lambda: hip.mlr_grad(P, X),
     }
     rounds = {k: [] for k in variants}
     for r in range(3):
