@@ -173,7 +173,7 @@ def test_mlr_fwd_fused_vs_ref():
     X = torch.randn(B, F, device="cuda")
     W = torch.randn(C, F, device="cuda") * 0.1
     y = torch.randint(0, C, (B,), device="cuda")
-    g, loss, cor = ops.mlr_forward(X, W, y)
+    g, loss, cor = ops._load_hip().mlr_fwd(X, W, y)
     logits = (X.cpu().double() @ W.cpu().double().t()).float()
     g_r, loss_r, cor_r = ops.softmax_grad_ce(logits, y.cpu())
     assert torch.allclose(g.cpu(), g_r, atol=2e-4)
@@ -187,7 +187,7 @@ def test_mlr_fwd_fused_odd_batch():
     X = torch.randn(B, F, device="cuda")
     W = torch.randn(C, F, device="cuda") * 0.1
     y = torch.randint(0, C, (B,), device="cuda")
-    g, loss, cor = ops.mlr_forward(X, W, y)
+    g, loss, cor = ops._load_hip().mlr_fwd(X, W, y)
     g_r, loss_r, cor_r = ops.softmax_grad_ce(
         (X.cpu().double() @ W.cpu().double().t()).float(), y.cpu())
     assert torch.allclose(g.cpu(), g_r, atol=2e-4)
@@ -241,6 +241,6 @@ def test_mlr_grad_gemm_vs_torch():
     B, F, C = 3000, 777, 10
     P = torch.randn(B, C, device="cuda")
     X = torch.randn(B, F, device="cuda")
-    g = ops.mlr_grad_gemm(P, X)
+    g = ops._load_hip().mlr_grad(P.contiguous(), X.contiguous())
     ref = P.cpu().double().t() @ X.cpu().double()
     assert torch.allclose(g.cpu().double(), ref, atol=1e-2, rtol=1e-4)
