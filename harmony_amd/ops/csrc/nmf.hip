@@ -103,22 +103,38 @@ __global__ void nmf_grad_e_kernel(const float* __restrict__ L,
   }
   float sq = 0.f;
   const int64_t p0 = row_ptr[row], p1 = row_ptr[row + 1];
-  for (int64_t p = p0; p < p1; ++p) {
-    const int64_t j = col_idx[p];
-    float r[MAXC];
-    float part = 0.f;
+  // software-pipelined 2-deep: issue the NEXT nonzero's R-row loads before
+  // the current dot's butterfly reduce — the serial per-nonzero chain
+  // (load R_j -> dot -> reduce) is latency-bound otherwise
+  float r[MAXC], rn[MAXC];
+  if (p0 < p1) {
+    const int64_t j0 = col_idx[p0];
 #pragma unroll
     for (int c = 0; c < MAXC; ++c) {
       int idx = c * WAVE + lane;
-      r[c] = (c < nchunk && idx < k) ? R[j * k + idx] : 0.f;
-      part += l[c] * r[c];
+      r[c] = (c < nchunk && idx < k) ? R[j0 * k + idx] : 0.f;
     }
+  }
+  for (int64_t p = p0; p < p1; ++p) {
+    if (p + 1 < p1) {
+      const int64_t jn = col_idx[p + 1];
+#pragma unroll
+      for (int c = 0; c < MAXC; ++c) {
+        int idx = c * WAVE + lane;
+        rn[c] = (c < nchunk && idx < k) ? R[jn * k + idx] : 0.f;
+      }
+    }
+    float part = 0.f;
+#pragma unroll
+    for (int c = 0; c < MAXC; ++c) part += l[c] * r[c];
     const float e = wave_reduce_sum(part) - vals[p];
     const float ge = 2.f * e;
 #pragma unroll
     for (int c = 0; c < MAXC; ++c) lg[c] += ge * r[c] + lam2 * l[c];
     if (lane == 0) e_out[p] = e;
     sq += (lane == 0) ? e * e : 0.f;
+#pragma unroll
+    for (int c = 0; c < MAXC; ++c) r[c] = rn[c];
   }
 #pragma unroll
   for (int c = 0; c < MAXC; ++c) {
@@ -144,14 +160,33 @@ __global__ void nmf_rgrad_kernel(const float* __restrict__ L,
 #pragma unroll
   for (int c = 0; c < MAXC; ++c) acc[c] = 0.f;
   const int64_t p0 = seg_ptr[col], p1 = seg_ptr[col + 1];
-  for (int64_t p = p0; p < p1; ++p) {
-    const float ge = 2.f * e_in[perm[p]];
-    const int64_t i = row_sorted[p];
+  // 2-deep pipeline over the segment's (e, L-row) loads
+  float lv[MAXC], lvn[MAXC];
+  float ge = 0.f, gen = 0.f;
+  if (p0 < p1) {
+    ge = 2.f * e_in[perm[p0]];
+    const int64_t i0 = row_sorted[p0];
 #pragma unroll
     for (int c = 0; c < MAXC; ++c) {
       int idx = c * WAVE + lane;
-      if (c < nchunk && idx < k) acc[c] += ge * L[i * k + idx];
+      lv[c] = (c < nchunk && idx < k) ? L[i0 * k + idx] : 0.f;
     }
+  }
+  for (int64_t p = p0; p < p1; ++p) {
+    if (p + 1 < p1) {
+      gen = 2.f * e_in[perm[p + 1]];
+      const int64_t in_ = row_sorted[p + 1];
+#pragma unroll
+      for (int c = 0; c < MAXC; ++c) {
+        int idx = c * WAVE + lane;
+        lvn[c] = (c < nchunk && idx < k) ? L[in_ * k + idx] : 0.f;
+      }
+    }
+#pragma unroll
+    for (int c = 0; c < MAXC; ++c) acc[c] += ge * lv[c];
+#pragma unroll
+    for (int c = 0; c < MAXC; ++c) lv[c] = lvn[c];
+    ge = gen;
   }
   // L2 term: lam2 * (#nonzeros in this column) * R_j
   const float nl = lam2 * (float)(p1 - p0);
