@@ -73,14 +73,19 @@ class DataPlane:
         return recv
 
     def _exchange_counts(self, counts: torch.Tensor) -> torch.Tensor:
-        """counts[i] = #elems this rank sends to rank i; returns recv counts."""
+        """counts[i] = #elems this rank sends to rank i; returns recv counts.
+        NCCL/RCCL requires device tensors; gloo requires host tensors — the
+        input is moved to the backend's device, the result returned on CPU
+        (split sizes must be host ints anyway)."""
+        dev = self.device if self.backend == "nccl" else torch.device("cpu")
+        counts = counts.to(dev)
         if self._supports_a2a():
             recv = torch.empty_like(counts)
             dist.all_to_all_single(recv, counts, group=self.group)
-            return recv
+            return recv.cpu()
         gathered = [torch.empty_like(counts) for _ in range(self.world_size)]
         dist.all_gather(gathered, counts, group=self.group)
-        return torch.stack(gathered)[:, self.rank].contiguous()
+        return torch.stack(gathered)[:, self.rank].contiguous().cpu()
 
     def _gather_perm(self, table) -> torch.Tensor:
         """Permutation p with full_rows[p] = concat(shards in rank order):
@@ -181,7 +186,7 @@ class DataPlane:
     def pull_keys(self, table, keys: torch.Tensor) -> torch.Tensor:
         keys = keys.to(self.device, torch.int64)
         sorted_keys, order, send_counts = self._route(table, keys)
-        recv_counts = self._exchange_counts(send_counts.to("cpu"))
+        recv_counts = self._exchange_counts(send_counts)
         ssp, rsp = send_counts.tolist(), recv_counts.tolist()
         req_keys = self._all_to_all_v(sorted_keys, ssp, rsp)      # keys we serve
         served = table.get_local(req_keys)                        # [n_req, vdim]
@@ -208,7 +213,7 @@ class DataPlane:
             uniq, agg = merge_key_deltas(keys, deltas, table.cfg.update_fn)
         sorted_keys, order, send_counts = self._route(table, uniq)
         sorted_deltas = agg[order]
-        recv_counts = self._exchange_counts(send_counts.to("cpu"))
+        recv_counts = self._exchange_counts(send_counts)
         ssp, rsp = send_counts.tolist(), recv_counts.tolist()
         recv_keys = self._all_to_all_v(sorted_keys, ssp, rsp)
         recv_deltas = self._all_to_all_v(sorted_deltas, ssp, rsp)
@@ -230,7 +235,7 @@ class DataPlane:
         keys = keys.to(self.device, torch.int64)
         sorted_keys, order, send_counts = self._route(table, keys)
         sorted_payload = payload[order]
-        recv_counts = self._exchange_counts(send_counts.to("cpu"))
+        recv_counts = self._exchange_counts(send_counts)
         ssp, rsp = send_counts.tolist(), recv_counts.tolist()
         recv_keys = self._all_to_all_v(sorted_keys, ssp, rsp)
         recv_payload = self._all_to_all_v(sorted_payload, ssp, rsp)
