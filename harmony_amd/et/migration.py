@@ -56,23 +56,28 @@ def migrate(table, moves: Dict[int, int], rank: int, world_size: int,
     table.ownership.update_many(moves)
     if world_size == 1 or not dist.is_initialized():
         return
-    # 2. batched p2p data movement over xGMI
+    # 2. batched p2p data movement over xGMI. All sends+recvs go in ONE
+    # batch_isend_irecv group: individually-enqueued NCCL p2p kernels on a
+    # shared comm stream deadlock when two ranks exchange blocks (each
+    # rank's send kernel waits for the peer's recv, which is queued behind
+    # the peer's own send).
     vdim = table.cfg.value_dim
     bs = table.block_size
-    reqs = []
+    p2p_ops = []
     recv_bufs: Dict[int, Tuple[List[int], torch.Tensor]] = {}
     for src in sorted(recvs):
         blocks = sorted(recvs[src])
         buf = torch.empty((len(blocks) * bs, vdim), dtype=table.dtype,
                           device=table.device)
         recv_bufs[src] = (blocks, buf)
-        reqs.append(dist.irecv(buf, src=src, group=group))
+        p2p_ops.append(dist.P2POp(dist.irecv, buf, src, group=group))
     for dst in sorted(sends):
         blocks = sorted(sends[dst])
         buf = torch.cat([table.local_block_view(b) for b in blocks]).contiguous()
-        reqs.append(dist.isend(buf, dst=dst, group=group))
-    for r in reqs:
-        r.wait()
+        p2p_ops.append(dist.P2POp(dist.isend, buf, dst, group=group))
+    if p2p_ops:
+        for r in dist.batch_isend_irecv(p2p_ops):
+            r.wait()
     # 3. adopt + drop
     all_sent = [b for bs_ in sends.values() for b in bs_]
     if all_sent:
