@@ -145,7 +145,13 @@ class NMFTrainer(Trainer):
         self.R_batch = None
 
     def pull_model(self) -> None:
-        self.R_batch = self.accessor.pull(self.batch.uniq_cols)
+        b = self.batch
+        # when a batch touches most of the column space (dense-ish batches),
+        # one all-gather beats the two-sided all-to-all key exchange
+        if b.uniq_cols.shape[0] * 2 > self.a["num_cols"]:
+            self.R_batch = self.accessor.pull_all()[b.uniq_cols]
+        else:
+            self.R_batch = self.accessor.pull(b.uniq_cols)
 
     def local_compute(self) -> None:
         b = self.batch
