@@ -179,9 +179,30 @@ class GBTTrainer(Trainer):
         return self.batch[0].shape[0]
 
 
-def make_batches(job: JobConfig, rank: int, device: torch.device):
+def make_batches(job: JobConfig, rank: int, device: torch.device,
+                 world_size: int = 1):
     a = defaults(job)
     F = a["num_features"]
+    n_blocks = job.num_worker_blocks or job.num_mini_batches
+    if job.app_args.get("input"):
+        # sample_gbt libsvm-style rows (+ optional .meta feature types,
+        # reference GBTMetadataParser)
+        from harmony_amd import dataloader as dl
+
+        X, y = dl.parse_libsvm_split(job.app_args["input"], rank,
+                                     world_size, F)
+        meta_path = job.app_args.get("metadata",
+                                     job.app_args["input"] + ".meta")
+        try:
+            meta = dl.parse_gbt_meta(dl.read_split(meta_path, (0, 1 << 50)))
+        except OSError:
+            meta = {}
+        _ = meta  # feature types: categorical handling is binning-equivalent
+        blocks = []
+        for xb, yb in zip(torch.chunk(X, n_blocks), torch.chunk(y, n_blocks)):
+            bins = quantize(xb, a["num_bins"])
+            blocks.append((bins.to(device), yb.to(device)))
+        return blocks
     g = torch.Generator().manual_seed(stable_seed(job.job_id, "data", rank))
     w = torch.randn(F, generator=g)
     blocks = []
@@ -206,5 +227,6 @@ def build(job: JobConfig, ctx, cp):
                           world_size=ctx.world_size, device=ctx.device,
                           tables={MODEL_TABLE: table}, app_args=job.app_args)
     trainer = GBTTrainer(tctx)
-    provider = TrainingDataProvider(make_batches(job, ctx.rank, ctx.device))
+    provider = TrainingDataProvider(
+        make_batches(job, ctx.rank, ctx.device, ctx.world_size))
     return {MODEL_TABLE: table}, trainer, provider

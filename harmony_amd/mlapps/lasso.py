@@ -52,9 +52,19 @@ def model_table_cfg(job: JobConfig, world_size: int) -> TableConfig:
     )
 
 
-def make_batches(job: JobConfig, rank: int, device: torch.device):
+def make_batches(job: JobConfig, rank: int, device: torch.device,
+                 world_size: int = 1):
     a = defaults(job)
     F = a["num_features"]
+    if job.app_args.get("input"):
+        # sample_lasso rows "label idx:val ..." (reference LassoParser)
+        from harmony_amd import dataloader as dl
+
+        X, y = dl.parse_libsvm_split(job.app_args["input"], rank,
+                                     world_size, F)
+        n_blocks = max(1, job.num_worker_blocks or job.num_mini_batches)
+        return [(xb.to(device), yb.to(device)) for xb, yb in
+                zip(torch.chunk(X, n_blocks), torch.chunk(y, n_blocks))], None
     g = torch.Generator().manual_seed(stable_seed(job.job_id, "data", rank))
     # sparse ground-truth weights
     w_true = torch.randn(F, generator=g)
@@ -114,7 +124,7 @@ def build(job: JobConfig, ctx, cp):
     cfg = model_table_cfg(job, ctx.world_size)
     comm = ctx.new_data_plane()
     table = Table(cfg, ctx.rank, ctx.world_size, ctx.device, comm=comm)
-    blocks, _ = make_batches(job, ctx.rank, ctx.device)
+    blocks, _ = make_batches(job, ctx.rank, ctx.device, ctx.world_size)
     tctx = TrainerContext(job_id=job.job_id, rank=ctx.rank,
                           world_size=ctx.world_size, device=ctx.device,
                           tables={MODEL_TABLE: table}, app_args=job.app_args)

@@ -83,8 +83,8 @@ def make_batches(job: JobConfig, rank: int, device: torch.device,
         # LDA data; each rank takes its file split, docs chunked to blocks)
         from harmony_amd import dataloader as dl
 
-        lines = dl.load_rank_lines(job.app_args["input"], rank, world_size)
-        offsets, words = dl.parse_lda(lines)
+        offsets, words = dl.parse_lda_split(job.app_args["input"], rank,
+                                            world_size)
         n_docs = offsets.shape[0] - 1
         blocks = []
         per = max(1, n_docs // n_blocks)

@@ -68,8 +68,8 @@ def make_batches(job: JobConfig, rank: int, device: torch.device,
     if a.get("input") or job.app_args.get("input"):
         from harmony_amd import dataloader as dl
 
-        lines = dl.load_rank_lines(job.app_args["input"], rank, world_size)
-        X, y = dl.parse_libsvm(lines, a["num_features"])
+        X, y = dl.parse_libsvm_split(job.app_args["input"], rank,
+                                     world_size, a["num_features"])
         n_blocks = max(1, job.num_worker_blocks or job.num_mini_batches)
         xs = torch.chunk(X, n_blocks)
         ys = torch.chunk(y.long(), n_blocks)

@@ -93,8 +93,8 @@ def make_batches(job: JobConfig, rank: int, device: torch.device,
         # into blocks
         from harmony_amd import dataloader as dl
 
-        lines = dl.load_rank_lines(job.app_args["input"], rank, world_size)
-        rows, cols, vals = dl.parse_nmf(lines)
+        rows, cols, vals = dl.parse_nmf_split(job.app_args["input"], rank,
+                                              world_size)
         uniq_rows, local_row = torch.unique(rows, return_inverse=True)
         n_rows = uniq_rows.shape[0]
         order = torch.argsort(local_row, stable=True)

@@ -28,6 +28,10 @@ void scatter_apply(torch::Tensor shard, torch::Tensor rows,
                    double maxval);
 void dense_apply(torch::Tensor shard, torch::Tensor delta, int64_t mode,
                  double step, double maxval);
+std::vector<torch::Tensor> parse_nmf_bytes(const std::string& buf);
+std::vector<torch::Tensor> parse_libsvm_bytes(const std::string& buf,
+                                              int64_t num_features);
+std::vector<torch::Tensor> parse_lda_bytes(const std::string& buf);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mlr_softmax_grad", &mlr_softmax_grad,
@@ -43,4 +47,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "apply TopicChanges +/-1 pairs to the word-topic shard (K9 sparse)");
   m.def("scatter_apply", &scatter_apply, "owner-side sparse update (K9)");
   m.def("dense_apply", &dense_apply, "owner-side dense update (K3)");
+  m.def("parse_nmf_bytes", &parse_nmf_bytes, "native NMF text parser");
+  m.def("parse_libsvm_bytes", &parse_libsvm_bytes,
+        "native libsvm-style parser (mlr/gbt/lasso)");
+  m.def("parse_lda_bytes", &parse_lda_bytes, "native LDA doc parser");
 }

@@ -18,7 +18,8 @@ from setuptools import setup  # noqa: E402
 from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E402
 
 ROOT = Path(__file__).resolve().parent
-SRC = sorted(str(p) for p in (ROOT / "harmony_amd/ops/csrc").glob("*.hip"))
+SRC = sorted(str(p) for p in (ROOT / "harmony_amd/ops/csrc").glob("*.hip")) + \
+      sorted(str(p) for p in (ROOT / "harmony_amd/ops/csrc").glob("*.cpp"))
 
 if __name__ == "__main__":
     if len(sys.argv) == 1:
