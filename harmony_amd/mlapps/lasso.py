@@ -85,7 +85,7 @@ class LassoTrainer(Trainer):
         self.a = defaults(JobConfig(job_id=ctx.job_id, app="lasso",
                                     app_args=ctx.app_args))
         self.accessor = ETModelAccessor(ctx.table(MODEL_TABLE))
-        self._loss = 0.0
+        self._loss = torch.zeros(())
 
     def pull_model(self) -> None:
         self.w = self.accessor.pull_all()[:self.a["num_parts"]].reshape(-1)
@@ -107,14 +107,14 @@ class LassoTrainer(Trainer):
             r = r + xi * (w[i] - wn)
             w[i] = wn
         self.delta = w - self.w
-        self._loss = float((r * r).mean())
+        self._loss = (r * r).mean()          # device scalar; no batch sync
 
     def push_update(self) -> None:
         P = self.a["num_parts"]
         self.accessor.push_dense(self.delta.reshape(P, -1))
 
     def evaluate_model(self):
-        return {"mse": self._loss}
+        return {"mse": float(self._loss)}
 
     def num_batch_examples(self) -> int:
         return self.batch[0].shape[0]

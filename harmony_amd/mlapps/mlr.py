@@ -96,9 +96,10 @@ class MLRTrainer(Trainer):
         self.accessor = ETModelAccessor(ctx.table(MODEL_TABLE))
         self.step_size = self.a["step_size"]
         self.W = None          # [C*P, F/P] pulled model
-        self._loss_sum = 0.0
+        # device-resident accumulators (no per-batch host syncs)
+        self._loss_sum = torch.zeros((), device=ctx.device)
         self._loss_n = 0
-        self._correct = 0
+        self._correct = torch.zeros((), dtype=torch.int64, device=ctx.device)
 
     def _w_matrix(self) -> torch.Tensor:
         C, F = self.a["num_classes"], self.a["num_features"]
@@ -116,9 +117,9 @@ class MLRTrainer(Trainer):
         grad = ops.mlr_grad_gemm(p, x)             # [C, F]
         grad = grad / x.shape[0] + self.a["lam"] * W
         self.grad_delta = (-self.step_size) * grad
-        self._loss_sum += float(loss)
+        self._loss_sum += loss
         self._loss_n += x.shape[0]
-        self._correct += int(correct)
+        self._correct += correct
 
     def push_update(self) -> None:
         C, P = self.a["num_classes"], self.a["num_parts_per_class"]
@@ -131,9 +132,12 @@ class MLRTrainer(Trainer):
     def evaluate_model(self):
         if not self._loss_n:
             return {}
-        out = {"cross_entropy": self._loss_sum / self._loss_n,
-               "accuracy": self._correct / self._loss_n}
-        self._loss_sum, self._loss_n, self._correct = 0.0, 0, 0
+        out = {"cross_entropy": float(self._loss_sum) / self._loss_n,
+               "accuracy": float(self._correct) / self._loss_n}
+        self._loss_sum = torch.zeros((), device=self.ctx.device)
+        self._correct = torch.zeros((), dtype=torch.int64,
+                                    device=self.ctx.device)
+        self._loss_n = 0
         return out
 
     def num_batch_examples(self) -> int:

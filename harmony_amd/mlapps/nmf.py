@@ -141,7 +141,9 @@ class NMFTrainer(Trainer):
         self.L = (torch.rand(num_local_rows, self.a["rank"], generator=g)
                   .to(ctx.device))
         self.step_size = self.a["step_size"]
-        self._sq_err = 0.0
+        # device-resident loss accumulator: a per-batch float() would force
+        # a host sync and serialize the async step pipeline
+        self._sq_err = torch.zeros((), device=ctx.device)
         self.R_batch = None
 
     def pull_model(self) -> None:
@@ -163,7 +165,7 @@ class NMFTrainer(Trainer):
         self.L[b.l_rows] = (L_batch - self.step_size * lgrad).clamp_(
             0.0, self.a["max_val"])
         self.rgrad = rgrad
-        self._sq_err += float(sq)
+        self._sq_err += sq
 
     def push_update(self) -> None:
         # uniq_cols are unique and rgrad is already per-key aggregated by K1
@@ -175,8 +177,8 @@ class NMFTrainer(Trainer):
             self.step_size *= self.a["decay_rate"]
 
     def evaluate_model(self):
-        out = {"sq_err": self._sq_err}
-        self._sq_err = 0.0
+        out = {"sq_err": float(self._sq_err)}
+        self._sq_err = torch.zeros((), device=self.ctx.device)
         return out
 
     def num_batch_examples(self) -> int:
