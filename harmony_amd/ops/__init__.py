@@ -235,6 +235,28 @@ def lda_apply_pairs(shard: torch.Tensor, rows: torch.Tensor,
     flat.scatter_add_(0, rows * K + new_t.long(), ones)
 
 
+def lda_apply_all(shard, word_rows, old_t, new_t, summary_row: int) -> None:
+    """Fused single-owner update: per token, if the topic changed, apply
+    ±1 to the word row and the summary row (K9; replaces the whole
+    nonzero/gather/bincount/scatter pipeline on the local path)."""
+    if _use_hip(shard):
+        _hip.lda_apply_all(shard, word_rows.contiguous(), old_t.contiguous(),
+                           new_t.contiguous(), int(summary_row))
+        return
+    changed = (old_t != new_t)
+    rows = word_rows[changed]
+    o = old_t[changed].long()
+    nw = new_t[changed].long()
+    K = shard.shape[1]
+    flat = shard.view(-1)
+    ones = torch.ones(rows.shape[0], dtype=shard.dtype, device=shard.device)
+    flat.scatter_add_(0, rows * K + o, -ones)
+    flat.scatter_add_(0, rows * K + nw, ones)
+    srow = summary_row * K
+    flat.scatter_add_(0, srow + o, -ones)
+    flat.scatter_add_(0, srow + nw, ones)
+
+
 # ---------------------------------------------------------------------------
 # K3/K9: fused owner-side update application ("server" compute)
 # ---------------------------------------------------------------------------

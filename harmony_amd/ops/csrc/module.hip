@@ -23,6 +23,9 @@ torch::Tensor lda_gibbs(torch::Tensor doc_topic, torch::Tensor word_topic,
                         int64_t seed);
 void lda_apply_pairs(torch::Tensor shard, torch::Tensor rows,
                      torch::Tensor old_t, torch::Tensor new_t);
+void lda_apply_all(torch::Tensor shard, torch::Tensor word_rows,
+                   torch::Tensor old_t, torch::Tensor new_t,
+                   int64_t summary_row);
 void scatter_apply(torch::Tensor shard, torch::Tensor rows,
                    torch::Tensor deltas, int64_t mode, double step,
                    double maxval);
@@ -45,6 +48,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("lda_gibbs", &lda_gibbs, "LDA collapsed Gibbs sweep (K7)");
   m.def("lda_apply_pairs", &lda_apply_pairs,
         "apply TopicChanges +/-1 pairs to the word-topic shard (K9 sparse)");
+  m.def("lda_apply_all", &lda_apply_all,
+        "fused local token-delta + summary update (K9, single-owner path)");
   m.def("scatter_apply", &scatter_apply, "owner-side sparse update (K9)");
   m.def("dense_apply", &dense_apply, "owner-side dense update (K3)");
   m.def("parse_nmf_bytes", &parse_nmf_bytes, "native NMF text parser");
