@@ -70,7 +70,9 @@ class CachedModelAccessor(ETModelAccessor):
 
     def pull_all(self) -> torch.Tensor:
         if self._cache is None or self._age >= self.refresh_batches:
-            self._cache = super().pull_all()
+            # clone: the local-mode pull returns the live shard, and a cache
+            # aliasing the table would see (and double-apply) every update
+            self._cache = super().pull_all().clone()
             self._age = 0
         self._age += 1
         return self._cache

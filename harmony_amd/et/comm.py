@@ -224,6 +224,34 @@ class DataPlane:
                                     table.cfg.update_fn)
         table.update_local(u2, agg2)
 
+    # ------------------------------------------------------------------ put
+
+    def put_keys(self, table, keys: torch.Tensor, values: torch.Tensor) -> None:
+        """multiPut (reference TableImpl.multiPut:156): route (key, value)
+        pairs to owners; owners overwrite (no update function). Duplicate
+        keys within one call: last writer wins (any is valid)."""
+        keys = keys.to(self.device, torch.int64)
+        values = values.to(self.device)
+        sorted_keys, order, send_counts = self._route(table, keys)
+        sorted_vals = values[order]
+        recv_counts = self._exchange_counts(send_counts)
+        ssp, rsp = send_counts.tolist(), recv_counts.tolist()
+        recv_keys = self._all_to_all_v(sorted_keys, ssp, rsp)
+        recv_vals = self._all_to_all_v(sorted_vals, ssp, rsp)
+        if recv_keys.numel():
+            table.put_local(recv_keys, recv_vals)
+
+    def remove_keys(self, table, keys: torch.Tensor) -> None:
+        """multiRemove (reference TableImpl.remove:513): owners reset the
+        rows to their deterministic init values (see Table.remove)."""
+        keys = keys.to(self.device, torch.int64)
+        sorted_keys, order, send_counts = self._route(table, keys)
+        recv_counts = self._exchange_counts(send_counts)
+        ssp, rsp = send_counts.tolist(), recv_counts.tolist()
+        recv_keys = self._all_to_all_v(sorted_keys, ssp, rsp)
+        if recv_keys.numel():
+            table.remove_local(recv_keys)
+
     # ------------------------------------------------------------ pair push
 
     def push_pairs(self, table, keys: torch.Tensor,

@@ -160,6 +160,36 @@ class Table:
     # get_or_init is get: the shard is always fully initialized.
     get_or_init = get
 
+    def put(self, keys: torch.Tensor, values: torch.Tensor) -> None:
+        """multiPut (reference TableImpl.java:156): overwrite values at the
+        owners (no update function)."""
+        if self._local_only():
+            self.put_local(keys, values)
+            return
+        self.comm.put_keys(self, keys, values)
+
+    def remove(self, keys: torch.Tensor) -> None:
+        """multiRemove (reference TableImpl.java:513). Dense tables have a
+        value for every key by construction, so remove restores the
+        deterministic initial value (the state a fresh getOrInit would see —
+        observationally the reference's remove-then-getOrInit)."""
+        if self._local_only():
+            self.remove_local(keys)
+            return
+        self.comm.remove_keys(self, keys)
+
+    def remove_local(self, keys: torch.Tensor) -> None:
+        keys = keys.to(self.device, torch.int64)
+        rows = self.local_rows_of(keys)
+        blocks = self.part.block_of(keys)
+        init = uf.init_fn(self.cfg.init_fn)
+        # regenerate per-block init rows and scatter the removed offsets
+        for b in torch.unique(blocks).tolist():
+            fresh = self.init_block_tensor(int(b))
+            sel = blocks == b
+            offs = self.part.offset_in_block(keys[sel])
+            self.shard[rows[sel]] = fresh[offs]
+
     def update(self, keys: torch.Tensor, deltas: torch.Tensor,
                assume_unique: bool = False) -> None:
         """multiUpdate (reference TableImpl.java:460): route deltas to owner
