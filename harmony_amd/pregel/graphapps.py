@@ -86,18 +86,27 @@ class ShortestPathComputation(Computation):
 def make_ring_plus_random_graph(num_vertices: int, out_degree: int,
                                 vertex_lo: int, vertex_hi: int,
                                 device, seed: int) -> LocalGraph:
-    """Synthetic graph: ring edge (connectivity) + random extra edges."""
+    """Synthetic graph: ring edge (connectivity) + random extra edges.
+
+    Edges are a pure function of (seed, vertex id) via the counter RNG, so
+    the GLOBAL graph is identical for every world size / partitioning — a
+    2-rank run computes exactly the same PageRank as a 1-rank run."""
+    from harmony_amd.ops.rng import rng_u32
+
     n_local = vertex_hi - vertex_lo
-    g = torch.Generator().manual_seed(seed)
     deg = out_degree
-    dsts = []
-    for v in range(vertex_lo, vertex_hi):
-        nbrs = [(v + 1) % num_vertices]
-        extra = torch.randint(0, num_vertices, (deg - 1,), generator=g).tolist()
-        nbrs += extra
-        dsts.append(nbrs)
+    vs = torch.arange(vertex_lo, vertex_hi, dtype=torch.int64)
+    ring = (vs + 1) % num_vertices                       # [n_local]
+    if deg > 1:
+        ctr = (vs.unsqueeze(1) * (deg - 1)
+               + torch.arange(deg - 1, dtype=torch.int64))  # [n_local, deg-1]
+        extra = rng_u32(seed & 0xFFFFFFFF, ctr.reshape(-1)) % num_vertices
+        edge_dst = torch.cat([ring.unsqueeze(1),
+                              extra.view(n_local, deg - 1)], dim=1).reshape(-1)
+    else:
+        edge_dst = ring
     row_ptr = torch.arange(0, (n_local + 1) * deg, deg, device=device)
-    edge_dst = torch.tensor([d for row in dsts for d in row], device=device)
     out_deg = torch.full((n_local,), float(deg), device=device)
-    return LocalGraph(vertex_lo=vertex_lo, row_ptr=row_ptr, edge_dst=edge_dst,
-                      out_degree=out_deg, num_vertices_global=num_vertices)
+    return LocalGraph(vertex_lo=vertex_lo, row_ptr=row_ptr,
+                      edge_dst=edge_dst.to(device), out_degree=out_deg,
+                      num_vertices_global=num_vertices)

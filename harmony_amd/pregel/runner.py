@@ -47,7 +47,7 @@ def run_pregel_job(job: JobConfig, ctx, cp: Optional[ControlPlane] = None,
     lo, hi = engine.local_vertex_range()
     engine.set_graph(make_ring_plus_random_graph(
         n, int(a["out_degree"]), lo, hi, ctx.device,
-        stable_seed(job.job_id, "graph", ctx.rank)))
+        stable_seed(job.job_id, "graph")))   # rank-independent graph
     values = engine.run()
     dt = time.perf_counter() - t0
     return {

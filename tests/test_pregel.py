@@ -104,3 +104,38 @@ def test_shortestpath_matches_bfs():
                 q.append(v)
     ref = torch.tensor([float(dist.get(v, float("inf"))) for v in range(n)])
     assert torch.equal(vals, ref)
+
+
+def _pr_values_worker(rank, world):
+    """PageRank values must be identical for any world size (the graph is a
+    pure function of (seed, vertex id))."""
+    from harmony_amd.config import JobConfig, RuntimeConfig
+    from harmony_amd.pregel.engine import PregelEngine
+    from harmony_amd.pregel.graphapps import (PageRankComputation,
+                                              make_ring_plus_random_graph)
+    from harmony_amd.runtime.bootstrap import init_executor
+    from harmony_amd.runtime.control import ControlPlane
+
+    ctx = init_executor(RuntimeConfig(device="cpu"))
+    n = 64
+    job = JobConfig(job_id="prw", app="pagerank", app_args={})
+    cp = ControlPlane(ctx.store, ctx.rank, ctx.world_size)
+    comp = PageRankComputation(num_iters=12)
+    engine = PregelEngine(job, comp, n, ctx, cp)
+    lo, hi = engine.local_vertex_range()
+    engine.set_graph(make_ring_plus_random_graph(n, 3, lo, hi, ctx.device,
+                                                 12345))
+    vals = engine.run().squeeze(1)
+    return (lo, vals.tolist())
+
+
+def test_pagerank_world_size_invariant():
+    res2 = run_dist(_pr_values_worker, world=2, timeout=120)
+    # single-process reference
+    lo1, v1 = _pr_values_worker(0, 1)
+    combined = []
+    for lo, vals in sorted(res2):
+        combined.extend(vals)
+    assert len(combined) == len(v1)
+    for a, b in zip(combined, v1):
+        assert abs(a - b) < 1e-6
