@@ -15,9 +15,31 @@
 
 namespace {
 
-constexpr int MAXC = 16;           // K <= 1024
-constexpr int WAVES_PER_BLOCK = 4;
+constexpr int MAXC = 16;           // K <= 64*G_chunks
+constexpr int BLOCK_THREADS = 256;
 
+// Templated on the lane-group width G: one document per G-lane group.
+// G=64 (one doc per wave) for large K; G=32 (two docs per wave) halves the
+// butterfly/scan depth and doubles docs in flight — wins when K/G chunks
+// still fit the register budget (K <= 512).
+template <int G>
+__device__ __forceinline__ float group_reduce_sum(float v) {
+#pragma unroll
+  for (int m = G / 2; m > 0; m >>= 1) v += __shfl_xor(v, m, G);
+  return v;
+}
+
+template <int G>
+__device__ __forceinline__ float group_inclusive_scan(float v, int lane) {
+#pragma unroll
+  for (int d = 1; d < G; d <<= 1) {
+    float o = __shfl_up(v, d, G);
+    if (lane >= d) v += o;
+  }
+  return v;
+}
+
+template <int G>
 __global__ void lda_gibbs_kernel(int* __restrict__ doc_topic,
                                  const int* __restrict__ word_topic,
                                  const int* __restrict__ topic_sum,
@@ -26,31 +48,33 @@ __global__ void lda_gibbs_kernel(int* __restrict__ doc_topic,
                                  int* __restrict__ z,
                                  float alpha, float beta, float vbeta,
                                  int n_docs, int K, unsigned int seed) {
-  extern __shared__ int smem[];                    // [WAVES][K] nd + [K] invden
-  const int wave = threadIdx.x / WAVE;
-  const int lane = threadIdx.x & (WAVE - 1);
-  int* nd = smem + wave * K;
-  float* invden = (float*)(smem + WAVES_PER_BLOCK * K);
-  const int nchunk = (K + WAVE - 1) / WAVE;
+  constexpr int GROUPS = BLOCK_THREADS / G;
+  extern __shared__ int smem[];                  // [GROUPS][K] nd + [K] invden
+  const int group = threadIdx.x / G;
+  const int lane = threadIdx.x % G;
+  const int sub = (threadIdx.x % WAVE) / G;      // group index within wave
+  int* nd = smem + group * K;
+  float* invden = (float*)(smem + GROUPS * K);
+  const int nchunk = (K + G - 1) / G;
 
   for (int idx = threadIdx.x; idx < K; idx += blockDim.x)
     invden[idx] = 1.0f / ((float)topic_sum[idx] + vbeta);
   __syncthreads();
 
-  const int doc = blockIdx.x * WAVES_PER_BLOCK + wave;
+  const int doc = blockIdx.x * GROUPS + group;
   if (doc >= n_docs) return;
 
 #pragma unroll
   for (int c = 0; c < MAXC; ++c) {
-    int idx = c * WAVE + lane;
+    int idx = c * G + lane;
     if (c < nchunk && idx < K) nd[idx] = doc_topic[(int64_t)doc * K + idx];
   }
 
   const int64_t p0 = doc_offsets[doc], p1 = doc_offsets[doc + 1];
   // NOTE: a 2-deep prefetch of the next token's word-topic row was tried
   // and measured SLOWER (0.51 -> 0.71 ms at 1M tokens, K=256): the extra
-  // 16 registers cost occupancy, which was already hiding the row-load
-  // latency across the 8 resident waves/SIMD. Keep the simple form.
+  // registers cost occupancy, which was already hiding the row-load
+  // latency across the resident waves. Keep the simple form.
   for (int64_t p = p0; p < p1; ++p) {
     const int64_t w = word_ids[p];
     const int old = z[p];
@@ -60,7 +84,7 @@ __global__ void lda_gibbs_kernel(int* __restrict__ doc_topic,
     float part = 0.f;
 #pragma unroll
     for (int c = 0; c < MAXC; ++c) {
-      int idx = c * WAVE + lane;
+      int idx = c * G + lane;
       if (c < nchunk && idx < K) {
         pr[c] = ((float)nd[idx] + alpha) *
                 ((float)word_topic[w * K + idx] + beta) * invden[idx];
@@ -69,7 +93,7 @@ __global__ void lda_gibbs_kernel(int* __restrict__ doc_topic,
       }
       part += pr[c];
     }
-    const float tot = wave_reduce_sum(part);
+    const float tot = group_reduce_sum<G>(part);
     const float u = rng_uniform(seed, (unsigned int)p) * tot;
 
     float run = 0.f;
@@ -77,12 +101,14 @@ __global__ void lda_gibbs_kernel(int* __restrict__ doc_topic,
 #pragma unroll
     for (int c = 0; c < MAXC; ++c) {
       if (c >= nchunk) break;
-      const float csum = wave_reduce_sum(pr[c]);
+      const float csum = group_reduce_sum<G>(pr[c]);
       if (knew < 0 && run + csum > u) {
-        const float pref = wave_inclusive_scan(pr[c]);
+        const float pref = group_inclusive_scan<G>(pr[c], lane);
         unsigned long long b = __ballot(run + pref > u);
-        int sel = (b != 0) ? (__ffsll((long long)b) - 1) : (WAVE - 1);
-        knew = c * WAVE + sel;
+        unsigned long long gm = (G == WAVE)
+            ? b : ((b >> (sub * G)) & ((1ull << G) - 1));
+        int sel = (gm != 0) ? (__ffsll((long long)gm) - 1) : (G - 1);
+        knew = c * G + sel;
       }
       run += csum;
     }
@@ -95,7 +121,7 @@ __global__ void lda_gibbs_kernel(int* __restrict__ doc_topic,
 
 #pragma unroll
   for (int c = 0; c < MAXC; ++c) {
-    int idx = c * WAVE + lane;
+    int idx = c * G + lane;
     if (c < nchunk && idx < K) doc_topic[(int64_t)doc * K + idx] = nd[idx];
   }
 }
@@ -198,17 +224,33 @@ torch::Tensor lda_gibbs(torch::Tensor doc_topic, torch::Tensor word_topic,
   const int D = doc_topic.size(0), K = doc_topic.size(1);
   TORCH_CHECK(K <= WAVE * MAXC, "num_topics > ", WAVE * MAXC, " unsupported");
   if (D == 0) return assignments;
-  dim3 blk(WAVE * WAVES_PER_BLOCK);
-  dim3 grid((D + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK);
-  const size_t shmem = (size_t)(WAVES_PER_BLOCK * K + K) * 4;
-  hipLaunchKernelGGL(lda_gibbs_kernel, grid, blk, shmem, current_stream(),
-                     doc_topic.data_ptr<int>(), word_topic.data_ptr<int>(),
-                     topic_sum.data_ptr<int>(),
-                     doc_offsets.data_ptr<int64_t>(),
-                     word_ids.data_ptr<int64_t>(),
-                     assignments.data_ptr<int>(),
-                     (float)alpha, (float)beta,
-                     (float)(num_vocabs * beta), D, K,
-                     (unsigned int)(seed & 0xffffffff));
+  const int G = (K <= 32 * MAXC) ? 32 : 64;       // 2 docs/wave when K fits
+  const int groups = BLOCK_THREADS / G;
+  dim3 blk(BLOCK_THREADS);
+  dim3 grid((D + groups - 1) / groups);
+  const size_t shmem = (size_t)(groups * K + K) * 4;
+  if (G == 32) {
+    hipLaunchKernelGGL(lda_gibbs_kernel<32>, grid, blk, shmem,
+                       current_stream(),
+                       doc_topic.data_ptr<int>(), word_topic.data_ptr<int>(),
+                       topic_sum.data_ptr<int>(),
+                       doc_offsets.data_ptr<int64_t>(),
+                       word_ids.data_ptr<int64_t>(),
+                       assignments.data_ptr<int>(),
+                       (float)alpha, (float)beta,
+                       (float)(num_vocabs * beta), D, K,
+                       (unsigned int)(seed & 0xffffffff));
+  } else {
+    hipLaunchKernelGGL(lda_gibbs_kernel<64>, grid, blk, shmem,
+                       current_stream(),
+                       doc_topic.data_ptr<int>(), word_topic.data_ptr<int>(),
+                       topic_sum.data_ptr<int>(),
+                       doc_offsets.data_ptr<int64_t>(),
+                       word_ids.data_ptr<int64_t>(),
+                       assignments.data_ptr<int>(),
+                       (float)alpha, (float)beta,
+                       (float)(num_vocabs * beta), D, K,
+                       (unsigned int)(seed & 0xffffffff));
+  }
   return assignments;
 }
