@@ -12,6 +12,7 @@
 // bit-reproduced by the torch reference for cross-checking.
 
 #include "hip_common.h"
+#include <cstdlib>
 
 namespace {
 
@@ -224,7 +225,10 @@ torch::Tensor lda_gibbs(torch::Tensor doc_topic, torch::Tensor word_topic,
   const int D = doc_topic.size(0), K = doc_topic.size(1);
   TORCH_CHECK(K <= WAVE * MAXC, "num_topics > ", WAVE * MAXC, " unsupported");
   if (D == 0) return assignments;
-  const int G = (K <= 32 * MAXC) ? 32 : 64;       // 2 docs/wave when K fits
+  int G = (K <= 32 * MAXC) ? 32 : 64;             // 2 docs/wave when K fits
+  const char* env = getenv("HARMONY_LDA_G");      // A/B override
+  if (env && atoi(env) == 64) G = 64;
+  if (env && atoi(env) == 32 && K <= 32 * MAXC) G = 32;
   const int groups = BLOCK_THREADS / G;
   dim3 blk(BLOCK_THREADS);
   dim3 grid((D + groups - 1) / groups);
