@@ -20,9 +20,11 @@ constexpr int MAXC = 16;           // K <= 64*G_chunks
 constexpr int BLOCK_THREADS = 256;
 
 // Templated on the lane-group width G: one document per G-lane group.
-// G=64 (one doc per wave) for large K; G=32 (two docs per wave) halves the
-// butterfly/scan depth and doubles docs in flight — wins when K/G chunks
-// still fit the register budget (K <= 512).
+// Measured (1M tokens, K=256, MI355X): G=64 is faster IN ISOLATION
+// (0.74 vs 0.93 ms step) but G=32 wins the 3-concurrent-job bench
+// (1.52 vs 1.72 ms/step) — the halved grid and lane width leave more CU
+// issue slots to the co-scheduled jobs' kernels. The default follows the
+// multi-tenant headline (G=32 when K <= 512); HARMONY_LDA_G overrides.
 template <int G>
 __device__ __forceinline__ float group_reduce_sum(float v) {
 #pragma unroll
