@@ -96,6 +96,17 @@ class MLRTrainer(Trainer):
         self.accessor = ETModelAccessor(ctx.table(MODEL_TABLE))
         self.step_size = self.a["step_size"]
         self.W = None          # [C*P, F/P] pulled model
+        # hipGraph capture of the compute phase (static per block); the
+        # pulled model lands in a stable buffer so graphs see fixed pointers
+        import os
+
+        from harmony_amd.utils.graphs import GraphRunner
+
+        self._graphs = GraphRunner(
+            enabled=ctx.device.type == "cuda"
+            and os.environ.get("HARMONY_NO_GRAPHS") != "1")
+        self._W_buf = None
+        self._bodies = {}
         # device-resident accumulators (no per-batch host syncs)
         self._loss_sum = torch.zeros((), device=ctx.device)
         self._loss_n = 0
@@ -107,23 +118,37 @@ class MLRTrainer(Trainer):
 
     def pull_model(self) -> None:
         C, P = self.a["num_classes"], self.a["num_parts_per_class"]
-        self.W = self.accessor.pull_all()[:C * P]
+        pulled = self.accessor.pull_all()[:C * P]
+        if self._W_buf is None:
+            self._W_buf = pulled.clone()
+        else:
+            self._W_buf.copy_(pulled)
+        self.W = self._W_buf
 
     def local_compute(self) -> None:
         x, y = self.batch
-        W = self._w_matrix()                       # [C, F]
-        # fused forward: GEMM + softmax + label-subtract + CE (one X read)
-        p, loss, correct = ops.mlr_forward(x, W, y)
-        grad = ops.mlr_grad_gemm(p, x)             # [C, F]
-        grad = grad / x.shape[0] + self.a["lam"] * W
-        self.grad_delta = (-self.step_size) * grad
-        self._loss_sum += loss
+        key = id(self.batch)
+        body = self._bodies.get(key)
+        if body is None:
+            W = self._w_matrix()                   # [C, F] view of W_buf
+
+            def body(x=x, y=y, W=W):
+                # forward GEMM + fused softmax + grad GEMM (one graph launch)
+                p, loss, correct = ops.mlr_forward(x, W, y)
+                g = ops.mlr_grad_gemm(p, x)        # [C, F]
+                self.grad_raw = g / x.shape[0] + self.a["lam"] * W
+                self._loss_sum += loss
+                self._correct += correct
+
+            self._bodies[key] = body
+        self._graphs.run(key, body, state=(self._loss_sum, self._correct))
         self._loss_n += x.shape[0]
-        self._correct += correct
 
     def push_update(self) -> None:
         C, P = self.a["num_classes"], self.a["num_parts_per_class"]
-        self.accessor.push_dense(self.grad_delta.view(C * P, -1))
+        # step size applied OUTSIDE the graph (it decays over epochs)
+        delta = (-self.step_size) * self.grad_raw
+        self.accessor.push_dense(delta.view(C * P, -1))
 
     def on_epoch_finished(self, epoch: int) -> None:
         if (epoch + 1) % self.a["decay_period"] == 0:

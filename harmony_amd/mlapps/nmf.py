@@ -145,27 +145,56 @@ class NMFTrainer(Trainer):
         # a host sync and serialize the async step pipeline
         self._sq_err = torch.zeros((), device=ctx.device)
         self.R_batch = None
+        # hipGraph capture of the compute phase: the pulled R rows land in a
+        # stable per-block buffer; step size lives in a device scalar so
+        # epoch decay does not invalidate the graph
+        import os
+
+        from harmony_amd.utils.graphs import GraphRunner
+
+        self._graphs = GraphRunner(
+            enabled=ctx.device.type == "cuda"
+            and os.environ.get("HARMONY_NO_GRAPHS") != "1")
+        self._R_bufs = {}
+        self._bodies = {}
+        self._step_t = torch.tensor(float(self.step_size), device=ctx.device)
 
     def pull_model(self) -> None:
         b = self.batch
         # when a batch touches most of the column space (dense-ish batches),
         # one all-gather beats the two-sided all-to-all key exchange
         if b.uniq_cols.shape[0] * 2 > self.a["num_cols"]:
-            self.R_batch = self.accessor.pull_all()[b.uniq_cols]
+            pulled = self.accessor.pull_all()[b.uniq_cols]
         else:
-            self.R_batch = self.accessor.pull(b.uniq_cols)
+            pulled = self.accessor.pull(b.uniq_cols)
+        buf = self._R_bufs.get(id(b))
+        if buf is None:
+            buf = self._R_bufs.setdefault(id(b), pulled.clone())
+        else:
+            buf.copy_(pulled)
+        self.R_batch = buf
 
     def local_compute(self) -> None:
         b = self.batch
-        L_batch = self.L[b.l_rows]
-        lgrad, rgrad, sq = ops.nmf_grad(L_batch, self.R_batch, b.row_ptr,
-                                        b.col_local, b.vals, self.a["lam"],
-                                        col_sorted=b.col_sorted)
-        # local L update (worker-side SGD apply, same rule as the server's)
-        self.L[b.l_rows] = (L_batch - self.step_size * lgrad).clamp_(
-            0.0, self.a["max_val"])
-        self.rgrad = rgrad
-        self._sq_err += sq
+        key = id(b)
+        body = self._bodies.get(key)
+        if body is None:
+            R_buf = self.R_batch
+
+            def body(b=b, R_buf=R_buf):
+                L_batch = self.L[b.l_rows]
+                lgrad, rgrad, sq = ops.nmf_grad(
+                    L_batch, R_buf, b.row_ptr, b.col_local, b.vals,
+                    self.a["lam"], col_sorted=b.col_sorted)
+                # local L update (worker-side SGD apply, same rule as the
+                # server's; step from a device scalar so decay stays live)
+                self.L[b.l_rows] = (L_batch - self._step_t * lgrad).clamp_(
+                    0.0, self.a["max_val"])
+                self.rgrad = rgrad
+                self._sq_err += sq
+
+            self._bodies[key] = body
+        self._graphs.run(key, body, state=(self.L, self._sq_err))
 
     def push_update(self) -> None:
         # uniq_cols are unique and rgrad is already per-key aggregated by K1
@@ -175,6 +204,7 @@ class NMFTrainer(Trainer):
     def on_epoch_finished(self, epoch: int) -> None:
         if (epoch + 1) % self.a["decay_period"] == 0:
             self.step_size *= self.a["decay_rate"]
+            self._step_t.fill_(float(self.step_size))
 
     def evaluate_model(self):
         out = {"sq_err": float(self._sq_err)}
