@@ -152,9 +152,12 @@ class NMFTrainer(Trainer):
 
         from harmony_amd.utils.graphs import GraphRunner
 
+        # measured on MI355X: graph replay was neutral in the 3-job bench
+        # and slightly SLOWER per-app (kernels are large; launches were not
+        # the bottleneck) — opt-in via HARMONY_GRAPHS=1
         self._graphs = GraphRunner(
             enabled=ctx.device.type == "cuda"
-            and os.environ.get("HARMONY_NO_GRAPHS") != "1")
+            and os.environ.get("HARMONY_GRAPHS") == "1")
         self._R_bufs = {}
         self._bodies = {}
         self._step_t = torch.tensor(float(self.step_size), device=ctx.device)
