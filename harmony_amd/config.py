@@ -97,6 +97,9 @@ class JobConfig:
     optimizer_period: int = 8          # batches between optimization windows
     dashboard_url: Optional[str] = None  # POST epoch metrics here
     trace_path: Optional[str] = None   # JSONL span output (rocTX on GPU)
+    model_chkp_per_epoch: bool = False # snapshot model tables every epoch
+    offline_model_eval: bool = False   # replay epoch snapshots after training
+    chkp_path: str = "/tmp/harmony_chkp_temp"  # reference ChkpTempPath
 
     def to_json(self) -> str:
         return json.dumps(dataclasses.asdict(self))

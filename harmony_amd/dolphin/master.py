@@ -78,10 +78,34 @@ def run_job(job: JobConfig, ctx: ExecutorContext,
                             ctx.rank, ctx.world_size, stream=stream,
                             orchestrator=orch, tracer=tracer,
                             dashboard=dashboard)
+    chkp_mgr = None
+    if job.model_chkp_per_epoch or job.offline_model_eval:
+        # per-epoch model snapshots for offline evaluation (reference
+        # ModelChkpManager, DolphinMaster.java:187-189)
+        from harmony_amd.dolphin.model_eval import ModelChkpManager
+        from harmony_amd.et.checkpoint import CheckpointManager
+
+        cm = CheckpointManager(temp_root=job.chkp_path)
+        chkp_mgr = ModelChkpManager(cm, job.job_id, tables)
+        orig_hook = trainer.on_epoch_finished
+
+        def _hook(epoch, _orig=orig_hook, _mgr=chkp_mgr):
+            _orig(epoch)
+            _mgr.on_epoch_finished(epoch)
+
+        trainer.on_epoch_finished = _hook
     metrics = tasklet.run()
     if tracer is not None:
         tracer.flush()
     ev = trainer.evaluate_model()
     for k, v in (ev or {}).items():
         metrics.add_custom(k, float(v))
+    if job.offline_model_eval and chkp_mgr is not None:
+        from harmony_amd.dolphin.model_eval import ModelEvaluator
+
+        evaluator = ModelEvaluator(chkp_mgr.cm, job.job_id, tables, trainer,
+                                   provider)
+        for cid, res in evaluator.evaluate_all(chkp_mgr.chkp_ids).items():
+            for k, v in (res or {}).items():
+                metrics.add_custom(f"offline/{cid}/{k}", float(v))
     return metrics

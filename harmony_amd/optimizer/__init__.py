@@ -5,11 +5,14 @@ SampleOptimizers, MetricProcessor) + dolphin/plan (PlanCompiler) +
 services/et plan engine (ETPlan op-DAG, PlanExecutorImpl).
 """
 
-from harmony_amd.optimizer.plan import MoveOp, Plan, PlanExecutor, SetBatchShareOp
+from harmony_amd.optimizer.plan import (DropTableOp, MoveOp, Plan,
+                                        PlanExecutor, SetBatchShareOp,
+                                        StartWorkerOp, StopWorkerOp)
 from harmony_amd.optimizer.optimizers import (HomogeneousCostOptimizer,
                                               Optimizer, SampleOptimizers)
 from harmony_amd.optimizer.orchestrator import OptimizationOrchestrator
 
-__all__ = ["Plan", "MoveOp", "SetBatchShareOp", "PlanExecutor", "Optimizer",
+__all__ = ["Plan", "MoveOp", "SetBatchShareOp", "StartWorkerOp",
+           "StopWorkerOp", "DropTableOp", "PlanExecutor", "Optimizer",
            "HomogeneousCostOptimizer", "SampleOptimizers",
            "OptimizationOrchestrator"]

@@ -50,13 +50,22 @@ class OptimizationOrchestrator:
 
     # ---------------------------------------------------------- metric flow
 
+    def _own_version(self) -> int:
+        return sum(t.ownership.version for t in self.tables.values()
+                   if hasattr(t, "ownership"))
+
     def report_batch(self, batch_time: float, comp: float, pull: float,
                      push: float, n_examples: int) -> None:
         self._batches += 1
         key = f"opt/{self.job_id}/m/{self.rank}/{self._batches}"
         if self._batches % self.period == 0:
+            # tagged with the ownership version: metrics measured against a
+            # different block layout are stale and must not steer a plan
+            # (reference loadMetricValidationInfo,
+            # ETOptimizationOrchestrator.java:164-166)
             self.cp.store.set(key, json.dumps(
-                [batch_time, comp, pull, push, n_examples]))
+                [batch_time, comp, pull, push, n_examples,
+                 self._own_version()]))
 
     def _gather_metrics(self, at_batch: int) -> Optional[List[RankMetrics]]:
         out = []
@@ -64,7 +73,10 @@ class OptimizationOrchestrator:
             key = f"opt/{self.job_id}/m/{r}/{at_batch}"
             if not self.cp.flag_set(key):
                 return None   # skip round if any executor's window missing
-            b, c, pl, ps, n = json.loads(self.cp.store.get(key))
+            rec = json.loads(self.cp.store.get(key))
+            b, c, pl, ps, n = rec[:5]
+            if len(rec) > 5 and rec[5] != self._own_version():
+                return None   # stale: measured under a different layout
             prev = self._ema.get(r)
             if prev is None:
                 m = RankMetrics(r, b, c, pl, ps, n)

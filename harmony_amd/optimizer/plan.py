@@ -43,11 +43,50 @@ class SetBatchShareOp:
         return {"op": "share", "shares": list(self.shares)}
 
 
+@dataclass(frozen=True)
+class StopWorkerOp:
+    """Stop a rank's worker role: its batch share drops to zero — it keeps
+    serving its table blocks (pure server), exactly the reference's
+    worker->server switch (PlanCompiler.translateToSwitch)."""
+
+    rank: int
+
+    def to_json(self):
+        return {"op": "stop_worker", "rank": self.rank}
+
+
+@dataclass(frozen=True)
+class StartWorkerOp:
+    rank: int
+    num_batches: int
+
+    def to_json(self):
+        return {"op": "start_worker", "rank": self.rank,
+                "num_batches": self.num_batches}
+
+
+@dataclass(frozen=True)
+class DropTableOp:
+    """Free a table's device shard on every rank (reference DropOp; the
+    jobserver drops per-job tables when a job finishes)."""
+
+    table_id: str
+
+    def to_json(self):
+        return {"op": "drop", "table_id": self.table_id}
+
+
 def op_from_json(d: dict):
     if d["op"] == "move":
         return MoveOp(d["table_id"], tuple(tuple(m) for m in d["moves"]))
     if d["op"] == "share":
         return SetBatchShareOp(tuple(tuple(s) for s in d["shares"]))
+    if d["op"] == "stop_worker":
+        return StopWorkerOp(d["rank"])
+    if d["op"] == "start_worker":
+        return StartWorkerOp(d["rank"], d["num_batches"])
+    if d["op"] == "drop":
+        return DropTableOp(d["table_id"])
     raise ValueError(d)
 
 
@@ -107,4 +146,16 @@ class PlanExecutor:
                             self.world_size, group=self.group)
             elif isinstance(op, SetBatchShareOp):
                 self.batch_shares = dict(op.shares)
+            elif isinstance(op, StopWorkerOp):
+                if self.batch_shares is None:
+                    self.batch_shares = {}
+                self.batch_shares[op.rank] = 0
+            elif isinstance(op, StartWorkerOp):
+                if self.batch_shares is None:
+                    self.batch_shares = {}
+                self.batch_shares[op.rank] = op.num_batches
+            elif isinstance(op, DropTableOp):
+                table = self.tables.pop(op.table_id, None)
+                if table is not None and hasattr(table, "drop_blocks"):
+                    table.drop_blocks(list(table.owned_blocks))
             ready = sorted(ready + dag.on_complete(i))

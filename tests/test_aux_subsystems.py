@@ -163,3 +163,21 @@ def test_hetero_milp_optimizer():
     assert not plan.empty()
     kinds = {type(op).__name__ for op in plan.ops}
     assert "SetBatchShareOp" in kinds
+
+
+def test_offline_eval_via_job_flags(tmp_path):
+    """-model_chkp_per_epoch/-offline_model_eval wired through run_job."""
+    from harmony_amd.config import JobConfig, RuntimeConfig
+    from harmony_amd.dolphin.master import run_job
+    from harmony_amd.runtime.bootstrap import init_executor
+
+    job = JobConfig(job_id="oev", app="mlr", max_num_epochs=2,
+                    num_mini_batches=2, model_chkp_per_epoch=True,
+                    offline_model_eval=True, chkp_path=str(tmp_path),
+                    app_args={"num_classes": 3, "num_features": 16,
+                              "num_parts_per_class": 2, "batch_size": 64,
+                              "step_size": 0.5})
+    ctx = init_executor(RuntimeConfig(device="cpu"))
+    s = run_job(job, ctx).summary()
+    assert "offline/epoch0/accuracy" in s
+    assert "offline/epoch1/accuracy" in s
