@@ -19,11 +19,18 @@ from typing import Any, Dict, List, Tuple
 
 from harmony_amd.config import DEFAULT_JOBSERVER_PORT, JobConfig
 
+def _bool(s: str) -> bool:
+    return s.lower() in ("true", "1", "yes")
+
+
 # JobConfig-level flags (reference DolphinParameters.java); everything else
 # goes into app_args for the app to interpret.
 _JOB_FIELDS = {"max_num_epochs": int, "num_mini_batches": int,
                "num_worker_blocks": int, "clock_slack": int,
-               "num_trainer_threads": int}
+               "num_trainer_threads": int, "optimizer": str,
+               "optimizer_period": int, "dashboard_url": str,
+               "trace_path": str, "chkp_path": str,
+               "model_chkp_per_epoch": _bool, "offline_model_eval": _bool}
 
 
 def _parse_flags(argv: List[str]) -> Tuple[Dict[str, Any], Dict[str, Any], bool]:
