@@ -175,9 +175,8 @@ class LDATrainer(Trainer):
 
     def pull_model(self) -> None:
         pulled = self.accessor.pull(self.batch.pull_keys)
-        # fold the batch-constant (n_wk+b)/(n_k+Vb) term once per pull
-        self.wtprime = ops.lda_wtprime(pulled[:-1], pulled[-1],
-                                       self.a["beta"], self.a["num_vocabs"])
+        self.word_topic = pulled[:-1]          # [n_uniq_words, K]
+        self.topic_sum = pulled[-1]            # [K]
 
     def local_compute(self) -> None:
         b = self.batch
@@ -185,8 +184,10 @@ class LDATrainer(Trainer):
         old = z.clone()
         self._step += 1
         dt = self.doc_topic[b.doc_ids]          # gather copy
-        ops.lda_gibbs(dt, self.wtprime, b.doc_offsets, b.word_local, z,
-                      self.a["alpha"], self._epoch_seed + self._step)
+        ops.lda_gibbs(dt, self.word_topic,
+                      self.topic_sum, b.doc_offsets, b.word_local, z,
+                      self.a["alpha"], self.a["beta"], self.a["num_vocabs"],
+                      self._epoch_seed + self._step)
         self.doc_topic[b.doc_ids] = dt          # write back
         self._old_z = old
         self._new_z = z

@@ -161,46 +161,41 @@ def nmf_grad(L: torch.Tensor, R: torch.Tensor, row_ptr: torch.Tensor,
 # K7: LDA collapsed Gibbs sampling over a doc block
 # ---------------------------------------------------------------------------
 
-def lda_wtprime(word_topic: torch.Tensor, topic_sum: torch.Tensor,
-                beta: float, num_vocabs: int) -> torch.Tensor:
-    """Batch-constant sampler term: wtprime[w][k] = (n_wk + b)/(n_k + V b).
-    Precomputed once per pull so the per-token kernel loop does one
-    convert+add+mul per topic (measured VALU cut on the Gibbs kernel)."""
-    inv = 1.0 / (topic_sum.float() + num_vocabs * beta)
-    return (word_topic.float() + beta) * inv
-
-
-def lda_gibbs(doc_topic: torch.Tensor, wtprime: torch.Tensor,
-              doc_offsets: torch.Tensor, word_ids: torch.Tensor,
-              assignments: torch.Tensor, alpha: float, seed: int
+def lda_gibbs(doc_topic: torch.Tensor, word_topic: torch.Tensor,
+              topic_sum: torch.Tensor, doc_offsets: torch.Tensor,
+              word_ids: torch.Tensor, assignments: torch.Tensor,
+              alpha: float, beta: float, num_vocabs: int, seed: int
               ) -> torch.Tensor:
     """One Gibbs sweep over the batch's tokens; returns new assignments.
 
-    doc_topic: [n_docs, K] int32 (updated in place); wtprime: [n_words, K]
-    float32 from lda_wtprime (batch-stale snapshot); doc_offsets: [n_docs+1]
-    CSR over tokens; word_ids: LOCAL word index per token; assignments:
-    [n_tokens] int32 (updated in place).
+    doc_topic: [n_docs, K] int32 (updated in place),
+    word_topic: [n_words_in_batch, K] (pulled rows, LOCAL indices; treated as
+    fixed within the sweep — batch-stale counts, see mlapps/lda.py),
+    topic_sum: [K], doc_offsets: [n_docs+1] CSR over tokens, word_ids: local
+    word index per token, assignments: [n_tokens] int32 (updated in place).
 
-    p(k) = (n_dk + alpha) * wtprime[w][k]
-    (reference SparseLDASampler.java:141-274's s/r/q bucket decomposition is
-    a CPU sparsity optimization; on CDNA4 the dense K-way distribution is
-    computed by a lane group per document — see ops/csrc/lda.hip.)
+    p(k) ∝ (n_dk + α) (n_wk + β) / (n_k + V β)
+    (reference SparseLDASampler.java:141-274's s/r/q bucket decomposition is a
+    CPU sparsity optimization; on CDNA4 the dense K-way distribution is
+    computed by a wave per document — see ops/csrc/lda.hip.)
 
-    The torch path mirrors the device kernel: same counter-based RNG keyed
-    by (seed, token index), same float32 terms, same "first k with
+    The torch path mirrors the device kernel: same counter-based RNG keyed by
+    (seed, token index), same float32 probability terms, same "first k with
     cumsum > u" draw — CPU and GPU agree except at float summation-order
     tie-breaks (tests require >=99% identical samples + exact invariants)."""
-    if _use_hip(wtprime):
-        return _hip.lda_gibbs(doc_topic, wtprime.contiguous(), doc_offsets,
-                              word_ids, assignments, float(alpha), int(seed))
+    if _use_hip(word_topic):
+        return _hip.lda_gibbs(doc_topic, word_topic, topic_sum, doc_offsets,
+                              word_ids, assignments, float(alpha), float(beta),
+                              int(num_vocabs), int(seed))
     from harmony_amd.ops.rng import rng_uniform
 
     n_docs = doc_topic.shape[0]
-    lengths = doc_offsets[1:] - doc_offsets[:-1]
-    max_len = int(lengths.max()) if n_docs else 0
+    inv_den = (1.0 / (topic_sum.float() + num_vocabs * beta)).float()   # [K]
     # Lockstep over token positions: docs advance one token per step so the
     # per-doc sequential dependency (n_dk) is honored while steps stay
     # vectorized over docs.
+    lengths = doc_offsets[1:] - doc_offsets[:-1]
+    max_len = int(lengths.max()) if n_docs else 0
     for pos in range(max_len):
         active = (lengths > pos).nonzero(as_tuple=True)[0]
         tok = doc_offsets[active] + pos
@@ -209,18 +204,35 @@ def lda_gibbs(doc_topic: torch.Tensor, wtprime: torch.Tensor,
         ar = torch.arange(active.shape[0])
         dt = doc_topic[active]
         dt[ar, old] -= 1
-        probs = (dt.float() + alpha) * wtprime[w]    # [n_active, K]
+        probs = ((dt.float() + alpha) * (word_topic[w].float() + beta)
+                 * inv_den)                                  # [n_active, K]
         tot = probs.sum(dim=1)
         u = rng_uniform(seed & 0xFFFFFFFF, tok.long()) * tot
         cdf = probs.cumsum(dim=1)
         new = torch.searchsorted(cdf, u.unsqueeze(1).to(cdf.dtype),
                                  right=True).squeeze(1)
         bad = new >= probs.shape[1]
-        new = torch.where(bad, old, new)             # numeric edge
+        new = torch.where(bad, old, new)                     # numeric edge
         dt[ar, new] += 1
         doc_topic[active] = dt
         assignments[tok] = new.to(assignments.dtype)
     return assignments
+
+
+def lda_apply_pairs(shard: torch.Tensor, rows: torch.Tensor,
+                    old_t: torch.Tensor, new_t: torch.Tensor) -> None:
+    """Apply (row, old_topic, new_topic) ±1 count pairs to the word-topic
+    shard (the reference's TopicChanges wire format — 12 B per changed token
+    instead of a dense K-int row per touched word)."""
+    if _use_hip(shard):
+        _hip.lda_apply_pairs(shard, rows.contiguous(),
+                             old_t.contiguous(), new_t.contiguous())
+        return
+    K = shard.shape[1]
+    flat = shard.view(-1)
+    ones = torch.ones(rows.shape[0], dtype=shard.dtype, device=shard.device)
+    flat.scatter_add_(0, rows * K + old_t.long(), -ones)
+    flat.scatter_add_(0, rows * K + new_t.long(), ones)
 
 
 def lda_apply_all(shard, word_rows, old_t, new_t, summary_row: int) -> None:

@@ -44,23 +44,25 @@ __device__ __forceinline__ float group_inclusive_scan(float v, int lane) {
 
 template <int G>
 __global__ void lda_gibbs_kernel(int* __restrict__ doc_topic,
-                                 const float* __restrict__ wtprime,
+                                 const int* __restrict__ word_topic,
+                                 const int* __restrict__ topic_sum,
                                  const int64_t* __restrict__ doc_offsets,
                                  const int64_t* __restrict__ word_ids,
                                  int* __restrict__ z,
-                                 float alpha,
+                                 float alpha, float beta, float vbeta,
                                  int n_docs, int K, unsigned int seed) {
-  // wtprime[w][k] = (n_wk + beta) / (n_k + V*beta), precomputed once per
-  // pull (it is batch-constant): the per-token inner loop is then one
-  // int-convert + add + mul per topic instead of two converts, an LDS
-  // denominator read and two muls.
   constexpr int GROUPS = BLOCK_THREADS / G;
-  extern __shared__ int smem[];                  // [GROUPS][K] nd
+  extern __shared__ int smem[];                  // [GROUPS][K] nd + [K] invden
   const int group = threadIdx.x / G;
   const int lane = threadIdx.x % G;
   const int sub = (threadIdx.x % WAVE) / G;      // group index within wave
   int* nd = smem + group * K;
+  float* invden = (float*)(smem + GROUPS * K);
   const int nchunk = (K + G - 1) / G;
+
+  for (int idx = threadIdx.x; idx < K; idx += blockDim.x)
+    invden[idx] = 1.0f / ((float)topic_sum[idx] + vbeta);
+  __syncthreads();
 
   const int doc = blockIdx.x * GROUPS + group;
   if (doc >= n_docs) return;
@@ -87,7 +89,8 @@ __global__ void lda_gibbs_kernel(int* __restrict__ doc_topic,
     for (int c = 0; c < MAXC; ++c) {
       int idx = c * G + lane;
       if (c < nchunk && idx < K) {
-        pr[c] = ((float)nd[idx] + alpha) * wtprime[w * K + idx];
+        pr[c] = ((float)nd[idx] + alpha) *
+                ((float)word_topic[w * K + idx] + beta) * invden[idx];
       } else {
         pr[c] = 0.f;
       }
@@ -212,14 +215,15 @@ void lda_apply_pairs(torch::Tensor shard, torch::Tensor rows,
                      old_t.data_ptr<int>(), new_t.data_ptr<int>(), n, K);
 }
 
-torch::Tensor lda_gibbs(torch::Tensor doc_topic, torch::Tensor wtprime,
-                        torch::Tensor doc_offsets,
+torch::Tensor lda_gibbs(torch::Tensor doc_topic, torch::Tensor word_topic,
+                        torch::Tensor topic_sum, torch::Tensor doc_offsets,
                         torch::Tensor word_ids, torch::Tensor assignments,
-                        double alpha, int64_t seed) {
-  CHECK_IN(doc_topic); CHECK_IN(wtprime);
+                        double alpha, double beta, int64_t num_vocabs,
+                        int64_t seed) {
+  CHECK_IN(doc_topic); CHECK_IN(word_topic); CHECK_IN(topic_sum);
   CHECK_IN(doc_offsets); CHECK_IN(word_ids); CHECK_IN(assignments);
   TORCH_CHECK(doc_topic.dtype() == torch::kInt32);
-  TORCH_CHECK(wtprime.dtype() == torch::kFloat32);
+  TORCH_CHECK(word_topic.dtype() == torch::kInt32);
   const int D = doc_topic.size(0), K = doc_topic.size(1);
   TORCH_CHECK(K <= WAVE * MAXC, "num_topics > ", WAVE * MAXC, " unsupported");
   if (D == 0) return assignments;
@@ -230,24 +234,28 @@ torch::Tensor lda_gibbs(torch::Tensor doc_topic, torch::Tensor wtprime,
   const int groups = BLOCK_THREADS / G;
   dim3 blk(BLOCK_THREADS);
   dim3 grid((D + groups - 1) / groups);
-  const size_t shmem = (size_t)(groups * K) * 4;
+  const size_t shmem = (size_t)(groups * K + K) * 4;
   if (G == 32) {
     hipLaunchKernelGGL(lda_gibbs_kernel<32>, grid, blk, shmem,
                        current_stream(),
-                       doc_topic.data_ptr<int>(), wtprime.data_ptr<float>(),
+                       doc_topic.data_ptr<int>(), word_topic.data_ptr<int>(),
+                       topic_sum.data_ptr<int>(),
                        doc_offsets.data_ptr<int64_t>(),
                        word_ids.data_ptr<int64_t>(),
                        assignments.data_ptr<int>(),
-                       (float)alpha, D, K,
+                       (float)alpha, (float)beta,
+                       (float)(num_vocabs * beta), D, K,
                        (unsigned int)(seed & 0xffffffff));
   } else {
     hipLaunchKernelGGL(lda_gibbs_kernel<64>, grid, blk, shmem,
                        current_stream(),
-                       doc_topic.data_ptr<int>(), wtprime.data_ptr<float>(),
+                       doc_topic.data_ptr<int>(), word_topic.data_ptr<int>(),
+                       topic_sum.data_ptr<int>(),
                        doc_offsets.data_ptr<int64_t>(),
                        word_ids.data_ptr<int64_t>(),
                        assignments.data_ptr<int>(),
-                       (float)alpha, D, K,
+                       (float)alpha, (float)beta,
+                       (float)(num_vocabs * beta), D, K,
                        (unsigned int)(seed & 0xffffffff));
   }
   return assignments;

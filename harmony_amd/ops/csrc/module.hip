@@ -16,10 +16,11 @@ std::vector<torch::Tensor> nmf_grad_twopass(
 std::vector<torch::Tensor> mlr_fwd(torch::Tensor X, torch::Tensor W,
                                    torch::Tensor labels);
 torch::Tensor mlr_grad(torch::Tensor P, torch::Tensor X);
-torch::Tensor lda_gibbs(torch::Tensor doc_topic, torch::Tensor wtprime,
-                        torch::Tensor doc_offsets,
+torch::Tensor lda_gibbs(torch::Tensor doc_topic, torch::Tensor word_topic,
+                        torch::Tensor topic_sum, torch::Tensor doc_offsets,
                         torch::Tensor word_ids, torch::Tensor assignments,
-                        double alpha, int64_t seed);
+                        double alpha, double beta, int64_t num_vocabs,
+                        int64_t seed);
 void lda_apply_pairs(torch::Tensor shard, torch::Tensor rows,
                      torch::Tensor old_t, torch::Tensor new_t);
 void lda_apply_all(torch::Tensor shard, torch::Tensor word_rows,

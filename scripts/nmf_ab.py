@@ -68,9 +68,8 @@ def main():
     wt.view(-1).scatter_add_(0, wl * K + z.long(),
                              torch.ones(D * T, dtype=torch.int32, device="cuda"))
     ts_sum = wt.sum(0).to(torch.int32)
-    wtp = ((wt.float() + 0.01) / (ts_sum.float() + V * 0.01))
     variants["lda_gibbs_sorted"] = lambda: hip.lda_gibbs(
-        dt, wtp, offs, wl, z, 0.1, 42)
+        dt, wt, ts_sum, offs, wl, z, 0.1, 0.01, V, 42)
     # unsorted tokens for comparison (cache-locality delta)
     w2 = (u * u * V).long().clamp_(0, V - 1)
     uw2, wl2 = torch.unique(w2, return_inverse=True)
@@ -79,9 +78,8 @@ def main():
     wt2 = torch.zeros(uw2.shape[0], K, dtype=torch.int32, device="cuda")
     wt2.view(-1).scatter_add_(0, wl2 * K + z2.long(),
                               torch.ones(D * T, dtype=torch.int32, device="cuda"))
-    wtp2 = ((wt2.float() + 0.01) / (ts_sum.float() + V * 0.01))
     variants["lda_gibbs_unsorted"] = lambda: hip.lda_gibbs(
-        dt, wtp2, offs, wl2, z2, 0.1, 42)
+        dt, wt2, ts_sum, offs, wl2, z2, 0.1, 0.01, V, 42)
 
     rounds = {kk: [] for kk in variants}
     for r in range(3):

@@ -72,8 +72,8 @@ def test_lda_gibbs_invariants():
         word_topic[w, z[i]] += 1
     topic_sum = word_topic.sum(0).to(torch.int32)
     z2 = z.clone()
-    wtp = ops.lda_wtprime(word_topic, topic_sum, 0.01, V)
-    ops.lda_gibbs(doc_topic, wtp, offsets, word_ids, z2, 0.1, seed=1234)
+    ops.lda_gibbs(doc_topic, word_topic, topic_sum, offsets, word_ids, z2,
+                  0.1, 0.01, V, seed=1234)
     # invariants: every doc still has tokens_per_doc assignments; all topics valid
     assert (doc_topic.sum(1) == tokens_per_doc).all()
     assert int(z2.min()) >= 0 and int(z2.max()) < K

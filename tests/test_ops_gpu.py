@@ -91,15 +91,15 @@ def test_lda_gibbs_gpu_matches_cpu_rng():
         0, word_ids * K + z0.long(), torch.ones(D * T, dtype=torch.int32))
     topic_sum = word_topic.sum(0).to(torch.int32)
 
-    wtp = ops.lda_wtprime(word_topic, topic_sum, 0.01, V)
     z_cpu = z0.clone()
     dt_cpu = doc_topic.clone()
-    ops.lda_gibbs(dt_cpu, wtp, offsets, word_ids, z_cpu, 0.1, seed=777)
+    ops.lda_gibbs(dt_cpu, word_topic, topic_sum, offsets, word_ids, z_cpu,
+                  0.1, 0.01, V, seed=777)
 
     z_gpu = z0.clone().cuda()
     dt_gpu = doc_topic.clone().cuda()
-    ops.lda_gibbs(dt_gpu, wtp.cuda(), offsets.cuda(),
-                  word_ids.cuda(), z_gpu, 0.1, seed=777)
+    ops.lda_gibbs(dt_gpu, word_topic.cuda(), topic_sum.cuda(), offsets.cuda(),
+                  word_ids.cuda(), z_gpu, 0.1, 0.01, V, seed=777)
 
     match = (z_gpu.cpu() == z_cpu).float().mean()
     # same counter RNG + same f32 terms; only summation-order ties differ
