@@ -229,7 +229,8 @@ class LDATrainer(Trainer):
                                   apply_pairs)
             summ = (torch.bincount(pn.long(), minlength=K)
                     - torch.bincount(po.long(), minlength=K)).to(torch.int32)
-            self.accessor.push(b.pull_keys[-1:], summ.unsqueeze(0))
+            self.accessor.push(b.pull_keys[-1:], summ.unsqueeze(0),
+                               assume_unique=True)
         self.accessor.metrics["total_push_time_sec"] += time.perf_counter() - t0
 
     def num_batch_examples(self) -> int:

@@ -170,12 +170,16 @@ class NMFTrainer(Trainer):
             pulled = self.accessor.pull_all()[b.uniq_cols]
         else:
             pulled = self.accessor.pull(b.uniq_cols)
-        buf = self._R_bufs.get(id(b))
-        if buf is None:
-            buf = self._R_bufs.setdefault(id(b), pulled.clone())
+        if self._graphs.enabled:
+            # graphs need a stable pointer -> stage into a per-block buffer
+            buf = self._R_bufs.get(id(b))
+            if buf is None:
+                buf = self._R_bufs.setdefault(id(b), pulled.clone())
+            else:
+                buf.copy_(pulled)
+            self.R_batch = buf
         else:
-            buf.copy_(pulled)
-        self.R_batch = buf
+            self.R_batch = pulled
 
     def local_compute(self) -> None:
         b = self.batch

@@ -231,11 +231,23 @@ torch::Tensor lda_gibbs(torch::Tensor doc_topic, torch::Tensor word_topic,
   const char* env = getenv("HARMONY_LDA_G");      // A/B override
   if (env && atoi(env) == 64) G = 64;
   if (env && atoi(env) == 32 && K <= 32 * MAXC) G = 32;
+  if (env && atoi(env) == 16 && K <= 16 * MAXC) G = 16;
   const int groups = BLOCK_THREADS / G;
   dim3 blk(BLOCK_THREADS);
   dim3 grid((D + groups - 1) / groups);
   const size_t shmem = (size_t)(groups * K + K) * 4;
-  if (G == 32) {
+  if (G == 16) {
+    hipLaunchKernelGGL(lda_gibbs_kernel<16>, grid, blk, shmem,
+                       current_stream(),
+                       doc_topic.data_ptr<int>(), word_topic.data_ptr<int>(),
+                       topic_sum.data_ptr<int>(),
+                       doc_offsets.data_ptr<int64_t>(),
+                       word_ids.data_ptr<int64_t>(),
+                       assignments.data_ptr<int>(),
+                       (float)alpha, (float)beta,
+                       (float)(num_vocabs * beta), D, K,
+                       (unsigned int)(seed & 0xffffffff));
+  } else if (G == 32) {
     hipLaunchKernelGGL(lda_gibbs_kernel<32>, grid, blk, shmem,
                        current_stream(),
                        doc_topic.data_ptr<int>(), word_topic.data_ptr<int>(),
