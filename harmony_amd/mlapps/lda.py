@@ -38,7 +38,8 @@ MODEL_TABLE = "lda_model"
 
 def defaults(job: JobConfig) -> dict:
     a = dict(num_docs=32768, num_vocabs=30000, num_topics=256,
-             tokens_per_doc=64, alpha=0.1, beta=0.01, docs_per_batch=4096)
+             tokens_per_doc=64, alpha=0.1, beta=0.01, docs_per_batch=4096,
+             sampler="exact")   # "exact" (dense Gibbs) | "alias" (MH, K7b)
     a.update(job.app_args)
     return a
 
@@ -177,6 +178,11 @@ class LDATrainer(Trainer):
         pulled = self.accessor.pull(self.batch.pull_keys)
         self.word_topic = pulled[:-1]          # [n_uniq_words, K]
         self.topic_sum = pulled[-1]            # [K]
+        if self.a["sampler"] == "alias":
+            # per-word Vose alias tables over the stale word factor (K7b)
+            self._alias = ops.lda_alias_build(
+                self.word_topic, self.topic_sum, self.a["beta"],
+                self.a["num_vocabs"])
 
     def local_compute(self) -> None:
         b = self.batch
@@ -184,10 +190,17 @@ class LDATrainer(Trainer):
         old = z.clone()
         self._step += 1
         dt = self.doc_topic[b.doc_ids]          # gather copy
-        ops.lda_gibbs(dt, self.word_topic,
-                      self.topic_sum, b.doc_offsets, b.word_local, z,
-                      self.a["alpha"], self.a["beta"], self.a["num_vocabs"],
-                      self._epoch_seed + self._step)
+        if self.a["sampler"] == "alias":
+            prob, alias, _, invden = self._alias
+            ops.lda_mh(dt, self.word_topic, invden, prob, alias,
+                       b.doc_offsets, b.word_local, z, self.a["alpha"],
+                       self.a["beta"], self._epoch_seed + self._step)
+        else:
+            ops.lda_gibbs(dt, self.word_topic,
+                          self.topic_sum, b.doc_offsets, b.word_local, z,
+                          self.a["alpha"], self.a["beta"],
+                          self.a["num_vocabs"],
+                          self._epoch_seed + self._step)
         self.doc_topic[b.doc_ids] = dt          # write back
         self._old_z = old
         self._new_z = z

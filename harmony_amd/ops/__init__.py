@@ -235,6 +235,108 @@ def lda_apply_pairs(shard: torch.Tensor, rows: torch.Tensor,
     flat.scatter_add_(0, rows * K + new_t.long(), ones)
 
 
+def lda_alias_build(word_topic: torch.Tensor, topic_sum: torch.Tensor,
+                    beta: float, num_vocabs: int):
+    """Per-word Vose alias tables over the (batch-stale) word factor
+    q_w(k) = (n_wk + b)/(n_k + V b). Returns (prob, alias, qsum, invden).
+    The construction order is deterministic and identical on CPU and GPU so
+    MH sweeps can be compared sample-by-sample."""
+    if _use_hip(word_topic):
+        return tuple(_hip.lda_alias_build(word_topic.contiguous(),
+                                          topic_sum.contiguous(),
+                                          float(beta), int(num_vocabs)))
+    rows, K = word_topic.shape
+    invden = (1.0 / (topic_sum.float() + num_vocabs * beta)).contiguous()
+    p0 = (word_topic.float() + beta) * invden
+    qsum = p0.sum(dim=1)
+    prob = p0 * (K / qsum).unsqueeze(1)
+    alias = torch.empty((rows, K), dtype=torch.int32)
+    probo = prob.clone()
+    for w in range(rows):
+        pr = probo[w]
+        small = [k for k in range(K) if float(pr[k]) < 1.0]
+        large = [k for k in range(K) if float(pr[k]) >= 1.0]
+        small.reverse(); large.reverse()   # match kernel's pop order
+        al = alias[w]
+        while small and large:
+            sm = small.pop()
+            lg = large[-1]
+            al[sm] = lg
+            rem = float(pr[lg]) + float(pr[sm]) - 1.0
+            pr[lg] = rem
+            large.pop()
+            if rem < 1.0:
+                small.append(lg)
+            else:
+                large.append(lg)
+        for lg in reversed(large):
+            pr[lg] = 1.0
+            al[lg] = lg
+        for sm in reversed(small):
+            pr[sm] = 1.0
+            al[sm] = sm
+    return probo, alias, qsum, invden
+
+
+def lda_mh(doc_topic: torch.Tensor, word_topic: torch.Tensor,
+           invden: torch.Tensor, prob: torch.Tensor, alias: torch.Tensor,
+           doc_offsets: torch.Tensor, word_ids: torch.Tensor,
+           assignments: torch.Tensor, alpha: float, beta: float,
+           seed: int) -> torch.Tensor:
+    """One Metropolis-Hastings alias sweep (K7b; see ops/csrc/lda_alias.hip
+    for the proposal/acceptance derivation). Same stationary distribution as
+    the exact sampler under the batch-stale word-topic snapshot; O(1) per
+    token. Torch path mirrors the kernel's RNG and float math."""
+    if _use_hip(word_topic):
+        return _hip.lda_mh(doc_topic, word_topic, invden, prob, alias,
+                           doc_offsets, word_ids, assignments, float(alpha),
+                           float(beta), int(seed))
+    from harmony_amd.ops.rng import rng_uniform
+
+    K = word_topic.shape[1]
+    lengths = doc_offsets[1:] - doc_offsets[:-1]
+    n_docs = doc_topic.shape[0]
+    max_len = int(lengths.max()) if n_docs else 0
+    aK = alpha * K
+    for pos in range(max_len):
+        active = (lengths > pos).nonzero(as_tuple=True)[0]
+        tok = doc_offsets[active] + pos
+        w = word_ids[tok].long()
+        s = assignments[tok].long()
+        ar = torch.arange(active.shape[0])
+        dt = doc_topic[active]
+        dt[ar, s] -= 1
+        c0 = (tok * 8).long()
+        sd = seed & 0xFFFFFFFF
+        # word proposal
+        u0 = rng_uniform(sd, c0) * K
+        bins = u0.long().clamp_(max=K - 1)
+        frac = u0 - bins.float()
+        t1 = torch.where(frac < prob[w, bins], bins, alias[w, bins].long())
+        a1 = (dt[ar, t1].float() + alpha) / (dt[ar, s].float() + alpha)
+        s = torch.where(rng_uniform(sd, c0 + 1) < a1, t1, s)
+        # doc proposal
+        Ld = lengths[active].float()
+        uni = rng_uniform(sd, c0 + 2) < (aK / (aK + Ld))
+        t_uni = (rng_uniform(sd, c0 + 3) * K).long().clamp_(max=K - 1)
+        j = doc_offsets[active] + (rng_uniform(sd, c0 + 4) * Ld).long()
+        j = torch.minimum(j, doc_offsets[active] + lengths[active] - 1)
+        t_copy = torch.where(j == tok, s, assignments[j].long())
+        t2 = torch.where(uni, t_uni, t_copy)
+        nds = dt[ar, s].float()
+        ndt = dt[ar, t2].float()
+        qs = nds + 1.0 + alpha
+        qt = ndt + (t2 == s).float() + alpha
+        pis = (nds + alpha) * (word_topic[w, s].float() + beta) * invden[s]
+        pit = (ndt + alpha) * (word_topic[w, t2].float() + beta) * invden[t2]
+        a2 = (pit * qs) / (pis * qt)
+        s = torch.where(rng_uniform(sd, c0 + 5) < a2, t2, s)
+        dt[ar, s] += 1
+        doc_topic[active] = dt
+        assignments[tok] = s.to(assignments.dtype)
+    return assignments
+
+
 def lda_apply_all(shard, word_rows, old_t, new_t, summary_row: int) -> None:
     """Fused single-owner update: per token, if the topic changed, apply
     ±1 to the word row and the summary row (K9; replaces the whole

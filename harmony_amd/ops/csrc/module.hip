@@ -26,6 +26,14 @@ void lda_apply_pairs(torch::Tensor shard, torch::Tensor rows,
 void lda_apply_all(torch::Tensor shard, torch::Tensor word_rows,
                    torch::Tensor old_t, torch::Tensor new_t,
                    int64_t summary_row);
+std::vector<torch::Tensor> lda_alias_build(torch::Tensor word_topic,
+                                           torch::Tensor topic_sum,
+                                           double beta, int64_t num_vocabs);
+torch::Tensor lda_mh(torch::Tensor doc_topic, torch::Tensor word_topic,
+                     torch::Tensor invden, torch::Tensor prob,
+                     torch::Tensor alias, torch::Tensor doc_offsets,
+                     torch::Tensor word_ids, torch::Tensor assignments,
+                     double alpha, double beta, int64_t seed);
 void scatter_apply(torch::Tensor shard, torch::Tensor rows,
                    torch::Tensor deltas, int64_t mode, double step,
                    double maxval);
@@ -50,6 +58,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "apply TopicChanges +/-1 pairs to the word-topic shard (K9 sparse)");
   m.def("lda_apply_all", &lda_apply_all,
         "fused local token-delta + summary update (K9, single-owner path)");
+  m.def("lda_alias_build", &lda_alias_build,
+        "per-word Vose alias tables over the stale word factor (K7b)");
+  m.def("lda_mh", &lda_mh,
+        "Metropolis-Hastings alias LDA sweep, thread-per-doc (K7b)");
   m.def("scatter_apply", &scatter_apply, "owner-side sparse update (K9)");
   m.def("dense_apply", &dense_apply, "owner-side dense update (K3)");
   m.def("parse_nmf_bytes", &parse_nmf_bytes, "native NMF text parser");

@@ -99,3 +99,47 @@ def test_rng_matches_scalar_reference():
     out = rng_u32(0xDEADBEEF, ctrs)
     for i, c in enumerate(ctrs.tolist()):
         assert int(out[i]) == ref(0xDEADBEEF, c), f"ctr {c}"
+
+
+def test_lda_alias_build_valid():
+    torch.manual_seed(3)
+    V, K = 40, 12
+    wt = torch.randint(0, 30, (V, K), dtype=torch.int32)
+    ts = wt.sum(0).to(torch.int32)
+    prob, alias, qsum, invden = ops.lda_alias_build(wt, ts, 0.01, V)
+    assert prob.shape == (V, K) and alias.shape == (V, K)
+    assert (prob >= 0).all() and (prob <= 1.0 + 1e-5).all()
+    assert int(alias.min()) >= 0 and int(alias.max()) < K
+    # alias sampling reproduces q_w: estimate via exhaustive bins
+    # P(k) = sum_bins [prob[b]*1{b==k} + (1-prob[b])*1{alias[b]==k}] / K
+    for w in range(0, V, 7):
+        est = torch.zeros(K)
+        for b in range(K):
+            est[b] += float(prob[w, b])
+            est[int(alias[w, b])] += 1.0 - float(prob[w, b])
+        est /= K
+        q = (wt[w].float() + 0.01) * invden
+        q = q / q.sum()
+        assert torch.allclose(est, q, atol=1e-4), w
+
+
+def test_lda_mh_invariants_and_mixing():
+    torch.manual_seed(4)
+    D, K, V, T = 32, 12, 60, 24
+    word_ids = torch.randint(0, V, (D * T,))
+    z = torch.randint(0, K, (D * T,), dtype=torch.int32)
+    offsets = torch.arange(0, (D + 1) * T, T)
+    dt = torch.zeros(D, K, dtype=torch.int32)
+    dt.view(-1).scatter_add_(0, torch.arange(D).repeat_interleave(T) * K
+                             + z.long(), torch.ones(D * T, dtype=torch.int32))
+    wt = torch.zeros(V, K, dtype=torch.int32)
+    wt.view(-1).scatter_add_(0, word_ids * K + z.long(),
+                             torch.ones(D * T, dtype=torch.int32))
+    ts = wt.sum(0).to(torch.int32)
+    prob, alias, _, invden = ops.lda_alias_build(wt, ts, 0.01, V)
+    z2 = z.clone()
+    ops.lda_mh(dt, wt, invden, prob, alias, offsets, word_ids, z2,
+               0.1, 0.01, seed=99)
+    assert (dt.sum(1) == T).all()          # token conservation per doc
+    assert int(z2.min()) >= 0 and int(z2.max()) < K
+    assert not torch.equal(z, z2)          # chain moved

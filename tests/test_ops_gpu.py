@@ -244,3 +244,48 @@ def test_mlr_grad_gemm_vs_torch():
     g = ops._load_hip().mlr_grad(P.contiguous(), X.contiguous())
     ref = P.cpu().double().t() @ X.cpu().double()
     assert torch.allclose(g.cpu().double(), ref, atol=1e-2, rtol=1e-4)
+
+
+def test_lda_mh_gpu_matches_cpu():
+    torch.manual_seed(12)
+    D, K, V, T = 128, 64, 500, 40
+    word_ids = torch.randint(0, V, (D * T,))
+    z0 = torch.randint(0, K, (D * T,), dtype=torch.int32)
+    offsets = torch.arange(0, (D + 1) * T, T)
+    dt = torch.zeros(D, K, dtype=torch.int32)
+    dt.view(-1).scatter_add_(0, torch.arange(D).repeat_interleave(T) * K
+                             + z0.long(), torch.ones(D * T, dtype=torch.int32))
+    wt = torch.zeros(V, K, dtype=torch.int32)
+    wt.view(-1).scatter_add_(0, word_ids * K + z0.long(),
+                             torch.ones(D * T, dtype=torch.int32))
+    ts = wt.sum(0).to(torch.int32)
+    prob_c, alias_c, _, inv_c = ops.lda_alias_build(wt, ts, 0.01, V)
+    prob_g, alias_g, _, inv_g = ops.lda_alias_build(wt.cuda(), ts.cuda(),
+                                                    0.01, V)
+    # alias tables must be bit-identical (deterministic construction)
+    assert torch.equal(alias_c, alias_g.cpu())
+    assert torch.allclose(prob_c, prob_g.cpu(), atol=1e-6)
+    z_cpu, dt_cpu = z0.clone(), dt.clone()
+    ops.lda_mh(dt_cpu, wt, inv_c, prob_c, alias_c, offsets, word_ids, z_cpu,
+               0.1, 0.01, seed=321)
+    z_gpu, dt_gpu = z0.clone().cuda(), dt.clone().cuda()
+    ops.lda_mh(dt_gpu, wt.cuda(), inv_g, prob_g, alias_g, offsets.cuda(),
+               word_ids.cuda(), z_gpu, 0.1, 0.01, seed=321)
+    match = (z_gpu.cpu() == z_cpu).float().mean()
+    assert float(match) > 0.99, float(match)
+    assert (dt_gpu.cpu().sum(1) == T).all()
+
+
+def test_lda_alias_job_end_to_end_gpu():
+    from harmony_amd.config import JobConfig, RuntimeConfig
+    from harmony_amd.dolphin.master import run_job
+    from harmony_amd.runtime.bootstrap import init_executor
+
+    ctx = init_executor(RuntimeConfig(device="cuda"))
+    job = JobConfig(job_id="g_lda_mh", app="lda", max_num_epochs=2,
+                    num_mini_batches=2,
+                    app_args={"num_vocabs": 2000, "num_topics": 64,
+                              "tokens_per_doc": 32, "docs_per_batch": 256,
+                              "sampler": "alias"})
+    m = run_job(job, ctx)
+    assert m.summary()["num_batches"] == 4
