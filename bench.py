@@ -47,6 +47,8 @@ def parse_args():
     p.add_argument("--lda-topics", type=int, default=256)
     p.add_argument("--lda-docs-per-batch", type=int, default=8192)
     p.add_argument("--lda-tokens-per-doc", type=int, default=128)
+    p.add_argument("--lda-sampler", type=str, default="exact",
+                   choices=["exact", "alias"])
     return p.parse_args()
 
 
@@ -78,7 +80,8 @@ def make_jobs(args, world: int):
             app_args={"num_vocabs": args.lda_vocab,
                       "num_topics": args.lda_topics,
                       "tokens_per_doc": args.lda_tokens_per_doc,
-                      "docs_per_batch": args.lda_docs_per_batch})
+                      "docs_per_batch": args.lda_docs_per_batch,
+                      "sampler": args.lda_sampler})
     return jobs
 
 
