@@ -191,10 +191,11 @@ class LDATrainer(Trainer):
         self._step += 1
         dt = self.doc_topic[b.doc_ids]          # gather copy
         if self.a["sampler"] == "alias":
-            prob, alias, _, invden = self._alias
-            ops.lda_mh(dt, self.word_topic, invden, prob, alias,
-                       b.doc_offsets, b.word_local, z, self.a["alpha"],
-                       self.a["beta"], self._epoch_seed + self._step)
+            prob, alias, tprob, talias, _, invden = self._alias
+            ops.lda_mh(dt, self.word_topic, invden, prob, alias, tprob,
+                       talias, b.doc_offsets, b.word_local, z,
+                       self.a["alpha"], self.a["beta"],
+                       self._epoch_seed + self._step)
         else:
             ops.lda_gibbs(dt, self.word_topic,
                           self.topic_sum, b.doc_offsets, b.word_local, z,

@@ -235,65 +235,108 @@ def lda_apply_pairs(shard: torch.Tensor, rows: torch.Tensor,
     flat.scatter_add_(0, rows * K + new_t.long(), ones)
 
 
+def _butterfly_sum64(x: torch.Tensor) -> torch.Tensor:
+    """Bitwise replica of the wave64 shfl_xor butterfly sum over dim -1."""
+    idx = torch.arange(64)
+    for m in (32, 16, 8, 4, 2, 1):
+        x = x + x[..., idx ^ m]
+    return x[..., 0]
+
+
+def _vose_serial(pr, al, lk, n):
+    """Python replica of the device vose_serial (same order)."""
+    small_top, large_top = -1, -1
+    for k in range(n - 1, -1, -1):
+        if float(pr[k]) < 1.0:
+            lk[k] = small_top
+            small_top = k
+        else:
+            lk[k] = large_top
+            large_top = k
+    while small_top >= 0 and large_top >= 0:
+        sm = small_top
+        small_top = lk[sm]
+        lg = large_top
+        al[sm] = lg
+        rem = float(pr[lg]) + float(pr[sm]) - 1.0
+        pr[lg] = rem
+        large_top = lk[lg]
+        if rem < 1.0:
+            lk[lg] = small_top
+            small_top = lg
+        else:
+            lk[lg] = large_top
+            large_top = lg
+    while large_top >= 0:
+        lg = large_top
+        large_top = lk[lg]
+        pr[lg] = 1.0
+        al[lg] = lg
+    while small_top >= 0:
+        sm = small_top
+        small_top = lk[sm]
+        pr[sm] = 1.0
+        al[sm] = sm
+
+
 def lda_alias_build(word_topic: torch.Tensor, topic_sum: torch.Tensor,
                     beta: float, num_vocabs: int):
-    """Per-word Vose alias tables over the (batch-stale) word factor
-    q_w(k) = (n_wk + b)/(n_k + V b). Returns (prob, alias, qsum, invden).
-    The construction order is deterministic and identical on CPU and GPU so
-    MH sweeps can be compared sample-by-sample."""
+    """Two-level per-word alias tables over the (batch-stale) word factor
+    q_w(k) = (n_wk + b)/(n_k + V b): a 64-entry top alias over lane-segment
+    masses + per-segment aliases (exact — see ops/csrc/lda_alias.hip).
+    Returns (prob, alias, top_prob, top_alias, qsum, invden). Construction
+    is deterministic and float-order-matched to the device kernel
+    (butterfly total, ascending serial segment sums) so CPU and GPU build
+    bit-identical tables."""
     if _use_hip(word_topic):
         return tuple(_hip.lda_alias_build(word_topic.contiguous(),
                                           topic_sum.contiguous(),
                                           float(beta), int(num_vocabs)))
     rows, K = word_topic.shape
+    W = 64
+    assert K % W == 0, "alias sampler requires K % 64 == 0"
+    S = K // W
     invden = (1.0 / (topic_sum.float() + num_vocabs * beta)).contiguous()
-    p0 = (word_topic.float() + beta) * invden
-    qsum = p0.sum(dim=1)
-    prob = p0 * (K / qsum).unsqueeze(1)
+    p = (word_topic.float() + beta) * invden            # [rows, K]
+    pseg = p.view(rows, W, S)
+    seg_mass = torch.zeros(rows, W)
+    for i in range(S):                                   # ascending = kernel
+        seg_mass = seg_mass + pseg[:, :, i]
+    total = _butterfly_sum64(seg_mass)
+    sscale = torch.where(seg_mass > 0, S / seg_mass,
+                         torch.zeros_like(seg_mass))
+    prob = (pseg * sscale.unsqueeze(2)).reshape(rows, K).contiguous()
     alias = torch.empty((rows, K), dtype=torch.int32)
-    probo = prob.clone()
-    for w in range(rows):
-        pr = probo[w]
-        small = [k for k in range(K) if float(pr[k]) < 1.0]
-        large = [k for k in range(K) if float(pr[k]) >= 1.0]
-        small.reverse(); large.reverse()   # match kernel's pop order
-        al = alias[w]
-        while small and large:
-            sm = small.pop()
-            lg = large[-1]
-            al[sm] = lg
-            rem = float(pr[lg]) + float(pr[sm]) - 1.0
-            pr[lg] = rem
-            large.pop()
-            if rem < 1.0:
-                small.append(lg)
-            else:
-                large.append(lg)
-        for lg in reversed(large):
-            pr[lg] = 1.0
-            al[lg] = lg
-        for sm in reversed(small):
-            pr[sm] = 1.0
-            al[sm] = sm
-    return probo, alias, qsum, invden
+    top_prob = (seg_mass * W / total.unsqueeze(1)).contiguous()
+    top_alias = torch.empty((rows, W), dtype=torch.int32)
+    lk = [0] * max(K, W)
+    for r in range(rows):
+        for g in range(W):
+            _vose_serial(prob[r, g * S:(g + 1) * S],
+                         alias[r, g * S:(g + 1) * S], lk, S)
+        _vose_serial(top_prob[r], top_alias[r], lk, W)
+    return prob, alias, top_prob, top_alias, total, invden
 
 
 def lda_mh(doc_topic: torch.Tensor, word_topic: torch.Tensor,
            invden: torch.Tensor, prob: torch.Tensor, alias: torch.Tensor,
+           top_prob: torch.Tensor, top_alias: torch.Tensor,
            doc_offsets: torch.Tensor, word_ids: torch.Tensor,
            assignments: torch.Tensor, alpha: float, beta: float,
            seed: int) -> torch.Tensor:
-    """One Metropolis-Hastings alias sweep (K7b; see ops/csrc/lda_alias.hip
-    for the proposal/acceptance derivation). Same stationary distribution as
+    """One Metropolis-Hastings alias sweep (K7b; proposal/acceptance
+    derivation in ops/csrc/lda_alias.hip). Same stationary distribution as
     the exact sampler under the batch-stale word-topic snapshot; O(1) per
     token. Torch path mirrors the kernel's RNG and float math."""
     if _use_hip(word_topic):
         return _hip.lda_mh(doc_topic, word_topic, invden, prob, alias,
-                           doc_offsets, word_ids, assignments, float(alpha),
-                           float(beta), int(seed))
+                           top_prob, top_alias, doc_offsets, word_ids,
+                           assignments, float(alpha), float(beta), int(seed))
     from harmony_amd.ops.rng import rng_uniform
 
     K = word_topic.shape[1]
+    W = 64
+    S = K // W
     lengths = doc_offsets[1:] - doc_offsets[:-1]
     n_docs = doc_topic.shape[0]
     max_len = int(lengths.max()) if n_docs else 0
@@ -308,11 +351,16 @@ def lda_mh(doc_topic: torch.Tensor, word_topic: torch.Tensor,
         dt[ar, s] -= 1
         c0 = (tok * 8).long()
         sd = seed & 0xFFFFFFFF
-        # word proposal
-        u0 = rng_uniform(sd, c0) * K
-        bins = u0.long().clamp_(max=K - 1)
-        frac = u0 - bins.float()
-        t1 = torch.where(frac < prob[w, bins], bins, alias[w, bins].long())
+        # word proposal: two-level alias draw
+        u1 = rng_uniform(sd, c0) * W
+        gb = u1.long().clamp_(max=W - 1)
+        g = torch.where(u1 - gb.float() < top_prob[w, gb], gb,
+                        top_alias[w, gb].long())
+        u2 = rng_uniform(sd, c0 + 6) * S
+        eb = u2.long().clamp_(max=S - 1)
+        ecol = g * S + eb
+        t1 = g * S + torch.where(u2 - eb.float() < prob[w, ecol], eb,
+                                 alias[w, ecol].long())
         a1 = (dt[ar, t1].float() + alpha) / (dt[ar, s].float() + alpha)
         s = torch.where(rng_uniform(sd, c0 + 1) < a1, t1, s)
         # doc proposal

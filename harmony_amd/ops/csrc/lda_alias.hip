@@ -5,9 +5,9 @@
 // a 2-proposal MH step per token has the SAME stationary distribution at
 // O(1) per token:
 //   word proposal  t1 ~ q_w(k) ∝ (n_wk + b)/(n_k + Vb)  via a per-word
-//                  alias table built once per pull; because q_w equals the
-//                  word factor of the (stale) posterior exactly, the
-//                  acceptance ratio collapses to the doc factor
+//                  TWO-LEVEL alias table built once per pull; because q_w
+//                  equals the word factor of the (stale) posterior exactly,
+//                  the acceptance ratio collapses to the doc factor
 //                  (n_d,t1 + a)/(n_d,s + a).
 //   doc proposal   t2 ~ q_d(k) ∝ n~_dk + a  (n~ includes the current token)
 //                  drawn in O(1): with prob aK/(aK+L_d) a uniform topic,
@@ -15,11 +15,20 @@
 //                  acceptance carries the full ratio with the doc-proposal
 //                  correction.
 //
-// Execution shape: ONE THREAD PER DOCUMENT (the per-token work is scalar),
-// the thread's doc-topic counts live as a u8 LDS row (counts <= doc length
-// <= 255), tokens walk serially per doc, thousands of docs in flight.
-// RNG: counter hash (hip_common.h) at ctr = token*8 + draw — mirrored by
-// the torch reference for sample-exact CPU/GPU tests.
+// Two-level alias (exact): the row's K probabilities split into 64 lane
+// segments; a 64-entry TOP alias selects the segment (mass-proportional),
+// a per-segment alias selects the entry. Sampling is exact:
+// P(k) = mass(g)/sum * p(k)/mass(g) = p(k)/sum. Construction is the win:
+// all 64 segment tables build CONCURRENTLY (one lane each, K/64 serial
+// steps) and only the 64-entry top table is a serial chain — the flat
+// one-wave Vose was a K-long dependent-LDS chain (measured 1.03 ms per
+// 46k x 256 build; thread-per-row before that: 2.26 ms).
+//
+// Sampler shape: ONE THREAD PER DOCUMENT (per-token work is scalar), the
+// doc-topic counts as a u8 LDS row (counts <= doc length <= 255), tokens
+// serial per doc, thousands of docs in flight. RNG: counter hash
+// (hip_common.h) at ctr = token*8 + draw — mirrored by the torch reference
+// for sample-exact CPU/GPU tests.
 
 #include "hip_common.h"
 #include <algorithm>
@@ -27,77 +36,86 @@
 namespace {
 
 constexpr int ALIAS_THREADS = 128;
-
-// Vose alias construction, one WAVE per word row: the row is staged
-// through LDS cooperatively (coalesced), lane 0 runs the serial two-queue
-// pairing on LDS (deterministic order, matching the torch reference), and
-// the wave writes prob/alias back coalesced. The first thread-per-row
-// version was 2.26 ms per build (uncoalesced serial global walks); this
-// shape is LDS-latency bound instead.
 constexpr int BUILD_WAVES = 4;
+
+// In-LDS serial Vose over n entries at pr[0..n), links lk, output al.
+// Deterministic order (descending init so pops ascend) — the torch
+// reference replicates it exactly.
+__device__ void vose_serial(float* pr, int* al, int* lk, int n) {
+  int small_top = -1, large_top = -1;
+  for (int k = n - 1; k >= 0; --k) {
+    if (pr[k] < 1.f) { lk[k] = small_top; small_top = k; }
+    else             { lk[k] = large_top; large_top = k; }
+  }
+  while (small_top >= 0 && large_top >= 0) {
+    const int sm = small_top; small_top = lk[sm];
+    const int lg = large_top;
+    al[sm] = lg;
+    const float rem = (pr[lg] + pr[sm]) - 1.f;
+    pr[lg] = rem;
+    large_top = lk[lg];
+    if (rem < 1.f) { lk[lg] = small_top; small_top = lg; }
+    else           { lk[lg] = large_top; large_top = lg; }
+  }
+  while (large_top >= 0) { const int lg = large_top; large_top = lk[lg];
+                           pr[lg] = 1.f; al[lg] = lg; }
+  while (small_top >= 0) { const int sm = small_top; small_top = lk[sm];
+                           pr[sm] = 1.f; al[sm] = sm; }
+}
 
 __global__ void alias_build_kernel(const int* __restrict__ word_topic,
                                    const float* __restrict__ invden,
                                    float beta,
                                    float* __restrict__ prob,     // [rows][K]
                                    int* __restrict__ alias,      // [rows][K]
+                                   float* __restrict__ top_prob, // [rows][64]
+                                   int* __restrict__ top_alias,  // [rows][64]
                                    float* __restrict__ qsum,     // [rows]
                                    int rows, int K) {
   extern __shared__ float smem_f[];
-  // layout per wave: pr[K] f32 | al[K] i32 | lk[K] i32
+  // per wave: pr[K] f32 | al[K] i32 | lk[K] i32 | tp[64] f32 | ta[64] i32 | tl[64] i32
   const int wave = threadIdx.x / WAVE;
   const int lane = threadIdx.x % WAVE;
-  float* pr = smem_f + (size_t)wave * 3 * K;
+  const size_t per_wave = 3 * (size_t)K + 3 * WAVE;
+  float* pr = smem_f + (size_t)wave * per_wave;
   int* al = (int*)(pr + K);
   int* lk = al + K;
-  const int nchunk = (K + WAVE - 1) / WAVE;
+  float* tp = (float*)(lk + K);
+  int* ta = (int*)(tp + WAVE);
+  int* tl = ta + WAVE;
+  const int S = K / WAVE;                  // entries per lane segment
   for (int row = blockIdx.x * BUILD_WAVES + wave; row < rows;
        row += gridDim.x * BUILD_WAVES) {
     const int64_t base = (int64_t)row * K;
-    float part = 0.f;
-    for (int c = 0; c < nchunk; ++c) {
-      const int k = c * WAVE + lane;
-      if (k < K) {
-        const float p = ((float)word_topic[base + k] + beta) * invden[k];
-        pr[k] = p;
-        part += p;
-      }
+    // lane's contiguous segment [lane*S, (lane+1)*S)
+    float seg_mass = 0.f;
+    for (int i = 0; i < S; ++i) {
+      const int k = lane * S + i;
+      const float p = ((float)word_topic[base + k] + beta) * invden[k];
+      pr[k] = p;
+      seg_mass += p;
     }
-    const float s = wave_reduce_sum(part);
-    const float scale = (float)K / s;
-    for (int c = 0; c < nchunk; ++c) {
-      const int k = c * WAVE + lane;
-      if (k < K) pr[k] *= scale;
-    }
+    const float total = wave_reduce_sum(seg_mass);
+    // per-segment alias: normalize within segment to mean 1 (concurrent
+    // across all 64 lanes; each runs a tiny S-entry serial Vose)
+    const float sscale = (seg_mass > 0.f) ? (float)S / seg_mass : 0.f;
+    for (int i = 0; i < S; ++i) pr[lane * S + i] *= sscale;
+    vose_serial(pr + lane * S, al + lane * S, lk + lane * S, S);
+    // top alias over segment masses (normalized to mean 1 across 64)
+    tp[lane] = seg_mass * (float)WAVE / total;
     if (lane == 0) {
-      qsum[row] = s;
-      int small_top = -1, large_top = -1;
-      for (int k = K - 1; k >= 0; --k) {
-        if (pr[k] < 1.f) { lk[k] = small_top; small_top = k; }
-        else             { lk[k] = large_top; large_top = k; }
-      }
-      while (small_top >= 0 && large_top >= 0) {
-        const int sm = small_top; small_top = lk[sm];
-        const int lg = large_top;
-        al[sm] = lg;
-        const float rem = (pr[lg] + pr[sm]) - 1.f;
-        pr[lg] = rem;
-        large_top = lk[lg];
-        if (rem < 1.f) { lk[lg] = small_top; small_top = lg; }
-        else           { lk[lg] = large_top; large_top = lg; }
-      }
-      while (large_top >= 0) { const int lg = large_top; large_top = lk[lg];
-                               pr[lg] = 1.f; al[lg] = lg; }
-      while (small_top >= 0) { const int sm = small_top; small_top = lk[sm];
-                               pr[sm] = 1.f; al[sm] = sm; }
+      vose_serial(tp, ta, tl, WAVE);
+      qsum[row] = total;
     }
+    // coalesced write-back
+    const int nchunk = K / WAVE;
     for (int c = 0; c < nchunk; ++c) {
       const int k = c * WAVE + lane;
-      if (k < K) {
-        prob[base + k] = pr[k];
-        alias[base + k] = al[k];
-      }
+      prob[base + k] = pr[k];
+      alias[base + k] = al[k];
     }
+    top_prob[(int64_t)row * WAVE + lane] = tp[lane];
+    top_alias[(int64_t)row * WAVE + lane] = ta[lane];
   }
 }
 
@@ -105,8 +123,10 @@ __global__ __launch_bounds__(ALIAS_THREADS)
 void lda_mh_kernel(int* __restrict__ doc_topic,        // [D][K] int32
                    const int* __restrict__ word_topic, // [rows][K] (stale)
                    const float* __restrict__ invden,   // [K]
-                   const float* __restrict__ prob,     // alias prob
-                   const int* __restrict__ alias,      // alias index
+                   const float* __restrict__ prob,     // entry alias prob
+                   const int* __restrict__ alias,      // entry alias index
+                   const float* __restrict__ top_prob, // [rows][64]
+                   const int* __restrict__ top_alias,  // [rows][64]
                    const int64_t* __restrict__ doc_offsets,
                    const int64_t* __restrict__ word_ids,
                    int* __restrict__ z,
@@ -123,19 +143,27 @@ void lda_mh_kernel(int* __restrict__ doc_topic,        // [D][K] int32
   const float Ld = (float)(p1 - p0);
   const float aK = alpha * (float)K;
   const float p_uniform = aK / (aK + Ld);
+  const int S = K / WAVE;
   for (int64_t p = p0; p < p1; ++p) {
     const int64_t w = word_ids[p];
     const int64_t wbase = w * K;
     int s = z[p];
     nd[s] -= 1;                                       // exclude the token
     const unsigned int c0 = (unsigned int)(p * 8);
-    // ---- word proposal (alias): acceptance = doc factor only -------
+    // ---- word proposal (two-level alias): acceptance = doc factor ---
     {
-      const float u = rng_uniform(seed, c0 + 0) * (float)K;
-      int bin = (int)u;
-      if (bin >= K) bin = K - 1;
-      const float frac = u - (float)bin;
-      const int t1 = (frac < prob[wbase + bin]) ? bin : alias[wbase + bin];
+      const float u1 = rng_uniform(seed, c0 + 0) * (float)WAVE;
+      int gb = (int)u1;
+      if (gb >= WAVE) gb = WAVE - 1;
+      const int64_t tbase = w * WAVE;
+      const int g = (u1 - (float)gb < top_prob[tbase + gb])
+                        ? gb : top_alias[tbase + gb];
+      const float u2 = rng_uniform(seed, c0 + 6) * (float)S;
+      int eb = (int)u2;
+      if (eb >= S) eb = S - 1;
+      const int64_t ebase = wbase + g * S;
+      const int t1 = g * S + ((u2 - (float)eb < prob[ebase + eb])
+                                  ? eb : alias[ebase + eb]);
       const float a1 = ((float)nd[t1] + alpha) / ((float)nd[s] + alpha);
       if (rng_uniform(seed, c0 + 1) < a1) s = t1;
     }
@@ -152,7 +180,8 @@ void lda_mh_kernel(int* __restrict__ doc_topic,        // [D][K] int32
       }
       const float nds = (float)nd[s], ndt = (float)nd[t2];
       // n~ (proposal counts) include the current assignment s
-      const float qs = nds + 1.f + alpha, qt = ndt + (t2 == s ? 1.f : 0.f) + alpha;
+      const float qs = nds + 1.f + alpha;
+      const float qt = ndt + (t2 == s ? 1.f : 0.f) + alpha;
       const float pis = (nds + alpha) *
           ((float)word_topic[wbase + s] + beta) * invden[s];
       const float pit = (ndt + alpha) *
@@ -175,34 +204,40 @@ std::vector<torch::Tensor> lda_alias_build(torch::Tensor word_topic,
   CHECK_IN(word_topic); CHECK_IN(topic_sum);
   TORCH_CHECK(word_topic.dtype() == torch::kInt32);
   const int rows = word_topic.size(0), K = word_topic.size(1);
+  TORCH_CHECK(K % WAVE == 0, "alias sampler requires K % 64 == 0");
   auto invden = 1.0 / (topic_sum.to(torch::kFloat32)
                        + (double)num_vocabs * beta);
   invden = invden.contiguous();
-  auto prob = torch::empty({rows, K}, word_topic.options()
-                                          .dtype(torch::kFloat32));
+  auto prob = torch::empty({rows, K},
+                           word_topic.options().dtype(torch::kFloat32));
   auto alias = torch::empty({rows, K}, word_topic.options());
+  auto top_prob = torch::empty({rows, WAVE},
+                               word_topic.options().dtype(torch::kFloat32));
+  auto top_alias = torch::empty({rows, WAVE}, word_topic.options());
   auto qsum = torch::empty({rows}, prob.options());
   if (rows > 0) {
     dim3 blk(WAVE * BUILD_WAVES);
-    dim3 grid(std::min((rows + BUILD_WAVES - 1) / BUILD_WAVES, 2048));
-    const size_t shmem = (size_t)BUILD_WAVES * 3 * K * 4;
+    dim3 grid(std::min((rows + BUILD_WAVES - 1) / BUILD_WAVES, 8192));
+    const size_t shmem = (size_t)BUILD_WAVES * (3 * K + 3 * WAVE) * 4;
     hipLaunchKernelGGL(alias_build_kernel, grid, blk, shmem, current_stream(),
                        word_topic.data_ptr<int>(), invden.data_ptr<float>(),
                        (float)beta, prob.data_ptr<float>(),
-                       alias.data_ptr<int>(), qsum.data_ptr<float>(),
+                       alias.data_ptr<int>(), top_prob.data_ptr<float>(),
+                       top_alias.data_ptr<int>(), qsum.data_ptr<float>(),
                        rows, K);
   }
-  return {prob, alias, qsum, invden};
+  return {prob, alias, top_prob, top_alias, qsum, invden};
 }
 
 torch::Tensor lda_mh(torch::Tensor doc_topic, torch::Tensor word_topic,
                      torch::Tensor invden, torch::Tensor prob,
-                     torch::Tensor alias, torch::Tensor doc_offsets,
+                     torch::Tensor alias, torch::Tensor top_prob,
+                     torch::Tensor top_alias, torch::Tensor doc_offsets,
                      torch::Tensor word_ids, torch::Tensor assignments,
                      double alpha, double beta, int64_t seed) {
   CHECK_IN(doc_topic); CHECK_IN(word_topic); CHECK_IN(invden);
-  CHECK_IN(prob); CHECK_IN(alias); CHECK_IN(doc_offsets);
-  CHECK_IN(word_ids); CHECK_IN(assignments);
+  CHECK_IN(prob); CHECK_IN(alias); CHECK_IN(top_prob); CHECK_IN(top_alias);
+  CHECK_IN(doc_offsets); CHECK_IN(word_ids); CHECK_IN(assignments);
   const int D = doc_topic.size(0), K = doc_topic.size(1);
   if (D == 0) return assignments;
   dim3 blk(ALIAS_THREADS), grid((D + ALIAS_THREADS - 1) / ALIAS_THREADS);
@@ -211,7 +246,9 @@ torch::Tensor lda_mh(torch::Tensor doc_topic, torch::Tensor word_topic,
   hipLaunchKernelGGL(lda_mh_kernel, grid, blk, shmem, current_stream(),
                      doc_topic.data_ptr<int>(), word_topic.data_ptr<int>(),
                      invden.data_ptr<float>(), prob.data_ptr<float>(),
-                     alias.data_ptr<int>(), doc_offsets.data_ptr<int64_t>(),
+                     alias.data_ptr<int>(), top_prob.data_ptr<float>(),
+                     top_alias.data_ptr<int>(),
+                     doc_offsets.data_ptr<int64_t>(),
                      word_ids.data_ptr<int64_t>(),
                      assignments.data_ptr<int>(),
                      (float)alpha, (float)beta, D, K,

@@ -103,29 +103,40 @@ def test_rng_matches_scalar_reference():
 
 def test_lda_alias_build_valid():
     torch.manual_seed(3)
-    V, K = 40, 12
+    V, K = 40, 128          # K % 64 == 0 (two-level alias)
+    W, S = 64, K // 64
     wt = torch.randint(0, 30, (V, K), dtype=torch.int32)
     ts = wt.sum(0).to(torch.int32)
-    prob, alias, qsum, invden = ops.lda_alias_build(wt, ts, 0.01, V)
+    prob, alias, tprob, talias, qsum, invden = ops.lda_alias_build(
+        wt, ts, 0.01, V)
     assert prob.shape == (V, K) and alias.shape == (V, K)
+    assert tprob.shape == (V, W) and talias.shape == (V, W)
     assert (prob >= 0).all() and (prob <= 1.0 + 1e-5).all()
-    assert int(alias.min()) >= 0 and int(alias.max()) < K
-    # alias sampling reproduces q_w: estimate via exhaustive bins
-    # P(k) = sum_bins [prob[b]*1{b==k} + (1-prob[b])*1{alias[b]==k}] / K
+    # exhaustive two-level enumeration reproduces q_w exactly:
+    # P(k) = P(seg g) * P(entry k | g)
     for w in range(0, V, 7):
-        est = torch.zeros(K)
-        for b in range(K):
-            est[b] += float(prob[w, b])
-            est[int(alias[w, b])] += 1.0 - float(prob[w, b])
-        est /= K
         q = (wt[w].float() + 0.01) * invden
-        q = q / q.sum()
-        assert torch.allclose(est, q, atol=1e-4), w
+        qn = q / q.sum()
+        # segment distribution from the top alias
+        pseg = torch.zeros(W)
+        for b in range(W):
+            pseg[b] += float(tprob[w, b])
+            pseg[int(talias[w, b])] += 1.0 - float(tprob[w, b])
+        pseg /= W
+        est = torch.zeros(K)
+        for g in range(W):
+            pe = torch.zeros(S)
+            for b in range(S):
+                pe[b] += float(prob[w, g * S + b])
+                pe[int(alias[w, g * S + b])] += 1.0 - float(prob[w, g * S + b])
+            pe /= S
+            est[g * S:(g + 1) * S] = pseg[g] * pe
+        assert torch.allclose(est, qn, atol=1e-4), w
 
 
 def test_lda_mh_invariants_and_mixing():
     torch.manual_seed(4)
-    D, K, V, T = 32, 12, 60, 24
+    D, K, V, T = 32, 64, 60, 24
     word_ids = torch.randint(0, V, (D * T,))
     z = torch.randint(0, K, (D * T,), dtype=torch.int32)
     offsets = torch.arange(0, (D + 1) * T, T)
@@ -136,10 +147,11 @@ def test_lda_mh_invariants_and_mixing():
     wt.view(-1).scatter_add_(0, word_ids * K + z.long(),
                              torch.ones(D * T, dtype=torch.int32))
     ts = wt.sum(0).to(torch.int32)
-    prob, alias, _, invden = ops.lda_alias_build(wt, ts, 0.01, V)
+    prob, alias, tprob, talias, _, invden = ops.lda_alias_build(
+        wt, ts, 0.01, V)
     z2 = z.clone()
-    ops.lda_mh(dt, wt, invden, prob, alias, offsets, word_ids, z2,
-               0.1, 0.01, seed=99)
+    ops.lda_mh(dt, wt, invden, prob, alias, tprob, talias, offsets,
+               word_ids, z2, 0.1, 0.01, seed=99)
     assert (dt.sum(1) == T).all()          # token conservation per doc
     assert int(z2.min()) >= 0 and int(z2.max()) < K
     assert not torch.equal(z, z2)          # chain moved
