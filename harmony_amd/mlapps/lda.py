@@ -39,7 +39,8 @@ MODEL_TABLE = "lda_model"
 def defaults(job: JobConfig) -> dict:
     a = dict(num_docs=32768, num_vocabs=30000, num_topics=256,
              tokens_per_doc=64, alpha=0.1, beta=0.01, docs_per_batch=4096,
-             sampler="exact")   # "exact" (dense Gibbs) | "alias" (MH, K7b)
+             sampler="exact",   # "exact" (dense Gibbs) | "alias" (MH, K7b)
+             alias_refresh=4)   # rebuild alias tables every N pulls
     a.update(job.app_args)
     return a
 
@@ -179,10 +180,23 @@ class LDATrainer(Trainer):
         self.word_topic = pulled[:-1]          # [n_uniq_words, K]
         self.topic_sum = pulled[-1]            # [K]
         if self.a["sampler"] == "alias":
-            # per-word Vose alias tables over the stale word factor (K7b)
-            self._alias = ops.lda_alias_build(
-                self.word_topic, self.topic_sum, self.a["beta"],
-                self.a["num_vocabs"])
+            # two-level alias tables over the word factor (K7b), rebuilt
+            # every alias_refresh uses OF THIS BLOCK (tables index the
+            # block's local word ids) — the MH acceptance corrects for
+            # table staleness with the stored proposal density qv
+            if not hasattr(self, "_alias_cache"):
+                self._alias_cache = {}
+                self._alias_age = {}
+            bid = self._block_idx
+            refresh = max(1, int(self.a["alias_refresh"]))
+            age = self._alias_age.get(bid, refresh)
+            if age >= refresh:
+                self._alias_cache[bid] = ops.lda_alias_build(
+                    self.word_topic, self.topic_sum, self.a["beta"],
+                    self.a["num_vocabs"])
+                age = 0
+            self._alias_age[bid] = age + 1
+            self._alias = self._alias_cache[bid]
 
     def local_compute(self) -> None:
         b = self.batch
@@ -191,9 +205,12 @@ class LDATrainer(Trainer):
         self._step += 1
         dt = self.doc_topic[b.doc_ids]          # gather copy
         if self.a["sampler"] == "alias":
-            prob, alias, tprob, talias, _, invden = self._alias
+            prob, alias, tprob, talias, qv, _, invden = self._alias
+            # NOTE: invden/qv index LOCAL word ids of the batch on which the
+            # tables were built; with one static key set per block this is
+            # consistent across refreshes of the same block
             ops.lda_mh(dt, self.word_topic, invden, prob, alias, tprob,
-                       talias, b.doc_offsets, b.word_local, z,
+                       talias, qv, b.doc_offsets, b.word_local, z,
                        self.a["alpha"], self.a["beta"],
                        self._epoch_seed + self._step)
         else:

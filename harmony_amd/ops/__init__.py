@@ -303,6 +303,7 @@ def lda_alias_build(word_topic: torch.Tensor, topic_sum: torch.Tensor,
     for i in range(S):                                   # ascending = kernel
         seg_mass = seg_mass + pseg[:, :, i]
     total = _butterfly_sum64(seg_mass)
+    qv = (p * (1.0 / total).unsqueeze(1)).contiguous()   # encoded density
     sscale = torch.where(seg_mass > 0, S / seg_mass,
                          torch.zeros_like(seg_mass))
     prob = (pseg * sscale.unsqueeze(2)).reshape(rows, K).contiguous()
@@ -315,22 +316,22 @@ def lda_alias_build(word_topic: torch.Tensor, topic_sum: torch.Tensor,
             _vose_serial(prob[r, g * S:(g + 1) * S],
                          alias[r, g * S:(g + 1) * S], lk, S)
         _vose_serial(top_prob[r], top_alias[r], lk, W)
-    return prob, alias, top_prob, top_alias, total, invden
+    return prob, alias, top_prob, top_alias, qv, total, invden
 
 
 def lda_mh(doc_topic: torch.Tensor, word_topic: torch.Tensor,
            invden: torch.Tensor, prob: torch.Tensor, alias: torch.Tensor,
            top_prob: torch.Tensor, top_alias: torch.Tensor,
-           doc_offsets: torch.Tensor, word_ids: torch.Tensor,
-           assignments: torch.Tensor, alpha: float, beta: float,
-           seed: int) -> torch.Tensor:
+           qv: torch.Tensor, doc_offsets: torch.Tensor,
+           word_ids: torch.Tensor, assignments: torch.Tensor, alpha: float,
+           beta: float, seed: int) -> torch.Tensor:
     """One Metropolis-Hastings alias sweep (K7b; proposal/acceptance
     derivation in ops/csrc/lda_alias.hip). Same stationary distribution as
     the exact sampler under the batch-stale word-topic snapshot; O(1) per
     token. Torch path mirrors the kernel's RNG and float math."""
     if _use_hip(word_topic):
         return _hip.lda_mh(doc_topic, word_topic, invden, prob, alias,
-                           top_prob, top_alias, doc_offsets, word_ids,
+                           top_prob, top_alias, qv, doc_offsets, word_ids,
                            assignments, float(alpha), float(beta), int(seed))
     from harmony_amd.ops.rng import rng_uniform
 
@@ -361,7 +362,13 @@ def lda_mh(doc_topic: torch.Tensor, word_topic: torch.Tensor,
         ecol = g * S + eb
         t1 = g * S + torch.where(u2 - eb.float() < prob[w, ecol], eb,
                                  alias[w, ecol].long())
-        a1 = (dt[ar, t1].float() + alpha) / (dt[ar, s].float() + alpha)
+        # stale-table-safe acceptance: pi from the CURRENT snapshot, q from
+        # the table's encoded density (cancels when the table is fresh)
+        pi_s = (dt[ar, s].float() + alpha) * \
+            (word_topic[w, s].float() + beta) * invden[s]
+        pi_t = (dt[ar, t1].float() + alpha) * \
+            (word_topic[w, t1].float() + beta) * invden[t1]
+        a1 = (pi_t * qv[w, s]) / (pi_s * qv[w, t1])
         s = torch.where(rng_uniform(sd, c0 + 1) < a1, t1, s)
         # doc proposal
         Ld = lengths[active].float()

@@ -70,6 +70,7 @@ __global__ void alias_build_kernel(const int* __restrict__ word_topic,
                                    int* __restrict__ alias,      // [rows][K]
                                    float* __restrict__ top_prob, // [rows][64]
                                    int* __restrict__ top_alias,  // [rows][64]
+                                   float* __restrict__ qv,       // [rows][K]
                                    float* __restrict__ qsum,     // [rows]
                                    int rows, int K) {
   extern __shared__ float smem_f[];
@@ -96,6 +97,11 @@ __global__ void alias_build_kernel(const int* __restrict__ word_topic,
       seg_mass += p;
     }
     const float total = wave_reduce_sum(seg_mass);
+    // proposal density actually encoded by the tables (for stale-table
+    // acceptance correction): qv = p / total
+    const float inv_total = 1.f / total;
+    for (int i = 0; i < S; ++i)
+      qv[base + lane * S + i] = pr[lane * S + i] * inv_total;
     // per-segment alias: normalize within segment to mean 1 (concurrent
     // across all 64 lanes; each runs a tiny S-entry serial Vose)
     const float sscale = (seg_mass > 0.f) ? (float)S / seg_mass : 0.f;
@@ -121,12 +127,13 @@ __global__ void alias_build_kernel(const int* __restrict__ word_topic,
 
 __global__ __launch_bounds__(ALIAS_THREADS)
 void lda_mh_kernel(int* __restrict__ doc_topic,        // [D][K] int32
-                   const int* __restrict__ word_topic, // [rows][K] (stale)
+                   const int* __restrict__ word_topic, // [rows][K] (fresh)
                    const float* __restrict__ invden,   // [K]
                    const float* __restrict__ prob,     // entry alias prob
                    const int* __restrict__ alias,      // entry alias index
                    const float* __restrict__ top_prob, // [rows][64]
                    const int* __restrict__ top_alias,  // [rows][64]
+                   const float* __restrict__ qv,       // proposal density
                    const int64_t* __restrict__ doc_offsets,
                    const int64_t* __restrict__ word_ids,
                    int* __restrict__ z,
@@ -164,7 +171,15 @@ void lda_mh_kernel(int* __restrict__ doc_topic,        // [D][K] int32
       const int64_t ebase = wbase + g * S;
       const int t1 = g * S + ((u2 - (float)eb < prob[ebase + eb])
                                   ? eb : alias[ebase + eb]);
-      const float a1 = ((float)nd[t1] + alpha) / ((float)nd[s] + alpha);
+      // acceptance: pi uses the CURRENT snapshot, q the (possibly older)
+      // table's encoded density — with a fresh table the word factors
+      // cancel mathematically; the explicit form stays correct when the
+      // tables are reused across several pulls (alias_refresh)
+      const float pi_s = ((float)nd[s] + alpha) *
+          ((float)word_topic[wbase + s] + beta) * invden[s];
+      const float pi_t = ((float)nd[t1] + alpha) *
+          ((float)word_topic[wbase + t1] + beta) * invden[t1];
+      const float a1 = (pi_t * qv[wbase + s]) / (pi_s * qv[wbase + t1]);
       if (rng_uniform(seed, c0 + 1) < a1) s = t1;
     }
     // ---- doc proposal: q_d ∝ n~_dk + a (n~ includes current token) --
@@ -214,6 +229,8 @@ std::vector<torch::Tensor> lda_alias_build(torch::Tensor word_topic,
   auto top_prob = torch::empty({rows, WAVE},
                                word_topic.options().dtype(torch::kFloat32));
   auto top_alias = torch::empty({rows, WAVE}, word_topic.options());
+  auto qv = torch::empty({rows, K},
+                         word_topic.options().dtype(torch::kFloat32));
   auto qsum = torch::empty({rows}, prob.options());
   if (rows > 0) {
     dim3 blk(WAVE * BUILD_WAVES);
@@ -223,21 +240,23 @@ std::vector<torch::Tensor> lda_alias_build(torch::Tensor word_topic,
                        word_topic.data_ptr<int>(), invden.data_ptr<float>(),
                        (float)beta, prob.data_ptr<float>(),
                        alias.data_ptr<int>(), top_prob.data_ptr<float>(),
-                       top_alias.data_ptr<int>(), qsum.data_ptr<float>(),
-                       rows, K);
+                       top_alias.data_ptr<int>(), qv.data_ptr<float>(),
+                       qsum.data_ptr<float>(), rows, K);
   }
-  return {prob, alias, top_prob, top_alias, qsum, invden};
+  return {prob, alias, top_prob, top_alias, qv, qsum, invden};
 }
 
 torch::Tensor lda_mh(torch::Tensor doc_topic, torch::Tensor word_topic,
                      torch::Tensor invden, torch::Tensor prob,
                      torch::Tensor alias, torch::Tensor top_prob,
-                     torch::Tensor top_alias, torch::Tensor doc_offsets,
+                     torch::Tensor top_alias, torch::Tensor qv,
+                     torch::Tensor doc_offsets,
                      torch::Tensor word_ids, torch::Tensor assignments,
                      double alpha, double beta, int64_t seed) {
   CHECK_IN(doc_topic); CHECK_IN(word_topic); CHECK_IN(invden);
   CHECK_IN(prob); CHECK_IN(alias); CHECK_IN(top_prob); CHECK_IN(top_alias);
-  CHECK_IN(doc_offsets); CHECK_IN(word_ids); CHECK_IN(assignments);
+  CHECK_IN(qv); CHECK_IN(doc_offsets); CHECK_IN(word_ids);
+  CHECK_IN(assignments);
   const int D = doc_topic.size(0), K = doc_topic.size(1);
   if (D == 0) return assignments;
   dim3 blk(ALIAS_THREADS), grid((D + ALIAS_THREADS - 1) / ALIAS_THREADS);
@@ -247,7 +266,7 @@ torch::Tensor lda_mh(torch::Tensor doc_topic, torch::Tensor word_topic,
                      doc_topic.data_ptr<int>(), word_topic.data_ptr<int>(),
                      invden.data_ptr<float>(), prob.data_ptr<float>(),
                      alias.data_ptr<int>(), top_prob.data_ptr<float>(),
-                     top_alias.data_ptr<int>(),
+                     top_alias.data_ptr<int>(), qv.data_ptr<float>(),
                      doc_offsets.data_ptr<int64_t>(),
                      word_ids.data_ptr<int64_t>(),
                      assignments.data_ptr<int>(),

@@ -32,7 +32,8 @@ std::vector<torch::Tensor> lda_alias_build(torch::Tensor word_topic,
 torch::Tensor lda_mh(torch::Tensor doc_topic, torch::Tensor word_topic,
                      torch::Tensor invden, torch::Tensor prob,
                      torch::Tensor alias, torch::Tensor top_prob,
-                     torch::Tensor top_alias, torch::Tensor doc_offsets,
+                     torch::Tensor top_alias, torch::Tensor qv,
+                     torch::Tensor doc_offsets,
                      torch::Tensor word_ids, torch::Tensor assignments,
                      double alpha, double beta, int64_t seed);
 void scatter_apply(torch::Tensor shard, torch::Tensor rows,

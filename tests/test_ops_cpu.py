@@ -107,7 +107,7 @@ def test_lda_alias_build_valid():
     W, S = 64, K // 64
     wt = torch.randint(0, 30, (V, K), dtype=torch.int32)
     ts = wt.sum(0).to(torch.int32)
-    prob, alias, tprob, talias, qsum, invden = ops.lda_alias_build(
+    prob, alias, tprob, talias, qv, qsum, invden = ops.lda_alias_build(
         wt, ts, 0.01, V)
     assert prob.shape == (V, K) and alias.shape == (V, K)
     assert tprob.shape == (V, W) and talias.shape == (V, W)
@@ -147,10 +147,10 @@ def test_lda_mh_invariants_and_mixing():
     wt.view(-1).scatter_add_(0, word_ids * K + z.long(),
                              torch.ones(D * T, dtype=torch.int32))
     ts = wt.sum(0).to(torch.int32)
-    prob, alias, tprob, talias, _, invden = ops.lda_alias_build(
+    prob, alias, tprob, talias, qv, _, invden = ops.lda_alias_build(
         wt, ts, 0.01, V)
     z2 = z.clone()
-    ops.lda_mh(dt, wt, invden, prob, alias, tprob, talias, offsets,
+    ops.lda_mh(dt, wt, invden, prob, alias, tprob, talias, qv, offsets,
                word_ids, z2, 0.1, 0.01, seed=99)
     assert (dt.sum(1) == T).all()          # token conservation per doc
     assert int(z2.min()) >= 0 and int(z2.max()) < K

@@ -259,9 +259,9 @@ def test_lda_mh_gpu_matches_cpu():
     wt.view(-1).scatter_add_(0, word_ids * K + z0.long(),
                              torch.ones(D * T, dtype=torch.int32))
     ts = wt.sum(0).to(torch.int32)
-    prob_c, alias_c, tp_c, ta_c, _, inv_c = ops.lda_alias_build(
+    prob_c, alias_c, tp_c, ta_c, qv_c, _, inv_c = ops.lda_alias_build(
         wt, ts, 0.01, V)
-    prob_g, alias_g, tp_g, ta_g, _, inv_g = ops.lda_alias_build(
+    prob_g, alias_g, tp_g, ta_g, qv_g, _, inv_g = ops.lda_alias_build(
         wt.cuda(), ts.cuda(), 0.01, V)
     # tables must be bit-identical (order-matched deterministic build)
     assert torch.equal(alias_c, alias_g.cpu())
@@ -269,12 +269,11 @@ def test_lda_mh_gpu_matches_cpu():
     assert torch.allclose(prob_c, prob_g.cpu(), atol=1e-5)
     assert torch.allclose(tp_c, tp_g.cpu(), atol=1e-5)
     z_cpu, dt_cpu = z0.clone(), dt.clone()
-    ops.lda_mh(dt_cpu, wt, inv_c, prob_c, alias_c, tp_c, ta_c, offsets,
-               word_ids, z_cpu, 0.1, 0.01, seed=321)
+    ops.lda_mh(dt_cpu, wt, inv_c, prob_c, alias_c, tp_c, ta_c, qv_c,
+               offsets, word_ids, z_cpu, 0.1, 0.01, seed=321)
     z_gpu, dt_gpu = z0.clone().cuda(), dt.clone().cuda()
-    ops.lda_mh(dt_gpu, wt.cuda(), inv_g, prob_g, alias_g, tp_g.cuda(),
-               ta_g.cuda(), offsets.cuda(),
-               word_ids.cuda(), z_gpu, 0.1, 0.01, seed=321)
+    ops.lda_mh(dt_gpu, wt.cuda(), inv_g, prob_g, alias_g, tp_g, ta_g, qv_g,
+               offsets.cuda(), word_ids.cuda(), z_gpu, 0.1, 0.01, seed=321)
     match = (z_gpu.cpu() == z_cpu).float().mean()
     assert float(match) > 0.99, float(match)
     assert (dt_gpu.cpu().sum(1) == T).all()
