@@ -45,7 +45,7 @@ def parse_args():
     p.add_argument("--mlr-batch", type=int, default=16384)
     p.add_argument("--lda-vocab", type=int, default=100000)
     p.add_argument("--lda-topics", type=int, default=256)
-    p.add_argument("--lda-docs-per-batch", type=int, default=8192)
+    p.add_argument("--lda-docs-per-batch", type=int, default=16384)
     p.add_argument("--lda-tokens-per-doc", type=int, default=128)
     p.add_argument("--lda-sampler", type=str, default="exact",
                    choices=["exact", "alias"])
@@ -56,6 +56,10 @@ def make_jobs(args, world: int):
     from harmony_amd.config import JobConfig
 
     n_blocks = 4  # resident synthetic blocks per rank, cycled
+    # LDA keeps 8 resident blocks: at the default 16384 docs/batch that is
+    # 131k docs per rank -> the BASELINE "LDA 1M-doc/100k-vocab" config at
+    # 8 GPUs (weak scaling)
+    lda_blocks = 8
     jobs = {}
     if "nmf" in args.apps:
         jobs["nmf"] = JobConfig(
@@ -75,8 +79,8 @@ def make_jobs(args, world: int):
                       "batch_size": args.mlr_batch, "step_size": 0.01})
     if "lda" in args.apps:
         jobs["lda"] = JobConfig(
-            job_id="bench_lda", app="lda", num_mini_batches=n_blocks,
-            num_worker_blocks=n_blocks,
+            job_id="bench_lda", app="lda", num_mini_batches=lda_blocks,
+            num_worker_blocks=lda_blocks,
             app_args={"num_vocabs": args.lda_vocab,
                       "num_topics": args.lda_topics,
                       "tokens_per_doc": args.lda_tokens_per_doc,
