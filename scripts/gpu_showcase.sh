@@ -22,7 +22,7 @@ echo
 timeout $T bin/submit_shortest_path.sh -port 7311 -job_id sc_sp -num_vertices 100000 -out_degree 4 --wait | head -c 400
 echo
 # elastic: addvector with the scripted rotating optimizer + validation
-timeout $T python -m harmony_amd.jobserver.client submit -port 7311 -app addvector -job_id sc_elastic -max_num_epochs 4 -num_mini_batches 6 -optimizer "harmony_amd.optimizer.optimizers:_RotFactory" --wait 2>/dev/null | head -c 200
+timeout $T python -m harmony_amd.jobserver.client submit -port 7311 -app addvector -job_id sc_elastic -max_num_epochs 4 -num_mini_batches 6 -optimizer homogeneous -optimizer_period 4 --wait 2>/dev/null | head -c 200
 echo
 bin/stop_jobserver.sh -port 7311 | head -c 100
 echo
