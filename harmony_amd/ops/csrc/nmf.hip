@@ -12,6 +12,7 @@
 // cheap on gfx950. The rank-dot is a 64-lane butterfly reduction.
 
 #include "hip_common.h"
+#include <cstdlib>
 
 namespace {
 
@@ -81,6 +82,14 @@ __global__ void nmf_grad_kernel(const float* __restrict__ L,
 // (NMFTrainer.aggregateGradient:375-406) with a segmented reduction and
 // removes 200M+ HBM/L2 atomics per batch.
 
+template <int G>
+__device__ __forceinline__ float group_sum(float v) {
+#pragma unroll
+  for (int m = G / 2; m > 0; m >>= 1) v += __shfl_xor(v, m, G);
+  return v;
+}
+
+template <int G>
 __global__ void nmf_grad_e_kernel(const float* __restrict__ L,
                                   const float* __restrict__ R,
                                   const int64_t* __restrict__ row_ptr,
@@ -90,14 +99,14 @@ __global__ void nmf_grad_e_kernel(const float* __restrict__ L,
                                   float* __restrict__ e_out,
                                   float* __restrict__ sqerr,
                                   int n_rows, int k, float lam2) {
-  const int lane = threadIdx.x & (WAVE - 1);
-  const int row = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  const int lane = threadIdx.x & (G - 1);
+  const int row = (blockIdx.x * blockDim.x + threadIdx.x) / G;
   if (row >= n_rows) return;
-  const int nchunk = (k + WAVE - 1) / WAVE;
+  const int nchunk = (k + G - 1) / G;
   float l[MAXC], lg[MAXC];
 #pragma unroll
   for (int c = 0; c < MAXC; ++c) {
-    int idx = c * WAVE + lane;
+    int idx = c * G + lane;
     l[c] = (c < nchunk && idx < k) ? L[(int64_t)row * k + idx] : 0.f;
     lg[c] = 0.f;
   }
@@ -111,7 +120,7 @@ __global__ void nmf_grad_e_kernel(const float* __restrict__ L,
     const int64_t j0 = col_idx[p0];
 #pragma unroll
     for (int c = 0; c < MAXC; ++c) {
-      int idx = c * WAVE + lane;
+      int idx = c * G + lane;
       r[c] = (c < nchunk && idx < k) ? R[j0 * k + idx] : 0.f;
     }
   }
@@ -120,14 +129,14 @@ __global__ void nmf_grad_e_kernel(const float* __restrict__ L,
       const int64_t jn = col_idx[p + 1];
 #pragma unroll
       for (int c = 0; c < MAXC; ++c) {
-        int idx = c * WAVE + lane;
+        int idx = c * G + lane;
         rn[c] = (c < nchunk && idx < k) ? R[jn * k + idx] : 0.f;
       }
     }
     float part = 0.f;
 #pragma unroll
     for (int c = 0; c < MAXC; ++c) part += l[c] * r[c];
-    const float e = wave_reduce_sum(part) - vals[p];
+    const float e = group_sum<G>(part) - vals[p];
     const float ge = 2.f * e;
 #pragma unroll
     for (int c = 0; c < MAXC; ++c) lg[c] += ge * r[c] + lam2 * l[c];
@@ -138,12 +147,13 @@ __global__ void nmf_grad_e_kernel(const float* __restrict__ L,
   }
 #pragma unroll
   for (int c = 0; c < MAXC; ++c) {
-    int idx = c * WAVE + lane;
+    int idx = c * G + lane;
     if (c < nchunk && idx < k) lgrad[(int64_t)row * k + idx] = lg[c];
   }
   if (lane == 0 && sq != 0.f) atomicAdd(sqerr, sq);
 }
 
+template <int G>
 __global__ void nmf_rgrad_kernel(const float* __restrict__ L,
                                  const float* __restrict__ R,
                                  const float* __restrict__ e_in,
@@ -152,10 +162,10 @@ __global__ void nmf_rgrad_kernel(const float* __restrict__ L,
                                  const int64_t* __restrict__ row_sorted,
                                  float* __restrict__ rgrad,
                                  int n_cols, int k, float lam2) {
-  const int lane = threadIdx.x & (WAVE - 1);
-  const int col = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  const int lane = threadIdx.x & (G - 1);
+  const int col = (blockIdx.x * blockDim.x + threadIdx.x) / G;
   if (col >= n_cols) return;
-  const int nchunk = (k + WAVE - 1) / WAVE;
+  const int nchunk = (k + G - 1) / G;
   float acc[MAXC];
 #pragma unroll
   for (int c = 0; c < MAXC; ++c) acc[c] = 0.f;
@@ -168,7 +178,7 @@ __global__ void nmf_rgrad_kernel(const float* __restrict__ L,
     const int64_t i0 = row_sorted[p0];
 #pragma unroll
     for (int c = 0; c < MAXC; ++c) {
-      int idx = c * WAVE + lane;
+      int idx = c * G + lane;
       lv[c] = (c < nchunk && idx < k) ? L[i0 * k + idx] : 0.f;
     }
   }
@@ -178,7 +188,7 @@ __global__ void nmf_rgrad_kernel(const float* __restrict__ L,
       const int64_t in_ = row_sorted[p + 1];
 #pragma unroll
       for (int c = 0; c < MAXC; ++c) {
-        int idx = c * WAVE + lane;
+        int idx = c * G + lane;
         lvn[c] = (c < nchunk && idx < k) ? L[in_ * k + idx] : 0.f;
       }
     }
@@ -192,7 +202,7 @@ __global__ void nmf_rgrad_kernel(const float* __restrict__ L,
   const float nl = lam2 * (float)(p1 - p0);
 #pragma unroll
   for (int c = 0; c < MAXC; ++c) {
-    int idx = c * WAVE + lane;
+    int idx = c * G + lane;
     if (c < nchunk && idx < k)
       rgrad[(int64_t)col * k + idx] = acc[c] + nl * R[(int64_t)col * k + idx];
   }
@@ -213,24 +223,52 @@ std::vector<torch::Tensor> nmf_grad_twopass(
   auto rgrad = torch::empty_like(R);
   auto e = torch::empty_like(vals);
   auto sqerr = torch::zeros({}, L.options());
-  const int wpb = 4;
+  // lane-group width: 32 (2 rows/wave) when the rank fits 32*MAXC —
+  // same co-scheduling reasoning as the LDA sampler; HARMONY_NMF_G overrides
+  int G = (k <= 32 * MAXC) ? 32 : 64;
+  const char* env = getenv("HARMONY_NMF_G");
+  if (env && atoi(env) == 64) G = 64;
+  if (env && atoi(env) == 32 && k <= 32 * MAXC) G = 32;
+  const int threads = 256;
   if (n > 0) {
-    hipLaunchKernelGGL(nmf_grad_e_kernel, dim3((n + wpb - 1) / wpb),
-                       dim3(WAVE * wpb), 0, current_stream(),
-                       L.data_ptr<float>(), R.data_ptr<float>(),
-                       row_ptr.data_ptr<int64_t>(), col_idx.data_ptr<int64_t>(),
-                       vals.data_ptr<float>(), lgrad.data_ptr<float>(),
-                       e.data_ptr<float>(), sqerr.data_ptr<float>(),
-                       n, k, 2.f * (float)lam);
+    dim3 grid((n + threads / G - 1) / (threads / G));
+    if (G == 32)
+      hipLaunchKernelGGL(nmf_grad_e_kernel<32>, grid, dim3(threads), 0,
+                         current_stream(),
+                         L.data_ptr<float>(), R.data_ptr<float>(),
+                         row_ptr.data_ptr<int64_t>(),
+                         col_idx.data_ptr<int64_t>(),
+                         vals.data_ptr<float>(), lgrad.data_ptr<float>(),
+                         e.data_ptr<float>(), sqerr.data_ptr<float>(),
+                         n, k, 2.f * (float)lam);
+    else
+      hipLaunchKernelGGL(nmf_grad_e_kernel<64>, grid, dim3(threads), 0,
+                         current_stream(),
+                         L.data_ptr<float>(), R.data_ptr<float>(),
+                         row_ptr.data_ptr<int64_t>(),
+                         col_idx.data_ptr<int64_t>(),
+                         vals.data_ptr<float>(), lgrad.data_ptr<float>(),
+                         e.data_ptr<float>(), sqerr.data_ptr<float>(),
+                         n, k, 2.f * (float)lam);
   }
   if (m > 0) {
-    hipLaunchKernelGGL(nmf_rgrad_kernel, dim3((m + wpb - 1) / wpb),
-                       dim3(WAVE * wpb), 0, current_stream(),
-                       L.data_ptr<float>(), R.data_ptr<float>(),
-                       e.data_ptr<float>(), perm.data_ptr<int64_t>(),
-                       seg_ptr.data_ptr<int64_t>(),
-                       row_sorted.data_ptr<int64_t>(),
-                       rgrad.data_ptr<float>(), m, k, 2.f * (float)lam);
+    dim3 grid((m + threads / G - 1) / (threads / G));
+    if (G == 32)
+      hipLaunchKernelGGL(nmf_rgrad_kernel<32>, grid, dim3(threads), 0,
+                         current_stream(),
+                         L.data_ptr<float>(), R.data_ptr<float>(),
+                         e.data_ptr<float>(), perm.data_ptr<int64_t>(),
+                         seg_ptr.data_ptr<int64_t>(),
+                         row_sorted.data_ptr<int64_t>(),
+                         rgrad.data_ptr<float>(), m, k, 2.f * (float)lam);
+    else
+      hipLaunchKernelGGL(nmf_rgrad_kernel<64>, grid, dim3(threads), 0,
+                         current_stream(),
+                         L.data_ptr<float>(), R.data_ptr<float>(),
+                         e.data_ptr<float>(), perm.data_ptr<int64_t>(),
+                         seg_ptr.data_ptr<int64_t>(),
+                         row_sorted.data_ptr<int64_t>(),
+                         rgrad.data_ptr<float>(), m, k, 2.f * (float)lam);
   }
   return {lgrad, rgrad, sqerr};
 }
