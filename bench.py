@@ -47,7 +47,7 @@ def parse_args():
     p.add_argument("--lda-topics", type=int, default=256)
     p.add_argument("--lda-docs-per-batch", type=int, default=16384)
     p.add_argument("--lda-tokens-per-doc", type=int, default=128)
-    p.add_argument("--lda-sampler", type=str, default="exact",
+    p.add_argument("--lda-sampler", type=str, default="alias",
                    choices=["exact", "alias"])
     return p.parse_args()
 
