@@ -264,6 +264,32 @@ class LDATrainer(Trainer):
                                assume_unique=True)
         self.accessor.metrics["total_push_time_sec"] += time.perf_counter() - t0
 
+    def evaluate_model(self):
+        """Collapsed LDA log-likelihood (reference LDAStatCalculator:
+        log p(w|z) from the word-topic table + log p(z) from the local
+        doc-topic counts; doc part is per-rank, word part global)."""
+        import torch as T
+
+        a = self.a
+        K, V = a["num_topics"], a["num_vocabs"]
+        alpha, beta = float(a["alpha"]), float(a["beta"])
+        full = self.accessor.table.pull_all()[:V + 1].float()
+        wt = full[:V]                              # [V, K]
+        nk = full[V]                               # [K] topic totals
+        lpw = (T.lgamma(wt + beta).sum()
+               - T.lgamma(nk + V * beta).sum()
+               + K * (T.lgamma(T.tensor(V * beta)) -
+                      V * T.lgamma(T.tensor(beta))))
+        dt = self.doc_topic.float()
+        nd = dt.sum(dim=1)
+        lpz = (T.lgamma(dt + alpha).sum()
+               - T.lgamma(nd + K * alpha).sum()
+               + dt.shape[0] * (T.lgamma(T.tensor(K * alpha)) -
+                                K * T.lgamma(T.tensor(alpha))))
+        return {"log_likelihood": float(lpw + lpz),
+                "log_pw_given_z": float(lpw),
+                "log_pz_local": float(lpz)}
+
     def num_batch_examples(self) -> int:
         return self.batch.num_examples
 
