@@ -33,6 +33,7 @@ from harmony_amd.dolphin.data_provider import TrainingDataProvider
 from harmony_amd.dolphin.trainer import Trainer, TrainerContext
 from harmony_amd.et.table import ObjectTable
 from harmony_amd.utils import stable_seed
+from harmony_amd import ops
 
 MODEL_TABLE = "gbt_model"
 
@@ -93,17 +94,8 @@ def build_tree(bins: torch.Tensor, resid: torch.Tensor, num_bins: int,
         first = 2 ** level - 1
         n_nodes = 2 ** level
         # histogram: [n_nodes, F, num_bins] counts and residual sums
-        local = (node - first).unsqueeze(1)                  # [B, 1]
-        idx = (local * F * num_bins + torch.arange(F, device=dev) * num_bins
-               + bins)                                       # [B, F]
-        cnt = torch.zeros(n_nodes * F * num_bins, device=dev)
-        s = torch.zeros(n_nodes * F * num_bins, device=dev)
-        ones = torch.ones(B, 1, device=dev).expand(B, F)
-        cnt.scatter_add_(0, idx.reshape(-1), ones.reshape(-1))
-        s.scatter_add_(0, idx.reshape(-1),
-                       resid.unsqueeze(1).expand(B, F).reshape(-1))
-        cnt = cnt.view(n_nodes, F, num_bins)
-        s = s.view(n_nodes, F, num_bins)
+        # (K10 LDS-privatized kernel on GPU, scatter_add reference on CPU)
+        cnt, s = ops.gbt_hist(bins, resid, node - first, n_nodes, num_bins)
         # prefix sums over bins: left side of split at bin b = bins <= b
         ccum = cnt.cumsum(dim=2)
         scum = s.cumsum(dim=2)

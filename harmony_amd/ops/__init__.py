@@ -422,6 +422,33 @@ def lda_apply_all(shard, word_rows, old_t, new_t, summary_row: int) -> None:
 _APPLY_MODES = {"add": 0, "assign": 1, "nmf_sgd": 2, "lda_counts": 3}
 
 
+def gbt_hist(bins: torch.Tensor, resid: torch.Tensor, node: torch.Tensor,
+             n_nodes: int, num_bins: int
+             ) -> Tuple[torch.Tensor, torch.Tensor]:
+    """GBT level histogram (K10): per-(node, feature, bin) sample counts and
+    residual sums for one tree level. Reference GBTTrainer.java:244+ scans
+    sorted raw values per node/feature on the CPU; the GPU histogram method
+    (pre-quantized bins) is the standard redesign. Returns (cnt, sum),
+    each float32 [n_nodes, F, num_bins]."""
+    B, F = bins.shape
+    if _use_hip(bins) and n_nodes * num_bins <= 8192:
+        cnt = torch.zeros(n_nodes, F, num_bins, device=bins.device)
+        s = torch.zeros_like(cnt)
+        _hip.gbt_hist(bins.int().contiguous(), resid.float().contiguous(),
+                      node.int().contiguous(), cnt, s)
+        return cnt, s
+    idx = (node.long().unsqueeze(1) * F * num_bins
+           + torch.arange(F, device=bins.device) * num_bins + bins)
+    cnt = torch.zeros(n_nodes * F * num_bins, device=bins.device)
+    s = torch.zeros_like(cnt)
+    cnt.scatter_add_(0, idx.reshape(-1),
+                     torch.ones(B, 1, device=bins.device).expand(B, F)
+                     .reshape(-1))
+    s.scatter_add_(0, idx.reshape(-1),
+                   resid.float().unsqueeze(1).expand(B, F).reshape(-1))
+    return (cnt.view(n_nodes, F, num_bins), s.view(n_nodes, F, num_bins))
+
+
 def fused_apply_supported(update_fn_name: str) -> bool:
     return update_fn_name in _APPLY_MODES
 

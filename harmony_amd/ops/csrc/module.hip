@@ -36,6 +36,8 @@ torch::Tensor lda_mh(torch::Tensor doc_topic, torch::Tensor word_topic,
                      torch::Tensor doc_offsets,
                      torch::Tensor word_ids, torch::Tensor assignments,
                      double alpha, double beta, int64_t seed);
+void gbt_hist(torch::Tensor bins, torch::Tensor resid, torch::Tensor node,
+              torch::Tensor cnt, torch::Tensor sum);
 void scatter_apply(torch::Tensor shard, torch::Tensor rows,
                    torch::Tensor deltas, int64_t mode, double step,
                    double maxval);
@@ -64,6 +66,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "per-word Vose alias tables over the stale word factor (K7b)");
   m.def("lda_mh", &lda_mh,
         "Metropolis-Hastings alias LDA sweep, thread-per-doc (K7b)");
+  m.def("gbt_hist", &gbt_hist, "GBT level histogram build (K10)");
   m.def("scatter_apply", &scatter_apply, "owner-side sparse update (K9)");
   m.def("dense_apply", &dense_apply, "owner-side dense update (K3)");
   m.def("parse_nmf_bytes", &parse_nmf_bytes, "native NMF text parser");

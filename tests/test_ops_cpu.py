@@ -209,3 +209,23 @@ def test_lda_samplers_both_recover_structure():
         purities[sampler] = float(purity)
     assert purities["exact"] > 0.9, purities
     assert purities["alias"] > 0.85, purities
+
+
+def test_gbt_hist_ref_shapes_and_totals():
+    import torch
+    from harmony_amd import ops
+
+    g = torch.Generator().manual_seed(3)
+    B, F, nb, n_nodes = 500, 7, 16, 4
+    bins = torch.randint(0, nb, (B, F), generator=g)
+    resid = torch.randn(B, generator=g)
+    node = torch.randint(0, n_nodes, (B,), generator=g)
+    cnt, s = ops.gbt_hist(bins, resid, node, n_nodes, nb)
+    assert cnt.shape == (n_nodes, F, nb) and s.shape == cnt.shape
+    assert int(cnt.sum().item()) == B * F
+    # per-feature sum of residual-sums equals the total residual restricted
+    # to each node's samples
+    for nd in range(n_nodes):
+        want = resid[node == nd].sum()
+        got = s[nd, 0].sum()
+        assert torch.allclose(got, want, atol=1e-4)

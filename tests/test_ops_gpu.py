@@ -292,3 +292,38 @@ def test_lda_alias_job_end_to_end_gpu():
                               "sampler": "alias"})
     m = run_job(job, ctx)
     assert m.summary()["num_batches"] == 4
+
+
+def test_gbt_hist_gpu_vs_ref():
+    # K10 LDS histogram vs the scatter_add reference, both on device
+    g = torch.Generator().manual_seed(7)
+    B, F, nb, n_nodes = 5000, 37, 64, 8
+    bins = torch.randint(0, nb, (B, F), generator=g).to("cuda")
+    resid = torch.randn(B, generator=g).to("cuda")
+    node = torch.randint(0, n_nodes, (B,), generator=g).to("cuda")
+    cnt, s = ops.gbt_hist(bins, resid, node, n_nodes, nb)
+    os.environ["HARMONY_FORCE_TORCH_OPS"] = "1"
+    try:
+        rcnt, rs = ops.gbt_hist(bins, resid, node, n_nodes, nb)
+    finally:
+        del os.environ["HARMONY_FORCE_TORCH_OPS"]
+    assert torch.equal(cnt, rcnt)
+    assert torch.allclose(s, rs, atol=1e-3)
+    assert int(cnt.sum().item()) == B * F
+
+
+def test_gbt_hist_gpu_deep_level_chunks():
+    # n_nodes * nb near the LDS cap -> feature chunking path (fc < F)
+    g = torch.Generator().manual_seed(8)
+    B, F, nb, n_nodes = 3000, 12, 64, 128   # 128*64 = 8192 entries -> fc=1
+    bins = torch.randint(0, nb, (B, F), generator=g).to("cuda")
+    resid = torch.randn(B, generator=g).to("cuda")
+    node = torch.randint(0, n_nodes, (B,), generator=g).to("cuda")
+    cnt, s = ops.gbt_hist(bins, resid, node, n_nodes, nb)
+    os.environ["HARMONY_FORCE_TORCH_OPS"] = "1"
+    try:
+        rcnt, rs = ops.gbt_hist(bins, resid, node, n_nodes, nb)
+    finally:
+        del os.environ["HARMONY_FORCE_TORCH_OPS"]
+    assert torch.equal(cnt, rcnt)
+    assert torch.allclose(s, rs, atol=1e-3)
