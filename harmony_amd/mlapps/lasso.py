@@ -25,6 +25,7 @@ from harmony_amd.dolphin.model_accessor import ETModelAccessor
 from harmony_amd.dolphin.trainer import Trainer, TrainerContext
 from harmony_amd.et.table import Table
 from harmony_amd.utils import stable_seed
+from harmony_amd import ops
 
 MODEL_TABLE = "lasso_model"
 ZERO_THRESHOLD = 1e-9
@@ -93,19 +94,16 @@ class LassoTrainer(Trainer):
     def local_compute(self) -> None:
         X, y = self.batch
         F = X.shape[1]
-        w = self.w.clone()
         lam_n = self.a["lam"] * X.shape[0]
-        r = y - X @ w                          # residual
-        col_sq = (X * X).sum(dim=0).clamp_min(ZERO_THRESHOLD)
-        # per-coordinate closed form, kept tensor-valued (no host syncs: the
-        # sweep queues F small device ops asynchronously)
-        for i in range(F):
-            xi = X[:, i]
-            c = xi @ r + w[i] * col_sq[i]
-            # soft-threshold (reference :164-190)
-            wn = torch.clamp(c.abs() - lam_n, min=0.0) * torch.sign(c) / col_sq[i]
-            r = r + xi * (w[i] - wn)
-            w[i] = wn
+        r = y - X @ self.w                     # residual
+        col_sq = getattr(X, "_harmony_colsq", None)
+        if col_sq is None:
+            col_sq = (X * X).sum(dim=0).clamp_min(ZERO_THRESHOLD)
+            X._harmony_colsq = col_sq          # batches are static per block
+        # full cyclic sweep in one persistent kernel on GPU (K11); the torch
+        # path queues F small device ops asynchronously (no host syncs)
+        w, r = ops.lasso_cd(X, r, self.w, col_sq, lam_n)
+        _ = F
         self.delta = w - self.w
         self._loss = (r * r).mean()          # device scalar; no batch sync
 

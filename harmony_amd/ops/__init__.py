@@ -449,6 +449,34 @@ def gbt_hist(bins: torch.Tensor, resid: torch.Tensor, node: torch.Tensor,
     return (cnt.view(n_nodes, F, num_bins), s.view(n_nodes, F, num_bins))
 
 
+def lasso_cd(X: torch.Tensor, r: torch.Tensor, w: torch.Tensor,
+             col_sq: torch.Tensor, lam_n: float
+             ) -> Tuple[torch.Tensor, torch.Tensor]:
+    """One full cyclic coordinate-descent sweep (K11): for each coordinate
+    the closed-form soft-threshold update (reference LassoTrainer.java:
+    164-190), residual maintained incrementally. Returns (w_new, r_new);
+    inputs are not mutated. GPU: single persistent workgroup, residual in
+    LDS (ops/csrc/lasso.hip); CPU: the tensor-op loop."""
+    F = X.shape[1]
+    w = w.clone()
+    r = r.clone()
+    if _use_hip(X) and X.shape[0] + 17 <= 16384:
+        Xt = getattr(X, "_harmony_xt", None)
+        if Xt is None:
+            Xt = X.t().contiguous()
+            X._harmony_xt = Xt          # batches are static per block
+        _hip.lasso_cd(Xt, r, w, col_sq.contiguous(), float(lam_n))
+        return w, r
+    for i in range(F):
+        xi = X[:, i]
+        c = xi @ r + w[i] * col_sq[i]
+        wn = (torch.clamp(c.abs() - lam_n, min=0.0) * torch.sign(c)
+              / col_sq[i])
+        r = r + xi * (w[i] - wn)
+        w[i] = wn
+    return w, r
+
+
 def fused_apply_supported(update_fn_name: str) -> bool:
     return update_fn_name in _APPLY_MODES
 

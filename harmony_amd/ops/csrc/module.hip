@@ -36,6 +36,8 @@ torch::Tensor lda_mh(torch::Tensor doc_topic, torch::Tensor word_topic,
                      torch::Tensor doc_offsets,
                      torch::Tensor word_ids, torch::Tensor assignments,
                      double alpha, double beta, int64_t seed);
+void lasso_cd(torch::Tensor Xt, torch::Tensor r, torch::Tensor w,
+              torch::Tensor col_sq, double lam_n);
 void gbt_hist(torch::Tensor bins, torch::Tensor resid, torch::Tensor node,
               torch::Tensor cnt, torch::Tensor sum);
 void scatter_apply(torch::Tensor shard, torch::Tensor rows,
@@ -67,6 +69,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("lda_mh", &lda_mh,
         "Metropolis-Hastings alias LDA sweep, thread-per-doc (K7b)");
   m.def("gbt_hist", &gbt_hist, "GBT level histogram build (K10)");
+  m.def("lasso_cd", &lasso_cd, "Lasso persistent CD sweep (K11)");
   m.def("scatter_apply", &scatter_apply, "owner-side sparse update (K9)");
   m.def("dense_apply", &dense_apply, "owner-side dense update (K3)");
   m.def("parse_nmf_bytes", &parse_nmf_bytes, "native NMF text parser");
