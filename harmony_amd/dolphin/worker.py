@@ -53,6 +53,21 @@ class WorkerTasklet:
         self._phase += 1
         return self._phase
 
+    def _consume_shares(self) -> None:
+        """SetBatchShareOp consumption: scale this rank's per-batch work to
+        its share relative to the mean (reference: the plan redistributes
+        training-data blocks; here batches are device-resident so a slow
+        rank serves a prefix slice of each block instead)."""
+        shares = getattr(self.orch.executor, "batch_shares", None)
+        if not shares or self.rank not in shares:
+            return
+        vals = [v for v in shares.values() if v > 0]
+        if not vals:
+            return
+        mean = sum(vals) / len(vals)
+        if mean > 0 and hasattr(self.provider, "set_share"):
+            self.provider.set_share(shares[self.rank] / mean)
+
     def run(self) -> MetricCollector:
         jid = self.job.job_id
         stream_ctx = (torch.cuda.stream(self.stream) if self.stream is not None
@@ -111,6 +126,7 @@ class WorkerTasklet:
                         if plan is not None:
                             with self.tus.net(jid, self._next_phase()):
                                 self.orch.apply(plan)
+                            self._consume_shares()
                 self.trainer.on_epoch_finished(epoch)
                 ep_dt = time.perf_counter() - ep_t0
                 self.metrics.add_epoch(EpochMetrics(

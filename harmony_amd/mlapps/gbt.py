@@ -219,6 +219,10 @@ def build(job: JobConfig, ctx, cp):
                           world_size=ctx.world_size, device=ctx.device,
                           tables={MODEL_TABLE: table}, app_args=job.app_args)
     trainer = GBTTrainer(tctx)
-    provider = TrainingDataProvider(
+    def _reslice(b, frac):
+        n = max(1, int(b[0].shape[0] * frac))
+        return (b[0][:n], b[1][:n])
+
+    provider = TrainingDataProvider(reslice=_reslice, local_blocks=
         make_batches(job, ctx.rank, ctx.device, ctx.world_size))
     return {MODEL_TABLE: table}, trainer, provider

@@ -127,5 +127,9 @@ def build(job: JobConfig, ctx, cp):
                           world_size=ctx.world_size, device=ctx.device,
                           tables={MODEL_TABLE: table}, app_args=job.app_args)
     trainer = LassoTrainer(tctx)
-    provider = TrainingDataProvider(blocks)
+    def _reslice(b, frac):
+        n = max(1, int(b[0].shape[0] * frac))
+        return (b[0][:n], b[1][:n])
+
+    provider = TrainingDataProvider(reslice=_reslice, local_blocks=blocks)
     return {MODEL_TABLE: table}, trainer, provider

@@ -232,5 +232,15 @@ def build(job: JobConfig, ctx, cp):
                           world_size=ctx.world_size, device=ctx.device,
                           tables={MODEL_TABLE: table}, app_args=job.app_args)
     trainer = NMFTrainer(tctx, rows_local)
-    provider = TrainingDataProvider(blocks)
+    def _reslice(b, frac):
+        # row-prefix re-slice for SetBatchShareOp: rebuild the static
+        # precomputes (uniq_cols, col_sorted) for the smaller batch once
+        n = max(1, int(b.l_rows.shape[0] * frac))
+        nnz = int(b.row_ptr[n])
+        nb = NMFBatch(b.l_rows[:n], b.row_ptr[:n + 1].contiguous(),
+                      b.col_idx[:nnz], b.vals[:nnz])
+        nb.block_idx = getattr(b, "block_idx", None)
+        return nb
+
+    provider = TrainingDataProvider(blocks, reslice=_reslice)
     return {MODEL_TABLE: table}, trainer, provider

@@ -156,6 +156,20 @@ class SampleOptimizers:
         return _Conc()
 
     @staticmethod
+    def batch_shares(shares):
+        """Fixed per-rank mini-batch shares (exercises SetBatchShareOp
+        consumption: WorkerTasklet._consume_shares -> provider.set_share)."""
+        from harmony_amd.optimizer.plan import SetBatchShareOp
+
+        class _Sh(Optimizer):
+            def optimize(self, metrics, owners, world_size):
+                p = Plan()
+                p.ops.append(SetBatchShareOp(tuple(shares)))
+                return p
+
+        return _Sh()
+
+    @staticmethod
     def even_rebalance(table_id: str):
         class _Even(Optimizer):
             def optimize(self, metrics, owners, world_size):

@@ -121,3 +121,49 @@ def _elastic_validation_worker(rank, world):
 
 def test_elastic_validation_no_lost_updates():
     assert all(run_dist(_elastic_validation_worker, world=2, timeout=180))
+
+
+def test_batch_share_consumed_end_to_end():
+    # SetBatchShareOp flows plan -> executor -> worker -> provider reslice:
+    # rank 0's share of 0.5 halves each batch's examples after the first
+    # optimization window
+    from harmony_amd.config import JobConfig, RuntimeConfig
+    from harmony_amd.dolphin.master import run_job
+    from harmony_amd.optimizer.optimizers import SampleOptimizers
+    from harmony_amd.runtime.bootstrap import init_executor
+
+    job = JobConfig(job_id="j_share", app="mlr", max_num_epochs=3,
+                    num_mini_batches=4, optimizer_period=2,
+                    app_args={"num_classes": 4, "num_features": 32,
+                              "num_parts_per_class": 2, "batch_size": 200,
+                              "step_size": 0.1})
+    ctx = init_executor(RuntimeConfig(device="cpu"))
+    opt = SampleOptimizers.batch_shares(((0, 1),))
+    m = run_job(job, ctx, optimizer=opt)
+    per_batch = [b.num_examples for b in m.batches]
+    assert per_batch[0] == 200          # before the first window
+    assert per_batch[-1] == 200         # single rank: share == mean -> no-op
+
+    opt2 = _HalfShare()
+    job2 = JobConfig(job_id="j_share2", app="mlr", max_num_epochs=3,
+                     num_mini_batches=4, optimizer_period=2,
+                     app_args={"num_classes": 4, "num_features": 32,
+                               "num_parts_per_class": 2, "batch_size": 200,
+                               "step_size": 0.1})
+    m2 = run_job(job2, ctx, optimizer=opt2)
+    per_batch2 = [b.num_examples for b in m2.batches]
+    assert per_batch2[0] == 200
+    assert per_batch2[-1] == 100        # share 0.5 vs mean 1.0
+
+
+class _HalfShare:
+    """Optimizer whose shares put this rank at half the mean."""
+
+    def optimize(self, metrics, owners, world_size):
+        from harmony_amd.optimizer.plan import Plan, SetBatchShareOp
+
+        p = Plan()
+        # ranks 0..world-1 get weight 1 except rank 0 gets 0.5 of the mean:
+        # with world=1 the executor sees {0: 1} mean 2 via a phantom entry
+        p.ops.append(SetBatchShareOp(((0, 1), (1, 3))))
+        return p
