@@ -167,3 +167,34 @@ class _HalfShare:
         # with world=1 the executor sees {0: 1} mean 2 via a phantom entry
         p.ops.append(SetBatchShareOp(((0, 1), (1, 3))))
         return p
+
+
+def _share_dist_worker(rank, world):
+    from harmony_amd.config import JobConfig, RuntimeConfig
+    from harmony_amd.dolphin.master import run_job
+    from harmony_amd.optimizer.optimizers import SampleOptimizers
+    from harmony_amd.runtime.bootstrap import init_executor
+
+    ctx = init_executor(RuntimeConfig(device="cpu"))
+    # mean share = 2 -> rank 0 serves half batches, rank 1 clamps at full
+    opt = SampleOptimizers.batch_shares(((0, 1), (1, 3)))
+    job = JobConfig(job_id="j_share_d", app="mlr", max_num_epochs=3,
+                    num_mini_batches=2, optimizer_period=2,
+                    app_args={"num_classes": 4, "num_features": 32,
+                              "num_parts_per_class": 2, "batch_size": 100,
+                              "step_size": 0.1})
+    m = run_job(job, ctx, optimizer=opt)
+    return [b.num_examples for b in m.batches]
+
+
+def test_batch_share_two_ranks_gloo():
+    # uneven per-rank batch sizes must not break the collective pull/push
+    res = run_dist_opt(_share_dist_worker, world=2, timeout=180)
+    assert res[0][0] == 100 and res[0][-1] == 50    # rank 0 halved
+    assert res[1][0] == 100 and res[1][-1] == 100   # rank 1 clamped at 1.0
+
+
+def run_dist_opt(fn, world, timeout):
+    from tests.dist_helper import run_dist
+
+    return run_dist(fn, world=world, timeout=timeout)
