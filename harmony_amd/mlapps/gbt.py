@@ -142,6 +142,11 @@ class GBTTrainer(Trainer):
         self.table: ObjectTable = ctx.table(MODEL_TABLE)
         self.forest: List[GBTree] = []
         self._mse = 0.0
+        # incremental forest predictions: the forest is append-only
+        # (update fn v + [d]), so each block's prediction is cached and only
+        # trees added since the block's last visit are applied — O(new)
+        # instead of O(total) trees per batch
+        self._pred_cache: dict = {}
 
     def pull_model(self) -> None:
         # pullAllTrees (reference :767): gather every label's forest
@@ -151,9 +156,17 @@ class GBTTrainer(Trainer):
     def local_compute(self) -> None:
         bins, y = self.batch
         a = self.a
-        pred = torch.zeros_like(y)
-        for t in self.forest:
+        done, last, pred = self._pred_cache.get(id(self.batch),
+                                                (0, None, None))
+        # valid only if the cached prefix is literally this forest's prefix
+        # (offline eval may reload an older/replaced forest into the table)
+        if pred is None or done > len(self.forest) or (
+                done > 0 and self.forest[done - 1] is not last):
+            done, pred = 0, torch.zeros_like(y)
+        for t in self.forest[done:]:
             pred = pred + a["step_size"] * t.predict_bins(bins)
+        self._pred_cache[id(self.batch)] = (
+            len(self.forest), self.forest[-1] if self.forest else None, pred)
         resid = y - pred
         self._mse = float((resid * resid).mean())
         self.new_tree = build_tree(bins, resid, a["num_bins"], a["max_depth"],

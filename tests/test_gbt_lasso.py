@@ -68,3 +68,48 @@ def test_gbt_two_ranks_forest_shared():
     for nb, nt in res:
         assert nb == 4
         assert nt == 6   # before the 4th batch's push: 3 batches * 2 ranks
+
+
+def test_gbt_incremental_pred_matches_full():
+    # the append-only prediction cache must give the same residuals as a
+    # full forest re-predict, and must reset when the forest is replaced
+    import torch
+
+    from harmony_amd.mlapps.gbt import GBTree
+
+    torch.manual_seed(0)
+    bins = torch.randint(0, 16, (256, 8))
+    y = torch.randn(256)
+    trees = []
+    for i in range(5):
+        t = GBTree(2, [i % 8, (i + 1) % 8, (i + 2) % 8], [3, 7, 11],
+                   [0.1 * i, -0.2, 0.3, 0.05])
+        trees.append(t)
+    full = torch.zeros_like(y)
+    for t in trees:
+        full = full + 0.1 * t.predict_bins(bins)
+
+    class _Tr:
+        pass
+
+    from harmony_amd.mlapps.gbt import GBTTrainer
+
+    tr = _Tr()
+    tr._pred_cache = {}
+    tr.forest = []
+    tr.batch = (bins, y)
+    tr.a = {"step_size": 0.1, "num_bins": 16, "max_depth": 2, "lam": 1.0}
+    inc = None
+    for k in (2, 4, 5):            # grow the forest incrementally
+        tr.forest = trees[:k]
+        GBTTrainer.local_compute(tr)
+        inc = tr._mse
+    want = float(((y - full) ** 2).mean())
+    assert abs(inc - want) < 1e-5
+    # replace the forest with a *different* object list of the same length:
+    # cache must reset, not reuse
+    clones = [GBTree(t.depth, t.feature, t.threshold, t.leaf_value)
+              for t in trees]
+    tr.forest = clones
+    GBTTrainer.local_compute(tr)
+    assert abs(tr._mse - want) < 1e-5
