@@ -56,7 +56,18 @@ def _stderr_tails(world: int, port: int, n: int = 40) -> str:
 
 
 def run_dist(fn, world: int = 2, args=(), timeout: int = 120):
-    """Run fn(rank, world, *args) in `world` processes; returns [out_rank0..]."""
+    """Run fn(rank, world, *args) in `world` processes; returns [out_rank0..].
+    Retries once on a TCPStore port collision (free_port is check-then-use,
+    so a rare EADDRINUSE race with another process is possible)."""
+    try:
+        return _run_dist_once(fn, world, args, timeout)
+    except RuntimeError as e:
+        if "EADDRINUSE" not in str(e):
+            raise
+        return _run_dist_once(fn, world, args, timeout)
+
+
+def _run_dist_once(fn, world: int = 2, args=(), timeout: int = 120):
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
     port = free_port()
