@@ -153,6 +153,9 @@ class ThreadLocalTCPStore:
     def compare_set(self, key, expected, desired):
         return self._client().compare_set(key, expected, desired)
 
+    def append(self, key, value):
+        return self._client().append(key, value)
+
     def wait(self, keys):
         return self._client().wait(keys)
 
@@ -204,6 +207,11 @@ class LocalStore:
                 self._cv.notify_all()
                 return _to_bytes(desired)
             return cur if cur is not None else _to_bytes(expected)
+
+    def append(self, key: str, value) -> None:
+        with self._cv:
+            self._d[key] = self._d.get(key, b"") + _to_bytes(value)
+            self._cv.notify_all()
 
     def wait(self, keys) -> None:
         for k in keys:

@@ -25,6 +25,8 @@ import threading
 import time
 from typing import Dict, Optional, Set
 
+from harmony_amd.utils import sanitize
+
 
 class ControlPlane:
     def __init__(self, store, rank: int, world_size: int):
@@ -197,6 +199,11 @@ class TaskUnitScheduler:
                     with self._cv:
                         self._done.add(s)
             if not blockers:
+                # record in ISSUE order (post-wait): this is the order the
+                # rank actually enqueues the phase's collectives
+                if sanitize.enabled():
+                    sanitize.record(self.cp.store, self.cp.rank, job_id,
+                                    phase_idx, seq)
                 return seq
             with self._cv:
                 self._cv.wait(timeout=0.001)
