@@ -27,6 +27,10 @@ class JobScheduler:
     def on_resource_change(self, pool: "ResourcePool") -> None:
         pass
 
+    def order(self, pending: List[JobConfig]) -> List[JobConfig]:
+        """Order in which queued jobs are offered resources (default FIFO)."""
+        return pending
+
 
 class ResourcePool:
     """Homogeneous executor pool shared by all jobs (reference
@@ -72,10 +76,41 @@ class FIFOExclusiveScheduler(JobScheduler):
         return pool.all_ranks
 
 
+class PriorityScheduler(JobScheduler):
+    """Exclusive one-job-at-a-time like fifo, but the queue drains in
+    priority order: higher `-priority` first (app_args, default 0), FIFO
+    within a priority level."""
+
+    def on_job_arrival(self, job, pool):
+        if pool.running:
+            return None
+        return pool.all_ranks
+
+    def order(self, pending):
+        return sorted(pending,
+                      key=lambda j: -int(j.app_args.get("priority", 0)))
+
+
+class GangScheduler(JobScheduler):
+    """Each job gets `num_executors` executors EXCLUSIVELY (no co-location);
+    queues until that many executors are idle. The gang analogue for jobs
+    that cannot share GPUs."""
+
+    def on_job_arrival(self, job, pool):
+        n = int(job.app_args.get("num_executors", pool.world_size))
+        n = max(1, min(n, pool.world_size))
+        free = [r for r in pool.all_ranks if pool.load(r) == 0]
+        if len(free) < n:
+            return None
+        return free[:n]
+
+
 _BUILTIN = {
     "default": DefaultScheduler,
     "least_loaded": LeastLoadedScheduler,
     "fifo": FIFOExclusiveScheduler,
+    "priority": PriorityScheduler,
+    "gang": GangScheduler,
 }
 
 

@@ -86,7 +86,7 @@ class JobServerDriver:
     def _try_schedule(self) -> None:
         with self._lock:
             still = []
-            for job in self._pending:
+            for job in self.scheduler.order(list(self._pending)):
                 ranks = self.scheduler.on_job_arrival(job, self.pool)
                 if ranks:
                     self._publish(job, sorted(set(ranks)))

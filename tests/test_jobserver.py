@@ -113,3 +113,37 @@ def _jobserver_subset_worker(rank, world):
 
 def test_jobserver_subset_scheduling():
     assert all(run_dist(_jobserver_subset_worker, world=2, timeout=150))
+
+
+def test_priority_and_gang_schedulers():
+    from harmony_amd.config import JobConfig
+    from harmony_amd.jobserver.scheduler import (GangScheduler,
+                                                 PriorityScheduler,
+                                                 ResourcePool,
+                                                 load_scheduler)
+
+    assert isinstance(load_scheduler("priority"), PriorityScheduler)
+    assert isinstance(load_scheduler("gang"), GangScheduler)
+
+    pool = ResourcePool(4)
+    pri = PriorityScheduler()
+    lo = JobConfig(job_id="lo", app="mlr", app_args={"priority": 1})
+    hi = JobConfig(job_id="hi", app="mlr", app_args={"priority": 9})
+    # exclusive: queues while something runs
+    pool.running["x"] = [0, 1, 2, 3]
+    assert pri.on_job_arrival(lo, pool) is None
+    # queue drains highest priority first
+    assert [j.job_id for j in pri.order([lo, hi])] == ["hi", "lo"]
+    pool.running.clear()
+    assert pri.on_job_arrival(hi, pool) == [0, 1, 2, 3]
+
+    gang = GangScheduler()
+    a = JobConfig(job_id="a", app="mlr", app_args={"num_executors": 2})
+    b = JobConfig(job_id="b", app="mlr", app_args={"num_executors": 3})
+    ranks_a = gang.on_job_arrival(a, pool)
+    assert ranks_a == [0, 1]
+    pool.running["a"] = ranks_a
+    # only 2 idle left -> b (needs 3) queues; a second 2-gang fits
+    assert gang.on_job_arrival(b, pool) is None
+    c = JobConfig(job_id="c", app="mlr", app_args={"num_executors": 2})
+    assert gang.on_job_arrival(c, pool) == [2, 3]
