@@ -29,21 +29,33 @@ CREATE TABLE IF NOT EXISTS metrics (
 
 _PAGE = """<!doctype html><html><head><title>harmony_amd dashboard</title>
 <style>body{font-family:monospace;margin:2em}svg{border:1px solid #ccc}
-.m{margin-bottom:1.5em}</style></head><body>
-<h2>harmony_amd — job metrics</h2><div id="root">loading...</div>
+.m{margin-bottom:1.5em}a{margin-right:1em}</style></head><body>
+<h2>harmony_amd — job metrics</h2><div id="jobs"></div>
+<div id="root">loading...</div>
 <script>
-fetch('/data').then(r=>r.json()).then(rows=>{
+const sel=new URLSearchParams(location.search).get('job');
+fetch('/jobs').then(r=>r.json()).then(jobs=>{
+  document.getElementById('jobs').innerHTML='jobs: '+
+    ['<a href="/">all</a>'].concat(jobs.map(j=>
+      `<a href="/?job=${j.job_id}">${j.job_id}</a> (${j.reports} reports, `+
+      `${new Date(j.t0*1000).toISOString().slice(0,19)})`)).join(' | ');
+});
+fetch('/data'+(sel?'?job='+sel:'')).then(r=>r.json()).then(rows=>{
   const byJob={};
   rows.forEach(r=>{(byJob[r.job_id] ||= []).push(r);});
+  const chart=(pts,color)=>{
+    const W=600,H=120,mx=Math.max(...pts,1e-12);
+    const poly=pts.map((v,i)=>`${i*W/Math.max(pts.length-1,1)},${H-v/mx*H}`).join(' ');
+    return `<svg width=${W} height=${H}><polyline fill=none stroke=${color}
+      stroke-width=2 points="${poly}"/></svg> max ${mx.toFixed(2)}`;
+  };
   let html='';
   for (const [job,rs] of Object.entries(byJob)) {
-    const pts=rs.map(r=>{const p=JSON.parse(r.payload);
-      return p.data_processing_rate||p.dataProcessingRate||0;});
-    const W=600,H=120,mx=Math.max(...pts,1);
-    const poly=pts.map((v,i)=>`${i*W/Math.max(pts.length-1,1)},${H-v/mx*H}`).join(' ');
-    html+=`<div class=m><b>${job}</b> (${pts.length} reports, max ${mx.toFixed(0)} ex/s)
-      <br><svg width=${W} height=${H}><polyline fill=none stroke=steelblue
-      stroke-width=2 points="${poly}"/></svg></div>`;
+    const get=k=>rs.map(r=>{const p=JSON.parse(r.payload);return p[k]||0;});
+    const rate=get('data_processing_rate'), et=get('epoch_time_sec');
+    html+=`<div class=m><b>${job}</b> (${rs.length} reports)<br>
+      rate ex/s: ${chart(rate,'steelblue')}<br>`+
+      (et.some(v=>v>0)?`epoch sec: ${chart(et,'indianred')}`:'')+`</div>`;
   }
   document.getElementById('root').innerHTML=html||'no metrics yet';
 });
@@ -62,11 +74,29 @@ class _Handler(BaseHTTPRequestHandler):
         return con
 
     def do_GET(self):
-        if self.path.startswith("/data"):
+        if self.path.startswith("/jobs"):
+            # historical browsing: every job ever reported to this db file
             con = self._db()
             rows = con.execute(
-                "SELECT time, job_id, rank, kind, payload FROM metrics "
-                "ORDER BY id").fetchall()
+                "SELECT job_id, COUNT(*), MIN(time), MAX(time) FROM metrics "
+                "GROUP BY job_id ORDER BY MIN(id)").fetchall()
+            body = json.dumps([
+                {"job_id": j, "reports": n, "t0": t0, "t1": t1}
+                for j, n, t0, t1 in rows]).encode()
+            ct = "application/json"
+        elif self.path.startswith("/data"):
+            from urllib.parse import parse_qs, urlparse
+
+            q = parse_qs(urlparse(self.path).query)
+            con = self._db()
+            if "job" in q:
+                rows = con.execute(
+                    "SELECT time, job_id, rank, kind, payload FROM metrics "
+                    "WHERE job_id = ? ORDER BY id", (q["job"][0],)).fetchall()
+            else:
+                rows = con.execute(
+                    "SELECT time, job_id, rank, kind, payload FROM metrics "
+                    "ORDER BY id").fetchall()
             body = json.dumps([
                 {"time": t, "job_id": j, "rank": r, "kind": k, "payload": p}
                 for t, j, r, k, p in rows]).encode()

@@ -262,3 +262,28 @@ def test_sanitizer_multi_job_two_ranks_clean():
 
     res = run_dist(_san_dist_worker, world=2, timeout=180)
     assert res[0] == []
+
+
+def test_dashboard_jobs_index_and_filter(tmp_path):
+    import json
+    import urllib.request
+
+    from harmony_amd.dashboard.server import (DashboardConnector,
+                                              DashboardServer)
+
+    srv = DashboardServer(db_path=str(tmp_path / "d.db"))
+    port = srv.start()
+    try:
+        c = DashboardConnector(f"http://127.0.0.1:{port}")
+        assert c.send("jobA", 0, {"data_processing_rate": 10.0}, t=1.0)
+        assert c.send("jobA", 0, {"data_processing_rate": 20.0}, t=2.0)
+        assert c.send("jobB", 0, {"data_processing_rate": 5.0}, t=3.0)
+        jobs = json.loads(urllib.request.urlopen(
+            f"http://127.0.0.1:{port}/jobs", timeout=5).read())
+        assert [j["job_id"] for j in jobs] == ["jobA", "jobB"]
+        assert jobs[0]["reports"] == 2 and jobs[0]["t0"] == 1.0
+        only_b = json.loads(urllib.request.urlopen(
+            f"http://127.0.0.1:{port}/data?job=jobB", timeout=5).read())
+        assert len(only_b) == 1 and only_b[0]["job_id"] == "jobB"
+    finally:
+        srv.stop()
