@@ -229,7 +229,9 @@ std::vector<torch::Tensor> nmf_grad_twopass(
   const char* env = getenv("HARMONY_NMF_G");
   if (env && atoi(env) == 64) G = 64;
   if (env && atoi(env) == 32 && k <= 32 * MAXC) G = 32;
-  const int threads = 256;
+  int threads = 256;   // HARMONY_NMF_THREADS: occupancy A/B knob
+  const char* te = getenv("HARMONY_NMF_THREADS");
+  if (te) { int t = atoi(te); if (t == 128 || t == 256 || t == 512 || t == 1024) threads = t; }
   if (n > 0) {
     dim3 grid((n + threads / G - 1) / (threads / G));
     if (G == 32)
