@@ -47,6 +47,7 @@ def parse_args():
     p.add_argument("--lda-topics", type=int, default=256)
     p.add_argument("--lda-docs-per-batch", type=int, default=16384)
     p.add_argument("--lda-tokens-per-doc", type=int, default=128)
+    p.add_argument("--lda-alias-refresh", type=int, default=4)
     p.add_argument("--lda-sampler", type=str, default="alias",
                    choices=["exact", "alias"])
     return p.parse_args()
@@ -85,7 +86,8 @@ def make_jobs(args, world: int):
                       "num_topics": args.lda_topics,
                       "tokens_per_doc": args.lda_tokens_per_doc,
                       "docs_per_batch": args.lda_docs_per_batch,
-                      "sampler": args.lda_sampler})
+                      "sampler": args.lda_sampler,
+                      "alias_refresh": args.lda_alias_refresh})
     return jobs
 
 
