@@ -53,6 +53,33 @@ def run_job(job: JobConfig, ctx: ExecutorContext,
     tus = tus or TaskUnitScheduler(cp, {job.job_id}, multi_job=False)
     app = mlapps.get_app(job.app)
     tables, trainer, provider = app.build(job, ctx, cp)
+    if job.restore_chkp:
+        # start from a model snapshot (reference ETMaster.createTable(chkpId):
+        # block files are rank-independent, so the restore re-partitions for
+        # whatever executor set this job runs on). Accepts either a plain
+        # checkpoint id or a ModelChkpManager epoch id (per-table subdirs).
+        from harmony_amd.dolphin.model_eval import _safe
+        from harmony_amd.et.checkpoint import CheckpointManager
+
+        cm = CheckpointManager(temp_root=job.chkp_path)
+        for t in tables.values():
+            if not hasattr(t, "cfg"):
+                continue
+            cid = f"{job.restore_chkp}/{_safe(t.cfg.table_id)}"
+            app_id, use = ((job.job_id, cid) if cm.exists(job.job_id, cid)
+                           else (None, None))
+            if app_id is None:
+                # cross-job restore: "<src_job>/<chkp_id>" syntax
+                src, _, rest = job.restore_chkp.partition("/")
+                for c in (f"{rest}/{_safe(t.cfg.table_id)}", rest):
+                    if rest and cm.exists(src, c):
+                        app_id, use = src, c
+                        break
+            if app_id is None:
+                raise FileNotFoundError(
+                    f"restore_chkp {job.restore_chkp!r}: no checkpoint for "
+                    f"table {t.cfg.table_id} under {job.chkp_path}")
+            cm.load_into(t, app_id, use)
     orch = None
     if optimizer is None and job.optimizer:
         optimizer = _make_optimizer(job.optimizer)

@@ -29,7 +29,7 @@ _JOB_FIELDS = {"max_num_epochs": int, "num_mini_batches": int,
                "num_worker_blocks": int, "clock_slack": int,
                "num_trainer_threads": int, "optimizer": str,
                "optimizer_period": int, "dashboard_url": str,
-               "trace_path": str, "chkp_path": str,
+               "trace_path": str, "chkp_path": str, "restore_chkp": str,
                "model_chkp_per_epoch": _bool, "offline_model_eval": _bool}
 
 
