@@ -31,6 +31,7 @@
 // for sample-exact CPU/GPU tests.
 
 #include "hip_common.h"
+#include <cstdlib>
 #include <algorithm>
 
 namespace {
@@ -125,7 +126,7 @@ __global__ void alias_build_kernel(const int* __restrict__ word_topic,
   }
 }
 
-__global__ __launch_bounds__(ALIAS_THREADS)
+__global__ __launch_bounds__(256)
 void lda_mh_kernel(int* __restrict__ doc_topic,        // [D][K] int32
                    const int* __restrict__ word_topic, // [rows][K] (fresh)
                    const float* __restrict__ invden,   // [K]
@@ -259,8 +260,11 @@ torch::Tensor lda_mh(torch::Tensor doc_topic, torch::Tensor word_topic,
   CHECK_IN(assignments);
   const int D = doc_topic.size(0), K = doc_topic.size(1);
   if (D == 0) return assignments;
-  dim3 blk(ALIAS_THREADS), grid((D + ALIAS_THREADS - 1) / ALIAS_THREADS);
-  const size_t shmem = (size_t)ALIAS_THREADS * K;     // u8 rows
+  int threads = ALIAS_THREADS;    // HARMONY_LDA_MH_THREADS: fill/LDS A/B
+  const char* te = getenv("HARMONY_LDA_MH_THREADS");
+  if (te) { int t = atoi(te); if (t==64 || t==128 || t==256) threads = t; }
+  dim3 blk(threads), grid((D + threads - 1) / threads);
+  const size_t shmem = (size_t)threads * K;           // u8 rows
   TORCH_CHECK(shmem <= 160 * 1024, "K too large for u8 LDS rows");
   hipLaunchKernelGGL(lda_mh_kernel, grid, blk, shmem, current_stream(),
                      doc_topic.data_ptr<int>(), word_topic.data_ptr<int>(),
