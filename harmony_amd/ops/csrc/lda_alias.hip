@@ -144,28 +144,56 @@ void lda_mh_kernel(int* __restrict__ doc_topic,        // [D][K] int32
   const int tid = threadIdx.x;
   const int doc = blockIdx.x * blockDim.x + tid;
   unsigned char* nd = nd8 + (size_t)tid * K;
-  if (doc >= n_docs) return;
-  for (int k = 0; k < K; ++k)
-    nd[k] = (unsigned char)doc_topic[(int64_t)doc * K + k];
-  const int64_t p0 = doc_offsets[doc], p1 = doc_offsets[doc + 1];
+  // coalesced block-wide row staging: the WG's docs are contiguous, so the
+  // [docs x K] slice loads as one flat stream (thread-private loops would
+  // put adjacent lanes 1 KB apart — every 4 B load its own cacheline)
+  const int64_t gbase = (int64_t)blockIdx.x * blockDim.x * K;
+  const int ndocs_wg =
+      min((int)blockDim.x, n_docs - (int)(blockIdx.x * blockDim.x));
+  for (int i = tid; i < ndocs_wg * K; i += blockDim.x)
+    nd8[i] = (unsigned char)doc_topic[gbase + i];
+  __syncthreads();
+  // inactive tail threads skip the token loop but MUST reach the final
+  // barrier (an early return would strand the block-wide __syncthreads)
+  const bool active = doc < n_docs;
+  const int64_t p0 = active ? doc_offsets[doc] : 0;
+  const int64_t p1 = active ? doc_offsets[doc + 1] : 0;
   const float Ld = (float)(p1 - p0);
   const float aK = alpha * (float)K;
   const float p_uniform = aK / (aK + Ld);
   const int S = K / WAVE;
+  // software-prefetched word-proposal top row: token p+1's first-level
+  // alias lookup depends only on (seed, p+1, word_ids[p+1]) — issue its
+  // two loads before token p's dependent acceptance chain
+  float u1_pf = 0.f; float tpv_pf = 0.f; int tav_pf = 0; int gb_pf = 0;
+  if (p0 < p1) {
+    const float u1 = rng_uniform(seed, (unsigned int)(p0 * 8)) * (float)WAVE;
+    int gb = (int)u1;
+    if (gb >= WAVE) gb = WAVE - 1;
+    const int64_t tbase = word_ids[p0] * WAVE;
+    u1_pf = u1; gb_pf = gb;
+    tpv_pf = top_prob[tbase + gb]; tav_pf = top_alias[tbase + gb];
+  }
   for (int64_t p = p0; p < p1; ++p) {
     const int64_t w = word_ids[p];
     const int64_t wbase = w * K;
+    const float u1_cur = u1_pf; const float tpv = tpv_pf;
+    const int tav = tav_pf; const int gb_cur = gb_pf;
+    if (p + 1 < p1) {
+      const float u1n =
+          rng_uniform(seed, (unsigned int)((p + 1) * 8)) * (float)WAVE;
+      int gbn = (int)u1n;
+      if (gbn >= WAVE) gbn = WAVE - 1;
+      const int64_t tbn = word_ids[p + 1] * WAVE;
+      u1_pf = u1n; gb_pf = gbn;
+      tpv_pf = top_prob[tbn + gbn]; tav_pf = top_alias[tbn + gbn];
+    }
     int s = z[p];
     nd[s] -= 1;                                       // exclude the token
     const unsigned int c0 = (unsigned int)(p * 8);
     // ---- word proposal (two-level alias): acceptance = doc factor ---
     {
-      const float u1 = rng_uniform(seed, c0 + 0) * (float)WAVE;
-      int gb = (int)u1;
-      if (gb >= WAVE) gb = WAVE - 1;
-      const int64_t tbase = w * WAVE;
-      const int g = (u1 - (float)gb < top_prob[tbase + gb])
-                        ? gb : top_alias[tbase + gb];
+      const int g = (u1_cur - (float)gb_cur < tpv) ? gb_cur : tav;
       const float u2 = rng_uniform(seed, c0 + 6) * (float)S;
       int eb = (int)u2;
       if (eb >= S) eb = S - 1;
@@ -208,8 +236,9 @@ void lda_mh_kernel(int* __restrict__ doc_topic,        // [D][K] int32
     nd[s] += 1;
     z[p] = s;
   }
-  for (int k = 0; k < K; ++k)
-    doc_topic[(int64_t)doc * K + k] = (int)nd[k];
+  __syncthreads();
+  for (int i = tid; i < ndocs_wg * K; i += blockDim.x)
+    doc_topic[gbase + i] = (int)nd8[i];
 }
 
 }  // namespace
