@@ -126,6 +126,7 @@ __global__ void alias_build_kernel(const int* __restrict__ word_topic,
   }
 }
 
+template <bool PF>
 __global__ __launch_bounds__(256)
 void lda_mh_kernel(int* __restrict__ doc_topic,        // [D][K] int32
                    const int* __restrict__ word_topic, // [rows][K] (fresh)
@@ -166,7 +167,7 @@ void lda_mh_kernel(int* __restrict__ doc_topic,        // [D][K] int32
   // alias lookup depends only on (seed, p+1, word_ids[p+1]) — issue its
   // two loads before token p's dependent acceptance chain
   float u1_pf = 0.f; float tpv_pf = 0.f; int tav_pf = 0; int gb_pf = 0;
-  if (p0 < p1) {
+  if (PF && p0 < p1) {
     const float u1 = rng_uniform(seed, (unsigned int)(p0 * 8)) * (float)WAVE;
     int gb = (int)u1;
     if (gb >= WAVE) gb = WAVE - 1;
@@ -179,7 +180,7 @@ void lda_mh_kernel(int* __restrict__ doc_topic,        // [D][K] int32
     const int64_t wbase = w * K;
     const float u1_cur = u1_pf; const float tpv = tpv_pf;
     const int tav = tav_pf; const int gb_cur = gb_pf;
-    if (p + 1 < p1) {
+    if (PF && p + 1 < p1) {
       const float u1n =
           rng_uniform(seed, (unsigned int)((p + 1) * 8)) * (float)WAVE;
       int gbn = (int)u1n;
@@ -193,7 +194,16 @@ void lda_mh_kernel(int* __restrict__ doc_topic,        // [D][K] int32
     const unsigned int c0 = (unsigned int)(p * 8);
     // ---- word proposal (two-level alias): acceptance = doc factor ---
     {
-      const int g = (u1_cur - (float)gb_cur < tpv) ? gb_cur : tav;
+      int g;
+      if (PF) {
+        g = (u1_cur - (float)gb_cur < tpv) ? gb_cur : tav;
+      } else {
+        const float u1 = rng_uniform(seed, c0 + 0) * (float)WAVE;
+        int gb = (int)u1;
+        if (gb >= WAVE) gb = WAVE - 1;
+        const int64_t tb = w * WAVE;
+        g = (u1 - (float)gb < top_prob[tb + gb]) ? gb : top_alias[tb + gb];
+      }
       const float u2 = rng_uniform(seed, c0 + 6) * (float)S;
       int eb = (int)u2;
       if (eb >= S) eb = S - 1;
@@ -295,7 +305,10 @@ torch::Tensor lda_mh(torch::Tensor doc_topic, torch::Tensor word_topic,
   dim3 blk(threads), grid((D + threads - 1) / threads);
   const size_t shmem = (size_t)threads * K;           // u8 rows
   TORCH_CHECK(shmem <= 160 * 1024, "K too large for u8 LDS rows");
-  hipLaunchKernelGGL(lda_mh_kernel, grid, blk, shmem, current_stream(),
+  const char* pe = getenv("HARMONY_LDA_MH_PREFETCH");
+  const bool pf = !(pe && atoi(pe) == 0);   // default: prefetch on
+  auto kern = pf ? lda_mh_kernel<true> : lda_mh_kernel<false>;
+  hipLaunchKernelGGL(kern, grid, blk, shmem, current_stream(),
                      doc_topic.data_ptr<int>(), word_topic.data_ptr<int>(),
                      invden.data_ptr<float>(), prob.data_ptr<float>(),
                      alias.data_ptr<int>(), top_prob.data_ptr<float>(),

@@ -37,11 +37,10 @@ def main():
     prob, alias, tp, ta, qv, qsum, invden = tabs
     torch.cuda.synchronize()
     print(f"uniq words {R}")
-    for thr in (None, "64", "128", "256"):
-        if thr is None:
-            os.environ.pop("HARMONY_LDA_MH_THREADS", None)
-        else:
-            os.environ["HARMONY_LDA_MH_THREADS"] = thr
+    for thr, pf in (("128", "0"), ("128", "1"), ("64", "0"), ("64", "1"),
+                    ("256", "1")):
+        os.environ["HARMONY_LDA_MH_THREADS"] = thr
+        os.environ["HARMONY_LDA_MH_PREFETCH"] = pf
         zz = z.clone()
         dtt = dt.clone()
         for _ in range(3):
@@ -56,7 +55,7 @@ def main():
             torch.cuda.synchronize()
             ts.append(time.perf_counter() - t0)
         ts.sort()
-        print(f"threads={thr or 'default'}: {ts[len(ts)//2]*1e3:.3f} ms")
+        print(f"threads={thr} prefetch={pf}: {ts[len(ts)//2]*1e3:.3f} ms")
 
 
 if __name__ == "__main__":
