@@ -36,7 +36,10 @@
 
 namespace {
 
-constexpr int ALIAS_THREADS = 128;  // A/B: 64 threads measured 1.74 ms 3-job vs 1.52 at 128
+constexpr int ALIAS_THREADS = 64;   // same-box interleaved A/B: 64 beats 128
+                                    // (isolated 0.59 vs 0.655 ms; 3-job bench
+                                    // 1.59/1.62 vs 1.63/1.66 ms/step) — the
+                                    // earlier '64 = 1.74 ms' was cross-box noise
 constexpr int BUILD_WAVES = 4;
 
 // In-LDS serial Vose over n entries at pr[0..n), links lk, output al.
