@@ -166,42 +166,53 @@ void lda_mh_kernel(int* __restrict__ doc_topic,        // [D][K] int32
   const float aK = alpha * (float)K;
   const float p_uniform = aK / (aK + Ld);
   const int S = K / WAVE;
-  // software-prefetched word proposal: token p+1's ENTIRE two-level alias
-  // draw (top row -> entry row, two dependent loads) depends only on
-  // (seed, p+1, word_ids[p+1]), so it pipelines one token ahead — both
-  // latencies overlap token p's acceptance + doc-proposal work
-  auto word_proposal = [&](int64_t pp) -> int {
-    const unsigned int cc = (unsigned int)(pp * 8);
-    const int64_t ww = word_ids[pp];
-    const float u1 = rng_uniform(seed, cc + 0) * (float)WAVE;
+  // software-prefetched word-proposal top row: token p+1's first-level
+  // alias lookup depends only on (seed, p+1, word_ids[p+1]) — issue its
+  // two loads before token p's dependent acceptance chain
+  float u1_pf = 0.f; float tpv_pf = 0.f; int tav_pf = 0; int gb_pf = 0;
+  if (PF && p0 < p1) {
+    const float u1 = rng_uniform(seed, (unsigned int)(p0 * 8)) * (float)WAVE;
     int gb = (int)u1;
     if (gb >= WAVE) gb = WAVE - 1;
-    const int64_t tb = ww * WAVE;
-    const int g = (u1 - (float)gb < top_prob[tb + gb])
-                      ? gb : top_alias[tb + gb];
-    const float u2 = rng_uniform(seed, cc + 6) * (float)S;
-    int eb = (int)u2;
-    if (eb >= S) eb = S - 1;
-    const int64_t eb2 = ww * (int64_t)K + g * S + eb;
-    return g * S + ((u2 - (float)eb < prob[eb2]) ? eb : alias[eb2]);
-  };
-  int t1_pf = 0;
-  if (PF && p0 < p1) t1_pf = word_proposal(p0);
+    const int64_t tbase = word_ids[p0] * WAVE;
+    u1_pf = u1; gb_pf = gb;
+    tpv_pf = top_prob[tbase + gb]; tav_pf = top_alias[tbase + gb];
+  }
   for (int64_t p = p0; p < p1; ++p) {
     const int64_t w = word_ids[p];
     const int64_t wbase = w * K;
-    int t1;
-    if (PF) {
-      t1 = t1_pf;
-      if (p + 1 < p1) t1_pf = word_proposal(p + 1);
-    } else {
-      t1 = word_proposal(p);
+    const float u1_cur = u1_pf; const float tpv = tpv_pf;
+    const int tav = tav_pf; const int gb_cur = gb_pf;
+    if (PF && p + 1 < p1) {
+      const float u1n =
+          rng_uniform(seed, (unsigned int)((p + 1) * 8)) * (float)WAVE;
+      int gbn = (int)u1n;
+      if (gbn >= WAVE) gbn = WAVE - 1;
+      const int64_t tbn = word_ids[p + 1] * WAVE;
+      u1_pf = u1n; gb_pf = gbn;
+      tpv_pf = top_prob[tbn + gbn]; tav_pf = top_alias[tbn + gbn];
     }
     int s = z[p];
     nd[s] -= 1;                                       // exclude the token
     const unsigned int c0 = (unsigned int)(p * 8);
     // ---- word proposal (two-level alias): acceptance = doc factor ---
     {
+      int g;
+      if (PF) {
+        g = (u1_cur - (float)gb_cur < tpv) ? gb_cur : tav;
+      } else {
+        const float u1 = rng_uniform(seed, c0 + 0) * (float)WAVE;
+        int gb = (int)u1;
+        if (gb >= WAVE) gb = WAVE - 1;
+        const int64_t tb = w * WAVE;
+        g = (u1 - (float)gb < top_prob[tb + gb]) ? gb : top_alias[tb + gb];
+      }
+      const float u2 = rng_uniform(seed, c0 + 6) * (float)S;
+      int eb = (int)u2;
+      if (eb >= S) eb = S - 1;
+      const int64_t ebase = wbase + g * S;
+      const int t1 = g * S + ((u2 - (float)eb < prob[ebase + eb])
+                                  ? eb : alias[ebase + eb]);
       // acceptance: pi uses the CURRENT snapshot, q the (possibly older)
       // table's encoded density — with a fresh table the word factors
       // cancel mathematically; the explicit form stays correct when the
