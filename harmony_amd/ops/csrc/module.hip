@@ -38,6 +38,12 @@ torch::Tensor lda_mh(torch::Tensor doc_topic, torch::Tensor word_topic,
                      double alpha, double beta, int64_t seed);
 void lasso_cd(torch::Tensor Xt, torch::Tensor r, torch::Tensor w,
               torch::Tensor col_sq, double lam_n);
+torch::Tensor os_shard_alloc(int64_t rows, int64_t k);
+torch::Tensor os_ipc_handle(torch::Tensor shard);
+int64_t os_ipc_open(torch::Tensor handle_bytes);
+void os_ipc_close(int64_t ptr);
+torch::Tensor os_gather(int64_t ptr, torch::Tensor idx, int64_t k);
+void os_scatter_add(int64_t ptr, torch::Tensor idx, torch::Tensor delta);
 void gbt_hist(torch::Tensor bins, torch::Tensor resid, torch::Tensor node,
               torch::Tensor cnt, torch::Tensor sum);
 void scatter_apply(torch::Tensor shard, torch::Tensor rows,
@@ -70,6 +76,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Metropolis-Hastings alias LDA sweep, thread-per-doc (K7b)");
   m.def("gbt_hist", &gbt_hist, "GBT level histogram build (K10)");
   m.def("lasso_cd", &lasso_cd, "Lasso persistent CD sweep (K11)");
+  m.def("os_shard_alloc", &os_shard_alloc, "one-sided shard (hipMalloc)");
+  m.def("os_ipc_handle", &os_ipc_handle, "export shard via hipIpc");
+  m.def("os_ipc_open", &os_ipc_open, "map a peer shard (xGMI)");
+  m.def("os_ipc_close", &os_ipc_close, "unmap a peer shard");
+  m.def("os_gather", &os_gather, "gather rows from a mapped shard (K12)");
+  m.def("os_scatter_add", &os_scatter_add,
+        "atomic scatter-add into a mapped shard (K12)");
   m.def("scatter_apply", &scatter_apply, "owner-side sparse update (K9)");
   m.def("dense_apply", &dense_apply, "owner-side dense update (K3)");
   m.def("parse_nmf_bytes", &parse_nmf_bytes, "native NMF text parser");

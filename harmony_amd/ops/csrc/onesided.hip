@@ -1,0 +1,120 @@
+// One-sided xGMI data plane primitives (K12).
+//
+// Reference: the ET remote-access path (RemoteAccessOpSender/Handler,
+// CommManager — SURVEY §2.1) gives workers ASYNC per-key access to remote
+// shards over sockets. The collective data plane (et/comm.py) replaces it
+// with bulk-synchronous RCCL ops; THIS file is the async analogue the
+// MI355X way: each rank's shard is exported with hipIpcGetMemHandle
+// (dmabuf mode — HSA_ENABLE_IPC_MODE_LEGACY=0), peers map it once and
+// dereference it directly from gather/scatter kernels. On one node the
+// mapping is peer HBM over xGMI (p2p loads/stores, ≈153 GB/s per link);
+// no message, no rendezvous, no collective — a pull is a kernel.
+//
+// Shards for one-sided tables are allocated with hipMalloc directly
+// (NOT the torch caching allocator: IPC handles need the allocation
+// base, and torch's arena offsets break hipIpcGetMemHandle).
+
+#include "hip_common.h"
+
+namespace {
+
+constexpr int GATHER_THREADS = 256;
+
+// rows from a (possibly remote) shard: out[i] = shard[idx[i]]
+__global__ void gather_rows_kernel(const float* __restrict__ shard,
+                                   const int64_t* __restrict__ idx,
+                                   float* __restrict__ out,
+                                   int n, int k) {
+  const int64_t t = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t total = (int64_t)n * k;
+  for (int64_t i = t; i < total; i += (int64_t)gridDim.x * blockDim.x) {
+    const int64_t r = i / k, c = i - r * k;
+    out[i] = shard[idx[r] * k + c];
+  }
+}
+
+// atomic add into a (possibly remote) shard: shard[idx[i]] += delta[i]
+__global__ void scatter_add_rows_kernel(float* __restrict__ shard,
+                                        const int64_t* __restrict__ idx,
+                                        const float* __restrict__ delta,
+                                        int n, int k) {
+  const int64_t t = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t total = (int64_t)n * k;
+  for (int64_t i = t; i < total; i += (int64_t)gridDim.x * blockDim.x) {
+    const int64_t r = i / k, c = i - r * k;
+    atomicAdd(&shard[idx[r] * k + c], delta[i]);
+  }
+}
+
+dim3 grid_for(int64_t total) {
+  int64_t blocks = (total + GATHER_THREADS - 1) / GATHER_THREADS;
+  if (blocks > 4096) blocks = 4096;   // grid-stride; >> 256 CUs
+  if (blocks < 1) blocks = 1;
+  return dim3((unsigned)blocks);
+}
+
+}  // namespace
+
+torch::Tensor os_shard_alloc(int64_t rows, int64_t k) {
+  // hipMalloc-backed f32 tensor usable as an IPC export base
+  void* p = nullptr;
+  size_t bytes = (size_t)rows * k * sizeof(float);
+  TORCH_CHECK(hipMalloc(&p, bytes) == hipSuccess, "hipMalloc failed");
+  TORCH_CHECK(hipMemset(p, 0, bytes) == hipSuccess);
+  auto opts = torch::TensorOptions()
+                  .dtype(torch::kFloat32)
+                  .device(torch::kCUDA, 0);
+  return torch::from_blob(
+      p, {rows, k}, [](void* q) { hipFree(q); }, opts);
+}
+
+torch::Tensor os_ipc_handle(torch::Tensor shard) {
+  CHECK_IN(shard);
+  hipIpcMemHandle_t h;
+  TORCH_CHECK(hipIpcGetMemHandle(&h, shard.data_ptr()) == hipSuccess,
+              "hipIpcGetMemHandle failed (needs dmabuf IPC: "
+              "HSA_ENABLE_IPC_MODE_LEGACY=0)");
+  auto out = torch::empty({(int64_t)sizeof(h)},
+                          torch::TensorOptions().dtype(torch::kUInt8));
+  memcpy(out.data_ptr(), &h, sizeof(h));
+  return out;
+}
+
+int64_t os_ipc_open(torch::Tensor handle_bytes) {
+  TORCH_CHECK(handle_bytes.numel() == (int64_t)sizeof(hipIpcMemHandle_t));
+  hipIpcMemHandle_t h;
+  memcpy(&h, handle_bytes.contiguous().data_ptr(), sizeof(h));
+  void* p = nullptr;
+  TORCH_CHECK(hipIpcOpenMemHandle(&p, h, hipIpcMemLazyEnablePeerAccess)
+                  == hipSuccess, "hipIpcOpenMemHandle failed");
+  return (int64_t)(uintptr_t)p;
+}
+
+void os_ipc_close(int64_t ptr) {
+  TORCH_CHECK(hipIpcCloseMemHandle((void*)(uintptr_t)ptr) == hipSuccess);
+}
+
+torch::Tensor os_gather(int64_t ptr, torch::Tensor idx, int64_t k) {
+  CHECK_IN(idx);
+  const int n = idx.numel();
+  auto out = torch::empty({(int64_t)n, k},
+                          idx.options().dtype(torch::kFloat32));
+  if (n == 0) return out;
+  hipLaunchKernelGGL(gather_rows_kernel, grid_for((int64_t)n * k),
+                     dim3(GATHER_THREADS), 0, current_stream(),
+                     (const float*)(uintptr_t)ptr,
+                     idx.data_ptr<int64_t>(), out.data_ptr<float>(),
+                     n, (int)k);
+  return out;
+}
+
+void os_scatter_add(int64_t ptr, torch::Tensor idx, torch::Tensor delta) {
+  CHECK_IN(idx); CHECK_IN(delta);
+  const int n = idx.numel();
+  if (n == 0) return;
+  const int k = delta.size(1);
+  hipLaunchKernelGGL(scatter_add_rows_kernel, grid_for((int64_t)n * k),
+                     dim3(GATHER_THREADS), 0, current_stream(),
+                     (float*)(uintptr_t)ptr, idx.data_ptr<int64_t>(),
+                     delta.data_ptr<float>(), n, k);
+}
