@@ -9,13 +9,22 @@ sys.path.insert(0, ".")
 
 
 def child(q_h, q_done):
+    try:
+        _child(q_h, q_done)
+    except Exception:  # noqa: BLE001
+        import traceback
+
+        q_done.put("CHILD FAILED:\n" + traceback.format_exc())
+
+
+def _child(q_h, q_done):
     import torch
 
     from harmony_amd import ops
 
     hip = ops._load_hip()
     torch.cuda.set_device(0)
-    handle = q_h.get(timeout=60)
+    handle = q_h.get(timeout=300)
     ptr = hip.os_ipc_open(torch.tensor(list(handle), dtype=torch.uint8))
     idx = torch.tensor([0, 2, 5], dtype=torch.int64, device="cuda")
     got = hip.os_gather(ptr, idx, 4)
@@ -46,7 +55,7 @@ def main():
     p = ctx.Process(target=child, args=(q_h, q_done))
     p.start()
     q_h.put(bytes(h.tolist()))
-    print(q_done.get(timeout=120))
+    print(q_done.get(timeout=300))
     p.join(timeout=30)
     torch.cuda.synchronize()
     # child's +10 scatter must be visible here
