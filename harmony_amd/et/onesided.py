@@ -1,0 +1,163 @@
+"""One-sided (async) table over hipIpc-mapped peer shards.
+
+Reference: the ET remote-access path (RemoteAccessOpSender/Handler +
+CommManager) lets any worker pull/push any key at any time —
+no bulk synchrony. The collective data plane (et/comm.py) is the
+throughput path; THIS is its async analogue, the MI355X way:
+
+  * each rank's shard lives in hipMalloc memory (NOT the torch caching
+    allocator — IPC needs the allocation base) and is exported once with
+    `hipIpcGetMemHandle` (dmabuf mode: HSA_ENABLE_IPC_MODE_LEGACY=0);
+  * every peer maps every shard once at setup; a PULL is then just a
+    gather kernel dereferencing the peer pointer (peer HBM over xGMI,
+    p2p loads), and a PUSH is an atomicAdd scatter kernel — no message,
+    no rendezvous, no matching collective on the owner;
+  * consistency is per-element atomic adds (the owner-side `add` update
+    function algebra); richer update functions need the collective path.
+
+This makes SSP slack > 0 a REAL bounded-async mode: workers proceed at
+their own pace (uneven batch counts, no deadlock) with the SSP clock as
+the only cross-worker coupling. v1 scope: float32 dense tables, static
+ownership (no live migration of one-sided tables), `add` update fn.
+"""
+
+from __future__ import annotations
+
+import torch
+
+from harmony_amd.config import TableConfig
+from harmony_amd.et.table import Table
+
+
+def _require_hip():
+    from harmony_amd import ops
+
+    hip = ops._load_hip()
+    if hip is None or not torch.cuda.is_available():
+        raise RuntimeError("one-sided tables need the HIP extension + a GPU")
+    return hip
+
+
+class OneSidedTable(Table):
+    """Async-access dense table. Construct on every rank, then `connect()`
+    collectively (exchanges IPC handles through the control store)."""
+
+    def __init__(self, cfg: TableConfig, rank: int, world_size: int,
+                 device: torch.device, store=None):
+        assert cfg.update_fn == "add", \
+            "one-sided push is atomicAdd: update_fn must be 'add'"
+        assert cfg.dtype == "float32", "one-sided tables are float32 (v1)"
+        self._hip = _require_hip()
+        super().__init__(cfg, rank, world_size, device)
+        self.store = store
+        self._peer_ptr = {}          # rank -> mapped device pointer (int)
+        # every rank's block->slot map is derivable from the static
+        # round-robin ownership, so remote row indices need no exchange
+        self._peer_slot = {}
+        for r in range(world_size):
+            slot = torch.full((cfg.num_blocks,), -1, dtype=torch.int64)
+            for i, b in enumerate(self.ownership.owned_blocks(r)):
+                slot[b] = i
+            self._peer_slot[r] = slot.to(device)
+
+    # ------------------------------------------------------------- lifecycle
+
+    def _alloc_shard(self) -> None:
+        # hipMalloc-backed shard (IPC-exportable base), then the normal
+        # deterministic per-block init copied in
+        owned = self.ownership.owned_blocks(self.rank)
+        self._local_blocks = list(owned)
+        rows = len(owned) * self.part.block_size
+        self.shard = self._hip.os_shard_alloc(rows, self.cfg.value_dim)
+        from harmony_amd.et import update_functions as uf
+        from harmony_amd.et.table import _block_seed
+
+        init = uf.init_fn(self.cfg.init_fn)
+        bs = self.part.block_size
+        for i, b in enumerate(owned):
+            t = init((bs, self.cfg.value_dim), self.dtype, self.device,
+                     seed=_block_seed(self.cfg.table_id, b),
+                     **self.cfg.init_args)
+            self.shard[i * bs:(i + 1) * bs] = t
+        slot = torch.full((self.cfg.num_blocks,), -1, dtype=torch.int64)
+        for i, b in enumerate(owned):
+            slot[b] = i
+        self._block_slot = slot.to(self.device)
+
+    def connect(self, store=None) -> None:
+        """Collective: export my shard, map every peer's (via the store)."""
+        store = store or self.store
+        assert store is not None
+        torch.cuda.synchronize()
+        h = self._hip.os_ipc_handle(self.shard)
+        key = f"os/{self.cfg.table_id}"
+        store.set(f"{key}/{self.rank}", bytes(h.tolist()).hex())
+        for r in range(self.world_size):
+            if r == self.rank:
+                continue
+            raw = store.get(f"{key}/{r}")
+            hb = torch.tensor(list(bytes.fromhex(raw.decode())),
+                              dtype=torch.uint8)
+            self._peer_ptr[r] = self._hip.os_ipc_open(hb)
+
+    def close(self) -> None:
+        for p in self._peer_ptr.values():
+            self._hip.os_ipc_close(p)
+        self._peer_ptr.clear()
+
+    # ------------------------------------------------------------ async ops
+
+    def _owner_of(self, blocks: torch.Tensor) -> torch.Tensor:
+        if not hasattr(self, "_owner_dev") or                 self._owner_dev.device != self.device:
+            self._owner_dev = self.ownership.owner.to(self.device,
+                                                      torch.int64)
+        return self._owner_dev[blocks]
+
+    def _rows_on(self, r: int, keys: torch.Tensor) -> torch.Tensor:
+        blocks = self.part.block_of(keys)
+        return (self._peer_slot[r][blocks] * self.part.block_size
+                + self.part.offset_in_block(keys))
+
+    def pull(self, keys: torch.Tensor) -> torch.Tensor:
+        """Async pull: gather rows straight out of each owner's HBM."""
+        keys = keys.to(self.device, torch.int64)
+        owner = self._owner_of(self.part.block_of(keys))
+        out = torch.empty((keys.shape[0], self.cfg.value_dim),
+                          dtype=self.dtype, device=self.device)
+        for r in range(self.world_size):
+            sel = owner == r
+            if not bool(sel.any()):
+                continue
+            rows = self._rows_on(r, keys[sel])
+            if r == self.rank:
+                out[sel] = self.shard[rows]
+            else:
+                out[sel] = self._hip.os_gather(self._peer_ptr[r], rows,
+                                               self.cfg.value_dim)
+        return out
+
+    def push(self, keys: torch.Tensor, deltas: torch.Tensor) -> None:
+        """Async push: atomicAdd scatter into each owner's HBM."""
+        keys = keys.to(self.device, torch.int64)
+        deltas = deltas.to(self.device, self.dtype).contiguous()
+        owner = self._owner_of(self.part.block_of(keys))
+        for r in range(self.world_size):
+            sel = owner == r
+            if not bool(sel.any()):
+                continue
+            rows = self._rows_on(r, keys[sel])
+            d = deltas[sel].contiguous()
+            if r == self.rank:
+                self.shard.view(-1, self.cfg.value_dim).index_add_(
+                    0, rows, d)
+            else:
+                self._hip.os_scatter_add(self._peer_ptr[r], rows, d)
+
+    def pull_full(self) -> torch.Tensor:
+        """Async full-table pull (dense apps: MLR/Lasso pull everything)."""
+        keys = torch.arange(self.cfg.num_keys, device=self.device)
+        return self.pull(keys)
+
+    def fence(self) -> None:
+        """Make my issued pushes visible device-wide before a clock tick."""
+        torch.cuda.synchronize()

@@ -61,9 +61,11 @@ torch::Tensor os_shard_alloc(int64_t rows, int64_t k) {
   size_t bytes = (size_t)rows * k * sizeof(float);
   TORCH_CHECK(hipMalloc(&p, bytes) == hipSuccess, "hipMalloc failed");
   TORCH_CHECK(hipMemset(p, 0, bytes) == hipSuccess);
+  int dev = 0;
+  TORCH_CHECK(hipGetDevice(&dev) == hipSuccess);
   auto opts = torch::TensorOptions()
                   .dtype(torch::kFloat32)
-                  .device(torch::kCUDA, 0);
+                  .device(torch::kCUDA, dev);
   return torch::from_blob(
       p, {rows, k}, [](void* q) { hipFree(q); }, opts);
 }
