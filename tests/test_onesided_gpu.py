@@ -94,8 +94,9 @@ def _async_mlr_worker(rank, world):
     t.connect()
     _barrier(store, "conn", rank, world)
 
+    gw = torch.Generator().manual_seed(stable_seed("os_mlr", "truth", 0))
+    W_true = torch.randn(C, F, generator=gw)   # SHARED truth across ranks
     g = torch.Generator().manual_seed(stable_seed("os_mlr", "data", rank))
-    W_true = torch.randn(C, F, generator=g)
     X = torch.randn(512, F, generator=g).cuda()
     y = (X @ W_true.t().cuda()).argmax(dim=1)
     keys = torch.arange(C, device="cuda")
