@@ -10,9 +10,13 @@
 // mapping is peer HBM over xGMI (p2p loads/stores, ≈153 GB/s per link);
 // no message, no rendezvous, no collective — a pull is a kernel.
 //
-// Shards for one-sided tables are allocated with hipMalloc directly
-// (NOT the torch caching allocator: IPC handles need the allocation
-// base, and torch's arena offsets break hipIpcGetMemHandle).
+// Shards for one-sided tables are allocated FINE-GRAINED
+// (hipExtMallocWithFlags(hipDeviceMallocFinegrained)) and NOT through the
+// torch caching allocator: IPC handles need the allocation base, and
+// cross-process visibility needs cache-coherent (fine-grained) memory —
+// MI355X L2s are per-XCD, so coarse-grained atomics from one process are
+// not reliably visible to another process's loads until a flush. The
+// scatter atomics are system-scope for the same reason.
 
 #include "hip_common.h"
 
@@ -42,7 +46,8 @@ __global__ void scatter_add_rows_kernel(float* __restrict__ shard,
   const int64_t total = (int64_t)n * k;
   for (int64_t i = t; i < total; i += (int64_t)gridDim.x * blockDim.x) {
     const int64_t r = i / k, c = i - r * k;
-    atomicAdd(&shard[idx[r] * k + c], delta[i]);
+    __hip_atomic_fetch_add(&shard[idx[r] * k + c], delta[i],
+                           __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_SYSTEM);
   }
 }
 
@@ -59,7 +64,8 @@ torch::Tensor os_shard_alloc(int64_t rows, int64_t k) {
   // hipMalloc-backed f32 tensor usable as an IPC export base
   void* p = nullptr;
   size_t bytes = (size_t)rows * k * sizeof(float);
-  TORCH_CHECK(hipMalloc(&p, bytes) == hipSuccess, "hipMalloc failed");
+  TORCH_CHECK(hipExtMallocWithFlags(&p, bytes, hipDeviceMallocFinegrained)
+                  == hipSuccess, "fine-grained hipMalloc failed");
   TORCH_CHECK(hipMemset(p, 0, bytes) == hipSuccess);
   int dev = 0;
   TORCH_CHECK(hipGetDevice(&dev) == hipSuccess);
