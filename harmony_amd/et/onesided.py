@@ -147,11 +147,13 @@ class OneSidedTable(Table):
                 continue
             rows = self._rows_on(r, keys[sel])
             d = deltas[sel].contiguous()
-            if r == self.rank:
-                self.shard.view(-1, self.cfg.value_dim).index_add_(
-                    0, rows, d)
-            else:
-                self._hip.os_scatter_add(self._peer_ptr[r], rows, d)
+            # local pushes use the SAME system-scope atomic kernel as
+            # remote ones: torch index_add_ is a plain read-modify-write,
+            # and racing it against another process's atomics on the same
+            # cells can drop updates
+            ptr = (self.shard.data_ptr() if r == self.rank
+                   else self._peer_ptr[r])
+            self._hip.os_scatter_add(ptr, rows, d)
 
     def pull_full(self) -> torch.Tensor:
         """Async full-table pull (dense apps: MLR/Lasso pull everything)."""
