@@ -90,3 +90,32 @@ class CachedModelAccessor(ETModelAccessor):
             rows = self._cache[keys]
             self._cache[keys] = fn(rows, deltas.to(rows.dtype),
                                    **self.table.cfg.update_args)
+
+
+class OneSidedAccessor:
+    """ETModelAccessor-shaped facade over an et.onesided.OneSidedTable:
+    pulls and pushes are direct xGMI kernels, never collectives, so a
+    worker using it steps at its own pace (true async PS; SSP slack is
+    the only cross-worker coupling). Reference: workers talking to remote
+    tablets through the async op queue (RemoteAccessOpSender)."""
+
+    def __init__(self, table):
+        self.table = table
+        self._keys = None
+
+    def pull_all(self):
+        return self.table.pull_full()
+
+    def pull(self, keys):
+        return self.table.pull(keys)
+
+    def push_dense(self, grad_full):
+        if self._keys is None or self._keys.shape[0] != grad_full.shape[0]:
+            import torch as _t
+
+            self._keys = _t.arange(grad_full.shape[0],
+                                   device=grad_full.device)
+        self.table.push(self._keys, grad_full)
+
+    def push(self, keys, deltas, assume_unique=False):
+        self.table.push(keys, deltas)

@@ -93,7 +93,17 @@ class MLRTrainer(Trainer):
         super().__init__(ctx)
         self.a = defaults(JobConfig(job_id=ctx.job_id, app="mlr",
                                     app_args=ctx.app_args))
-        self.accessor = ETModelAccessor(ctx.table(MODEL_TABLE))
+        table = ctx.table(MODEL_TABLE)
+        from harmony_amd.et.onesided import OneSidedTable
+
+        if isinstance(table, OneSidedTable):
+            # async mode (app arg one_sided=true): pull/push are direct
+            # xGMI kernels; the worker never issues a collective
+            from harmony_amd.dolphin.model_accessor import OneSidedAccessor
+
+            self.accessor = OneSidedAccessor(table)
+        else:
+            self.accessor = ETModelAccessor(table)
         self.step_size = self.a["step_size"]
         self.W = None          # [C*P, F/P] pulled model
         # hipGraph capture of the compute phase (static per block); the
@@ -174,8 +184,20 @@ class MLRTrainer(Trainer):
 
 def build(job: JobConfig, ctx, cp):
     cfg = model_table_cfg(job, ctx.world_size)
-    comm = ctx.new_data_plane()
-    table = Table(cfg, ctx.rank, ctx.world_size, ctx.device, comm=comm)
+    a = defaults(job)
+    if a.get("one_sided"):
+        # async PS mode: hipIpc/xGMI one-sided table — no collectives in
+        # the training loop, SSP slack is the only cross-worker coupling
+        from harmony_amd.et.onesided import OneSidedTable
+
+        table = OneSidedTable(cfg, ctx.rank, ctx.world_size, ctx.device,
+                              store=ctx.store)
+        cp.barrier(f"{job.job_id}/os_alloc", ctx.world_size)
+        table.connect()
+        cp.barrier(f"{job.job_id}/os_conn", ctx.world_size)
+    else:
+        comm = ctx.new_data_plane()
+        table = Table(cfg, ctx.rank, ctx.world_size, ctx.device, comm=comm)
     tctx = TrainerContext(job_id=job.job_id, rank=ctx.rank,
                           world_size=ctx.world_size, device=ctx.device,
                           tables={MODEL_TABLE: table}, app_args=job.app_args)

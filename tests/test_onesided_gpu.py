@@ -122,3 +122,28 @@ def test_async_mlr_training_converges():
     # both ranks trained against the shared async model: far above chance
     for acc in res:
         assert acc > 0.6, res
+
+
+def _async_job_worker(rank, world):
+    """Full run_job MLR in one-sided mode: the training loop issues zero
+    collectives (gloo group only for setup barriers), SSP slack bounds
+    worker skew."""
+    from harmony_amd.config import JobConfig, RuntimeConfig
+    from harmony_amd.dolphin.master import run_job
+    from harmony_amd.runtime.bootstrap import init_executor
+
+    ctx = init_executor(RuntimeConfig(device="cuda", backend="gloo"))
+    job = JobConfig(job_id="os_job", app="mlr", max_num_epochs=3,
+                    num_mini_batches=4, clock_slack=2,
+                    app_args={"num_classes": 5, "num_features": 256,
+                              "num_parts_per_class": 4, "batch_size": 512,
+                              "step_size": 0.5, "one_sided": True})
+    s = run_job(job, ctx).summary()
+    return (s["num_batches"], s["accuracy"])
+
+
+def test_async_mlr_job_one_sided():
+    res = run_dist(_async_job_worker, world=2, timeout=300)
+    for n, acc in res:
+        assert n == 12
+        assert acc > 0.5, res
