@@ -33,7 +33,10 @@ __global__ void gather_rows_kernel(const float* __restrict__ shard,
   const int64_t total = (int64_t)n * k;
   for (int64_t i = t; i < total; i += (int64_t)gridDim.x * blockDim.x) {
     const int64_t r = i / k, c = i - r * k;
-    out[i] = shard[idx[r] * k + c];
+    // system-scope load: another process's completed atomics must be
+    // visible even if this process cached the line on an earlier pull
+    out[i] = __hip_atomic_load(&shard[idx[r] * k + c], __ATOMIC_RELAXED,
+                               __HIP_MEMORY_SCOPE_SYSTEM);
   }
 }
 
