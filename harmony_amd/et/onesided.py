@@ -129,11 +129,12 @@ class OneSidedTable(Table):
             if not bool(sel.any()):
                 continue
             rows = self._rows_on(r, keys[sel])
-            if r == self.rank:
-                out[sel] = self.shard[rows]
-            else:
-                out[sel] = self._hip.os_gather(self._peer_ptr[r], rows,
-                                               self.cfg.value_dim)
+            # local reads go through the same system-scope gather kernel:
+            # a plain torch read can hit a line this process cached before
+            # a peer's system-scope atomic landed
+            ptr = (self.shard.data_ptr() if r == self.rank
+                   else self._peer_ptr[r])
+            out[sel] = self._hip.os_gather(ptr, rows, self.cfg.value_dim)
         return out
 
     def push(self, keys: torch.Tensor, deltas: torch.Tensor) -> None:
