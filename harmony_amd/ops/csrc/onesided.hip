@@ -67,8 +67,11 @@ torch::Tensor os_shard_alloc(int64_t rows, int64_t k) {
   // hipMalloc-backed f32 tensor usable as an IPC export base
   void* p = nullptr;
   size_t bytes = (size_t)rows * k * sizeof(float);
-  TORCH_CHECK(hipExtMallocWithFlags(&p, bytes, hipDeviceMallocFinegrained)
-                  == hipSuccess, "fine-grained hipMalloc failed");
+  // uncached (MTYPE_UC): no XCD-L2 lines exist for shard memory, so a
+  // plain write from one process can never linger dirty and later stomp
+  // a peer's committed atomic — fine-grained alone still flaked (~1/5)
+  TORCH_CHECK(hipExtMallocWithFlags(&p, bytes, hipDeviceMallocUncached)
+                  == hipSuccess, "uncached hipMalloc failed");
   TORCH_CHECK(hipMemset(p, 0, bytes) == hipSuccess);
   int dev = 0;
   TORCH_CHECK(hipGetDevice(&dev) == hipSuccess);
