@@ -65,16 +65,21 @@ def _async_worker(rank, world):
     full = t.pull_full()
     torch.cuda.synchronize()
     ok = bool((full == 15.0).all())
-    vals = full.unique().tolist()
+    bad = (full != 15.0).nonzero()
+    diag = []
+    for i in range(min(8, bad.shape[0])):
+        r_, c_ = int(bad[i, 0]), int(bad[i, 1])
+        owner = 0 if r_ < 32 else 1     # 8 blocks of 8 keys, even split
+        diag.append((r_, c_, float(full[r_, c_]), owner))
     _barrier(store, "checked", rank, world)
     t.close()
-    return (ok, vals)
+    return (ok, full.unique().tolist(), diag)
 
 
 def test_async_uneven_push_pull_two_procs_conserved():
     res = run_dist(_async_worker, world=2, timeout=300)
-    for ok, vals in res:
-        assert ok, f"lost/duplicated pushes: values {vals}"
+    for ok, vals, diag in res:
+        assert ok, f"values {vals}; (row, col, val, owner) {diag}"
 
 
 def _async_mlr_worker(rank, world):
