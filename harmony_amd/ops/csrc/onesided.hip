@@ -40,6 +40,25 @@ __global__ void gather_rows_kernel(const float* __restrict__ shard,
   }
 }
 
+// fp32 add via integer CAS at system scope: CDNA hardware silently DROPS
+// unsupported fp atomics on some memory types (the gfx90a-era
+// unsafe-fp-atomics behavior; box-dependent via XNACK/page setup) —
+// integer compare-exchange is architecturally guaranteed everywhere.
+__device__ __forceinline__ void sys_atomic_add_f32(float* addr, float v) {
+  unsigned int* a = reinterpret_cast<unsigned int*>(addr);
+  unsigned int old = __hip_atomic_load(a, __ATOMIC_RELAXED,
+                                       __HIP_MEMORY_SCOPE_SYSTEM);
+  while (true) {
+    unsigned int assumed = old;
+    const float f = __uint_as_float(assumed) + v;
+    const unsigned int want = __float_as_uint(f);
+    if (__hip_atomic_compare_exchange_strong(
+            a, &old, want, __ATOMIC_RELAXED, __ATOMIC_RELAXED,
+            __HIP_MEMORY_SCOPE_SYSTEM))
+      break;
+  }
+}
+
 // atomic add into a (possibly remote) shard: shard[idx[i]] += delta[i]
 __global__ void scatter_add_rows_kernel(float* __restrict__ shard,
                                         const int64_t* __restrict__ idx,
@@ -49,8 +68,7 @@ __global__ void scatter_add_rows_kernel(float* __restrict__ shard,
   const int64_t total = (int64_t)n * k;
   for (int64_t i = t; i < total; i += (int64_t)gridDim.x * blockDim.x) {
     const int64_t r = i / k, c = i - r * k;
-    __hip_atomic_fetch_add(&shard[idx[r] * k + c], delta[i],
-                           __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_SYSTEM);
+    sys_atomic_add_f32(&shard[idx[r] * k + c], delta[i]);
   }
 }
 
