@@ -185,7 +185,7 @@ class MLRTrainer(Trainer):
 def build(job: JobConfig, ctx, cp):
     cfg = model_table_cfg(job, ctx.world_size)
     a = defaults(job)
-    if a.get("one_sided"):
+    if str(a.get("one_sided", "")).lower() in ("true", "1"):
         # async PS mode: hipIpc/xGMI one-sided table — no collectives in
         # the training loop, SSP slack is the only cross-worker coupling
         from harmony_amd.et.onesided import OneSidedTable
