@@ -85,7 +85,15 @@ class LassoTrainer(Trainer):
         super().__init__(ctx)
         self.a = defaults(JobConfig(job_id=ctx.job_id, app="lasso",
                                     app_args=ctx.app_args))
-        self.accessor = ETModelAccessor(ctx.table(MODEL_TABLE))
+        table = ctx.table(MODEL_TABLE)
+        from harmony_amd.et.onesided import OneSidedTable
+
+        if isinstance(table, OneSidedTable):
+            from harmony_amd.dolphin.model_accessor import OneSidedAccessor
+
+            self.accessor = OneSidedAccessor(table)
+        else:
+            self.accessor = ETModelAccessor(table)
         self._loss = torch.zeros(())
 
     def pull_model(self) -> None:
@@ -120,8 +128,18 @@ class LassoTrainer(Trainer):
 
 def build(job: JobConfig, ctx, cp):
     cfg = model_table_cfg(job, ctx.world_size)
-    comm = ctx.new_data_plane()
-    table = Table(cfg, ctx.rank, ctx.world_size, ctx.device, comm=comm)
+    a = defaults(job)
+    if str(a.get("one_sided", "")).lower() in ("true", "1"):
+        from harmony_amd.et.onesided import OneSidedTable
+
+        table = OneSidedTable(cfg, ctx.rank, ctx.world_size, ctx.device,
+                              store=ctx.store)
+        cp.barrier(f"{job.job_id}/os_alloc", ctx.world_size)
+        table.connect()
+        cp.barrier(f"{job.job_id}/os_conn", ctx.world_size)
+    else:
+        comm = ctx.new_data_plane()
+        table = Table(cfg, ctx.rank, ctx.world_size, ctx.device, comm=comm)
     blocks, _ = make_batches(job, ctx.rank, ctx.device, ctx.world_size)
     tctx = TrainerContext(job_id=job.job_id, rank=ctx.rank,
                           world_size=ctx.world_size, device=ctx.device,

@@ -287,3 +287,18 @@ def test_dashboard_jobs_index_and_filter(tmp_path):
         assert len(only_b) == 1 and only_b[0]["job_id"] == "jobB"
     finally:
         srv.stop()
+
+
+def test_one_sided_requires_gpu_clear_error():
+    import pytest as _pt
+    import torch
+
+    from harmony_amd.config import TableConfig
+    from harmony_amd.et.onesided import OneSidedTable
+
+    if torch.cuda.is_available():
+        _pt.skip("CPU-only check")
+    cfg = TableConfig(table_id="x", num_keys=8, value_dim=2, num_blocks=2,
+                      update_fn="add", init_fn="zeros")
+    with _pt.raises(RuntimeError, match="HIP extension"):
+        OneSidedTable(cfg, 0, 1, torch.device("cpu"))

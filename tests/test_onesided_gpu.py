@@ -152,3 +152,25 @@ def test_async_mlr_job_one_sided():
     for n, acc in res:
         assert n == 12
         assert acc > 0.5, res
+
+
+def _async_lasso_worker(rank, world):
+    from harmony_amd.config import JobConfig, RuntimeConfig
+    from harmony_amd.dolphin.master import run_job
+    from harmony_amd.runtime.bootstrap import init_executor
+
+    ctx = init_executor(RuntimeConfig(device="cuda", backend="gloo"))
+    job = JobConfig(job_id="os_lasso", app="lasso", max_num_epochs=2,
+                    num_mini_batches=2,
+                    app_args={"num_features": 64, "num_parts": 8,
+                              "batch_size": 512, "lam": 0.02,
+                              "one_sided": True})
+    s = run_job(job, ctx).summary()
+    return (s["num_batches"], s["mse"])
+
+
+def test_async_lasso_job_one_sided():
+    res = run_dist(_async_lasso_worker, world=2, timeout=300)
+    for n, mse in res:
+        assert n == 4
+        assert mse < 2.0, res
