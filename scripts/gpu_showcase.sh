@@ -21,6 +21,12 @@ timeout $T bin/submit_pagerank.sh -port 7311 -job_id sc_pr -num_vertices 100000 
 echo
 timeout $T bin/submit_shortest_path.sh -port 7311 -job_id sc_sp -num_vertices 100000 -out_degree 4 --wait | head -c 400
 echo
+# async one-sided mode (hipIpc/xGMI data plane, no collectives)
+timeout $T bin/submit_mlr.sh -port 7311 -job_id sc_async -max_num_epochs 2 -num_mini_batches 4 -num_classes 5 -num_features 2048 -num_parts_per_class 8 -batch_size 2048 -step_size 0.5 -one_sided true --wait | head -c 300
+echo
+# restore-from-checkpoint (reference createTable(chkpId))
+timeout $T bin/submit_mlr.sh -port 7311 -job_id sc_restore -max_num_epochs 1 -num_mini_batches 4 -num_classes 10 -num_features 4096 -num_parts_per_class 8 -batch_size 4096 -restore_chkp sc_mlr/epoch2 --wait | head -c 300
+echo
 # elastic: addvector with the scripted rotating optimizer + validation
 timeout $T python -m harmony_amd.jobserver.client submit -port 7311 -app addvector -job_id sc_elastic -max_num_epochs 4 -num_mini_batches 6 -optimizer homogeneous -optimizer_period 4 --wait 2>/dev/null | head -c 200
 echo
