@@ -89,8 +89,9 @@ __global__ void alias_build_kernel(const int* __restrict__ word_topic,
   int* ta = (int*)(tp + WAVE);
   int* tl = ta + WAVE;
   const int S = K / WAVE;                  // entries per lane segment
-  for (int row = blockIdx.x * BUILD_WAVES + wave; row < rows;
-       row += gridDim.x * BUILD_WAVES) {
+  const int waves_wg = blockDim.x / WAVE;
+  for (int row = blockIdx.x * waves_wg + wave; row < rows;
+       row += gridDim.x * waves_wg) {
     const int64_t base = (int64_t)row * K;
     // lane's contiguous segment [lane*S, (lane+1)*S)
     float seg_mass = 0.f;
@@ -276,9 +277,12 @@ std::vector<torch::Tensor> lda_alias_build(torch::Tensor word_topic,
                          word_topic.options().dtype(torch::kFloat32));
   auto qsum = torch::empty({rows}, prob.options());
   if (rows > 0) {
-    dim3 blk(WAVE * BUILD_WAVES);
-    dim3 grid(std::min((rows + BUILD_WAVES - 1) / BUILD_WAVES, 8192));
-    const size_t shmem = (size_t)BUILD_WAVES * (3 * K + 3 * WAVE) * 4;
+    int waves = BUILD_WAVES;       // HARMONY_LDA_BUILD_WAVES: occupancy A/B
+    const char* bw = getenv("HARMONY_LDA_BUILD_WAVES");
+    if (bw) { int w = atoi(bw); if (w >= 1 && w <= 8) waves = w; }
+    dim3 blk(WAVE * waves);
+    dim3 grid(std::min((rows + waves - 1) / waves, 8192));
+    const size_t shmem = (size_t)waves * (3 * K + 3 * WAVE) * 4;
     hipLaunchKernelGGL(alias_build_kernel, grid, blk, shmem, current_stream(),
                        word_topic.data_ptr<int>(), invden.data_ptr<float>(),
                        (float)beta, prob.data_ptr<float>(),

@@ -33,6 +33,22 @@ def main():
     wl64 = wl.to(dev)
     offs = torch.arange(0, (D + 1) * T, T, dtype=torch.int64, device=dev)
     z = z.to(dev)
+    import os as _os
+
+    for bw in ("1", "2", "4", "8"):
+        _os.environ["HARMONY_LDA_BUILD_WAVES"] = bw
+        for _ in range(2):
+            hip.lda_alias_build(wt, nk, 0.01, V)
+        torch.cuda.synchronize()
+        ts = []
+        for _ in range(10):
+            t0 = time.perf_counter()
+            hip.lda_alias_build(wt, nk, 0.01, V)
+            torch.cuda.synchronize()
+            ts.append(time.perf_counter() - t0)
+        ts.sort()
+        print(f"alias_build waves={bw}: {ts[len(ts)//2]*1e3:.3f} ms")
+    _os.environ.pop("HARMONY_LDA_BUILD_WAVES", None)
     tabs = hip.lda_alias_build(wt, nk, 0.01, V)
     prob, alias, tp, ta, qv, qsum, invden = tabs
     torch.cuda.synchronize()
