@@ -147,3 +147,40 @@ def test_priority_and_gang_schedulers():
     assert gang.on_job_arrival(b, pool) is None
     c = JobConfig(job_id="c", app="mlr", app_args={"num_executors": 2})
     assert gang.on_job_arrival(c, pool) == [2, 3]
+
+
+def test_jobserver_error_paths():
+    # bad app name -> job reports failed, server survives and runs the
+    # next good job; STATUS lists world size; WAIT on unknown id times out
+    from harmony_amd.config import JobConfig, RuntimeConfig
+    from harmony_amd.jobserver import client
+    from harmony_amd.jobserver.server import JobServerDriver
+    from harmony_amd.runtime.bootstrap import init_executor
+
+    port = free_port()
+    ctx = init_executor(RuntimeConfig(device="cpu"))
+    driver = JobServerDriver(ctx, scheduler="default", port=port)
+    t = threading.Thread(target=driver.run, daemon=True)
+    t.start()
+    time.sleep(0.3)
+
+    bad = JobConfig(job_id="jsbad", app="no_such_app", max_num_epochs=1,
+                    num_mini_batches=1)
+    res = client.submit(bad, port=port, wait=True, timeout=60)
+    assert res["status"] == "failed", res
+    assert "no_such_app" in str(res)
+
+    st = client._send({"cmd": "STATUS"}, port)
+    assert st["world_size"] == 1
+
+    r = client._send({"cmd": "WAIT", "job_id": "never_submitted",
+                      "timeout": 1}, port)
+    assert r["status"] == "timeout"
+
+    good = JobConfig(job_id="jsok", app="addvector", max_num_epochs=1,
+                     num_mini_batches=2,
+                     app_args={"num_keys": 8, "vector_dim": 2})
+    res = client.submit(good, port=port, wait=True, timeout=60)
+    assert res["status"] == "done", res
+    client.shutdown(port=port)
+    t.join(timeout=30)
