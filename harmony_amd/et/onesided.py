@@ -164,3 +164,28 @@ class OneSidedTable(Table):
     def fence(self) -> None:
         """Make my issued pushes visible device-wide before a clock tick."""
         torch.cuda.synchronize()
+
+    # -------------------------------------------- inherited-API guard rails
+
+    def update(self, keys, deltas, assume_unique: bool = False) -> None:
+        # 'add' algebra == async push; other update fns were rejected at
+        # construction, so this is always safe
+        self.push(keys, deltas)
+
+    def push_dense(self, grad_full) -> None:
+        self.push(torch.arange(self.cfg.num_keys, device=self.device),
+                  grad_full)
+
+    def get(self, keys):
+        return self.pull(keys)
+
+    def pull_all(self):
+        return self.pull_full()
+
+    def drop_blocks(self, blocks) -> None:
+        raise RuntimeError("one-sided tables are statically owned (v1): "
+                           "peer IPC mappings pin the shard layout")
+
+    def adopt_blocks(self, blocks) -> None:
+        raise RuntimeError("one-sided tables are statically owned (v1): "
+                           "peer IPC mappings pin the shard layout")
