@@ -178,7 +178,10 @@ def _share_dist_worker(rank, world):
     ctx = init_executor(RuntimeConfig(device="cpu"))
     # mean share = 2 -> rank 0 serves half batches, rank 1 clamps at full
     opt = SampleOptimizers.batch_shares(((0, 1), (1, 3)))
-    job = JobConfig(job_id="j_share_d", app="mlr", max_num_epochs=3,
+    # 5 epochs x 2 batches: a late first metric report can slip the
+    # decision one period (plan applies at batch 6 at the latest), so the
+    # job must outlive that worst case for the share to show in the tail
+    job = JobConfig(job_id="j_share_d", app="mlr", max_num_epochs=5,
                     num_mini_batches=2, optimizer_period=2,
                     app_args={"num_classes": 4, "num_features": 32,
                               "num_parts_per_class": 2, "batch_size": 100,
