@@ -51,6 +51,9 @@ class OneSidedTable(Table):
         super().__init__(cfg, rank, world_size, device)
         self.store = store
         self._peer_ptr = {}          # rank -> mapped device pointer (int)
+        # per-table op stats (reference RemoteAccessOpStat): pulled rows /
+        # pushed rows / remote bytes moved over xGMI
+        self.stats = {"pull_rows": 0, "push_rows": 0, "remote_bytes": 0}
         # every rank's block->slot map is derivable from the static
         # round-robin ownership, so remote row indices need no exchange
         self._peer_slot = {}
@@ -121,6 +124,7 @@ class OneSidedTable(Table):
     def pull(self, keys: torch.Tensor) -> torch.Tensor:
         """Async pull: gather rows straight out of each owner's HBM."""
         keys = keys.to(self.device, torch.int64)
+        self.stats["pull_rows"] += keys.shape[0]
         owner = self._owner_of(self.part.block_of(keys))
         out = torch.empty((keys.shape[0], self.cfg.value_dim),
                           dtype=self.dtype, device=self.device)
@@ -129,6 +133,9 @@ class OneSidedTable(Table):
             if not bool(sel.any()):
                 continue
             rows = self._rows_on(r, keys[sel])
+            if r != self.rank:
+                self.stats["remote_bytes"] += (rows.shape[0]
+                                               * self.cfg.value_dim * 4)
             # local reads go through the same system-scope gather kernel:
             # a plain torch read can hit a line this process cached before
             # a peer's system-scope atomic landed
@@ -140,6 +147,7 @@ class OneSidedTable(Table):
     def push(self, keys: torch.Tensor, deltas: torch.Tensor) -> None:
         """Async push: atomicAdd scatter into each owner's HBM."""
         keys = keys.to(self.device, torch.int64)
+        self.stats["push_rows"] += keys.shape[0]
         deltas = deltas.to(self.device, self.dtype).contiguous()
         owner = self._owner_of(self.part.block_of(keys))
         for r in range(self.world_size):
@@ -152,6 +160,9 @@ class OneSidedTable(Table):
             # remote ones: torch index_add_ is a plain read-modify-write,
             # and racing it against another process's atomics on the same
             # cells can drop updates
+            if r != self.rank:
+                self.stats["remote_bytes"] += (rows.shape[0]
+                                               * self.cfg.value_dim * 4)
             ptr = (self.shard.data_ptr() if r == self.rank
                    else self._peer_ptr[r])
             self._hip.os_scatter_add(ptr, rows, d)
