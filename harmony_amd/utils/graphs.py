@@ -63,7 +63,10 @@ class GraphRunner:
                     body()
             torch.cuda.current_stream().wait_stream(self._stream)
             g = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(g):
+            # thread_local: co-located jobs' threads keep allocating while
+            # this one captures — the default "global" error mode hard-
+            # aborts the whole process on any cross-thread CUDA malloc
+            with torch.cuda.graph(g, capture_error_mode="thread_local"):
                 body()
             for t, s in zip(state, saved):
                 t.copy_(s)
