@@ -210,6 +210,7 @@ def main():
             "steps": args.steps,
             "warmup": args.warmup,
             "ms_per_step": elapsed / args.steps * 1000.0,
+            "makespan_sec": elapsed,   # BASELINE metric: + makespan
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
