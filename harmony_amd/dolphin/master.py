@@ -124,7 +124,14 @@ def run_job(job: JobConfig, ctx: ExecutorContext,
     metrics = tasklet.run()
     if tracer is not None:
         tracer.flush()
-    ev = trainer.evaluate_model()
+    # evaluate_model may issue collectives (e.g. LDA's log-likelihood does
+    # a pull_all) — under co-located jobs EVERY collective needs a global
+    # NET ticket or two jobs' post-run evals can interleave on the wire.
+    # All ranks reach here after the job's cleanup barrier, so a fixed
+    # out-of-band phase index is requested in the same order everywhere.
+    _EVAL_PHASE = 1 << 30
+    with tasklet.tus.net(job.job_id, _EVAL_PHASE):
+        ev = trainer.evaluate_model()
     for k, v in (ev or {}).items():
         metrics.add_custom(k, float(v))
     if job.offline_model_eval and chkp_mgr is not None:
@@ -132,7 +139,9 @@ def run_job(job: JobConfig, ctx: ExecutorContext,
 
         evaluator = ModelEvaluator(chkp_mgr.cm, job.job_id, tables, trainer,
                                    provider)
-        for cid, res in evaluator.evaluate_all(chkp_mgr.chkp_ids).items():
+        with tasklet.tus.net(job.job_id, _EVAL_PHASE + 1):
+            offline = evaluator.evaluate_all(chkp_mgr.chkp_ids)
+        for cid, res in offline.items():
             for k, v in (res or {}).items():
                 metrics.add_custom(f"offline/{cid}/{k}", float(v))
     return metrics

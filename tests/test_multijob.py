@@ -55,3 +55,62 @@ def test_two_concurrent_jobs_two_ranks():
     for r in res:
         assert r["mj_mlr"] == 6
         assert r["mj_nmf"] == 4
+
+
+def _three_jobs_worker(rank, world):
+    import random
+    import threading
+    import time as _time
+
+    from harmony_amd.config import JobConfig, RuntimeConfig
+    from harmony_amd.dolphin.master import run_job
+    from harmony_amd.runtime.bootstrap import init_executor
+    from harmony_amd.runtime.control import ControlPlane, TaskUnitScheduler
+
+    ctx = init_executor(RuntimeConfig(device="cpu"))
+    cp = ControlPlane(ctx.store, ctx.rank, ctx.world_size)
+    jobs = [
+        JobConfig(job_id="s3_mlr", app="mlr", max_num_epochs=3,
+                  num_mini_batches=3,
+                  app_args={"num_classes": 3, "num_features": 16,
+                            "num_parts_per_class": 2, "batch_size": 32}),
+        JobConfig(job_id="s3_nmf", app="nmf", max_num_epochs=2,
+                  num_mini_batches=4,
+                  app_args={"num_cols": 64, "rank": 8, "nnz_per_row": 4,
+                            "rows_per_batch": 32}),
+        JobConfig(job_id="s3_lda", app="lda", max_num_epochs=2,
+                  num_mini_batches=2,
+                  app_args={"num_vocabs": 300, "num_topics": 8,
+                            "tokens_per_doc": 12, "docs_per_batch": 32}),
+    ]
+    tus = TaskUnitScheduler(cp, {j.job_id for j in jobs}, multi_job=True)
+    errs, results = [], {}
+
+    def run_one(job, jitter):
+        try:
+            # stagger starts randomly: ticket ordering must hold whatever
+            # the interleaving
+            _time.sleep(jitter)
+            results[job.job_id] = run_job(job, ctx, cp=cp, tus=tus).summary()
+        except Exception:  # noqa: BLE001
+            import traceback
+
+            errs.append(traceback.format_exc())
+
+    rng = random.Random(rank * 7 + 1)
+    ts = [threading.Thread(target=run_one, args=(j, rng.random() * 0.3))
+          for j in jobs]
+    for t in ts:
+        t.start()
+    for t in ts:
+        t.join(timeout=180)
+    assert not errs, errs[0]
+    return {j: s["num_batches"] for j, s in results.items()}
+
+
+def test_three_concurrent_jobs_three_ranks_staggered():
+    from tests.dist_helper import run_dist
+
+    res = run_dist(_three_jobs_worker, world=3, timeout=300)
+    for r in res:
+        assert r == {"s3_mlr": 9, "s3_nmf": 8, "s3_lda": 4}
