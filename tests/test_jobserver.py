@@ -184,3 +184,55 @@ def test_jobserver_error_paths():
     assert res["status"] == "done", res
     client.shutdown(port=port)
     t.join(timeout=30)
+
+
+def _jobserver_concurrent_worker(rank, world):
+    import os
+
+    from harmony_amd.config import JobConfig, RuntimeConfig
+    from harmony_amd.jobserver import client
+    from harmony_amd.jobserver.server import JobServerDriver
+    from harmony_amd.runtime.bootstrap import init_executor
+
+    port = 7600 + int(os.environ["MASTER_PORT"]) % 300
+    ctx = init_executor(RuntimeConfig(device="cpu"))
+    driver = JobServerDriver(ctx, scheduler="default", port=port)
+
+    if rank == 0:
+        def submit_pair():
+            time.sleep(0.5)
+            jobs = [
+                JobConfig(job_id="cc_mlr", app="mlr", max_num_epochs=2,
+                          num_mini_batches=2,
+                          app_args={"num_classes": 3, "num_features": 16,
+                                    "num_parts_per_class": 2,
+                                    "batch_size": 32}),
+                JobConfig(job_id="cc_lda", app="lda", max_num_epochs=2,
+                          num_mini_batches=2,
+                          app_args={"num_vocabs": 200, "num_topics": 8,
+                                    "tokens_per_doc": 10,
+                                    "docs_per_batch": 16}),
+            ]
+            # submit both WITHOUT waiting -> they co-run on both executors
+            for j in jobs:
+                client.submit(j, port=port)
+            for j in jobs:
+                r = client._send({"cmd": "WAIT", "job_id": j.job_id,
+                                  "timeout": 120}, port)
+                assert r["status"] == "done", (j.job_id, r)
+                assert all(s["num_batches"] == 4 for s in r["per_rank"])
+            client.shutdown(port=port)
+
+        threading.Thread(target=submit_pair, daemon=True).start()
+    driver.run()
+    return True
+
+
+def test_jobserver_concurrent_jobs_two_ranks():
+    # two jobs co-scheduled through the jobserver on 2 executors: the
+    # admin tickets + per-job NET ordering + the post-run LDA collective
+    # eval must all serialize correctly
+    from tests.dist_helper import run_dist
+
+    res = run_dist(_jobserver_concurrent_worker, world=2, timeout=240)
+    assert res == [True, True]
