@@ -58,9 +58,12 @@ def test_two_concurrent_jobs_two_ranks():
 
 
 def _three_jobs_worker(rank, world):
+    import os
     import random
     import threading
     import time as _time
+
+    os.environ["HARMONY_SANITIZE"] = "1"     # record + validate NET order
 
     from harmony_amd.config import JobConfig, RuntimeConfig
     from harmony_amd.dolphin.master import run_job
@@ -105,6 +108,14 @@ def _three_jobs_worker(rank, world):
     for t in ts:
         t.join(timeout=180)
     assert not errs, errs[0]
+    import torch.distributed as dist
+
+    dist.barrier()
+    if rank == 0:
+        from harmony_amd.utils import sanitize
+
+        violations = sanitize.validate(ctx.store, world)
+        assert violations == [], violations
     return {j: s["num_batches"] for j, s in results.items()}
 
 
