@@ -100,8 +100,11 @@ class OneSidedAccessor:
     tablets through the async op queue (RemoteAccessOpSender)."""
 
     def __init__(self, table):
+        import collections
+
         self.table = table
         self._keys = None
+        self.metrics = collections.defaultdict(float)
 
     def pull_all(self):
         return self.table.pull_full()

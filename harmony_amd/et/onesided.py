@@ -44,9 +44,14 @@ class OneSidedTable(Table):
 
     def __init__(self, cfg: TableConfig, rank: int, world_size: int,
                  device: torch.device, store=None):
-        assert cfg.update_fn == "add", \
-            "one-sided push is atomicAdd: update_fn must be 'add'"
-        assert cfg.dtype == "float32", "one-sided tables are float32 (v1)"
+        # update fns with pure-add merge algebra work one-sided:
+        #   add        — atomic float/int add IS the update
+        #   lda_counts — ±1 count deltas; the reference's clamp>=0 is a
+        #                defensive no-op (counts conserved by construction,
+        #                docs/ROADMAP.md), so atomic int add is exact
+        assert cfg.update_fn in ("add", "lda_counts"), \
+            "one-sided push is atomic add: update_fn must be add-algebra"
+        assert cfg.dtype in ("float32", "int32")
         self._hip = _require_hip()
         super().__init__(cfg, rank, world_size, device)
         self.store = store
@@ -71,7 +76,8 @@ class OneSidedTable(Table):
         owned = self.ownership.owned_blocks(self.rank)
         self._local_blocks = list(owned)
         rows = len(owned) * self.part.block_size
-        self.shard = self._hip.os_shard_alloc(rows, self.cfg.value_dim)
+        self.shard = self._hip.os_shard_alloc(
+            rows, self.cfg.value_dim, 1 if self.cfg.dtype == "int32" else 0)
         from harmony_amd.et import update_functions as uf
         from harmony_amd.et.table import _block_seed
 
@@ -141,7 +147,9 @@ class OneSidedTable(Table):
             # a peer's system-scope atomic landed
             ptr = (self.shard.data_ptr() if r == self.rank
                    else self._peer_ptr[r])
-            out[sel] = self._hip.os_gather(ptr, rows, self.cfg.value_dim)
+            out[sel] = self._hip.os_gather(
+                ptr, rows, self.cfg.value_dim,
+                1 if self.cfg.dtype == "int32" else 0)
         return out
 
     def push(self, keys: torch.Tensor, deltas: torch.Tensor) -> None:

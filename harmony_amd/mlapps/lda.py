@@ -130,7 +130,17 @@ class LDATrainer(Trainer):
         super().__init__(ctx)
         self.a = defaults(JobConfig(job_id=ctx.job_id, app="lda",
                                     app_args=ctx.app_args))
-        self.accessor = ETModelAccessor(ctx.table(MODEL_TABLE))
+        table = ctx.table(MODEL_TABLE)
+        from harmony_amd.et.onesided import OneSidedTable
+
+        if isinstance(table, OneSidedTable):
+            # async mode: pulls/pushes are direct xGMI kernels (atomic int
+            # adds); LDA's count algebra is pure add (clamp is a no-op)
+            from harmony_amd.dolphin.model_accessor import OneSidedAccessor
+
+            self.accessor = OneSidedAccessor(table)
+        else:
+            self.accessor = ETModelAccessor(table)
         K = self.a["num_topics"]
         self.doc_topic = torch.zeros(num_local_docs, K, dtype=torch.int32,
                                      device=ctx.device)
@@ -230,7 +240,30 @@ class LDATrainer(Trainer):
         table = self.accessor.table
         b = self.batch
         old, z = self._old_z, self._new_z
-        if table.comm is None or table.world_size == 1:
+        from harmony_amd.et.onesided import OneSidedTable
+
+        if isinstance(table, OneSidedTable):
+            # async path: dense ±1 delta rows for CHANGED words, one atomic
+            # scatter into whichever rank owns each word (denser on the
+            # wire than the 12 B pair format, but requires no rendezvous)
+            K = self.a["num_topics"]
+            changed = (z != old).nonzero(as_tuple=True)[0]
+            if changed.numel():
+                pw = b.word_ids[changed]
+                uniqw, inv = torch.unique(pw, return_inverse=True)
+                delta = torch.zeros(uniqw.shape[0], K, dtype=torch.int32,
+                                    device=z.device)
+                ones = torch.ones_like(changed, dtype=torch.int32)
+                delta.view(-1).scatter_add_(
+                    0, inv * K + z[changed].long(), ones)
+                delta.view(-1).scatter_add_(
+                    0, inv * K + old[changed].long(), -ones)
+                summ = (torch.bincount(z[changed].long(), minlength=K)
+                        - torch.bincount(old[changed].long(), minlength=K)
+                        ).to(torch.int32)
+                table.push(torch.cat([uniqw, b.pull_keys[-1:]]),
+                           torch.cat([delta, summ.unsqueeze(0)]))
+        elif table.comm is None or table.world_size == 1:
             # single-owner path: ONE fused kernel applies every changed
             # token's ±1 to its word row AND the summary row — no host sync
             # (the distributed path's nonzero() is only needed to size the
@@ -298,8 +331,17 @@ class LDATrainer(Trainer):
 def build(job: JobConfig, ctx, cp):
     a = defaults(job)
     cfg = model_table_cfg(job, ctx.world_size)
-    comm = ctx.new_data_plane()
-    table = Table(cfg, ctx.rank, ctx.world_size, ctx.device, comm=comm)
+    if str(a.get("one_sided", "")).lower() in ("true", "1"):
+        from harmony_amd.et.onesided import OneSidedTable
+
+        table = OneSidedTable(cfg, ctx.rank, ctx.world_size, ctx.device,
+                              store=ctx.store)
+        cp.barrier(f"{job.job_id}/os_alloc", ctx.world_size)
+        table.connect()
+        cp.barrier(f"{job.job_id}/os_conn", ctx.world_size)
+    else:
+        comm = ctx.new_data_plane()
+        table = Table(cfg, ctx.rank, ctx.world_size, ctx.device, comm=comm)
     blocks, local_docs = make_batches(job, ctx.rank, ctx.device,
                                       ctx.world_size)
     tctx = TrainerContext(job_id=job.job_id, rank=ctx.rank,

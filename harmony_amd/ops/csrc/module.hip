@@ -38,11 +38,12 @@ torch::Tensor lda_mh(torch::Tensor doc_topic, torch::Tensor word_topic,
                      double alpha, double beta, int64_t seed);
 void lasso_cd(torch::Tensor Xt, torch::Tensor r, torch::Tensor w,
               torch::Tensor col_sq, double lam_n);
-torch::Tensor os_shard_alloc(int64_t rows, int64_t k);
+torch::Tensor os_shard_alloc(int64_t rows, int64_t k, int64_t dtype_i32);
 torch::Tensor os_ipc_handle(torch::Tensor shard);
 int64_t os_ipc_open(torch::Tensor handle_bytes);
 void os_ipc_close(int64_t ptr);
-torch::Tensor os_gather(int64_t ptr, torch::Tensor idx, int64_t k);
+torch::Tensor os_gather(int64_t ptr, torch::Tensor idx, int64_t k,
+                        int64_t dtype_i32);
 void os_scatter_add(int64_t ptr, torch::Tensor idx, torch::Tensor delta);
 void gbt_hist(torch::Tensor bins, torch::Tensor resid, torch::Tensor node,
               torch::Tensor cnt, torch::Tensor sum);

@@ -72,6 +72,36 @@ __global__ void scatter_add_rows_kernel(float* __restrict__ shard,
   }
 }
 
+// int32 variants (LDA count tables): integer atomics are natively
+// supported at system scope — no CAS loop needed
+__global__ void gather_rows_i32_kernel(const int* __restrict__ shard,
+                                       const int64_t* __restrict__ idx,
+                                       int* __restrict__ out,
+                                       int n, int k) {
+  const int64_t t = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t total = (int64_t)n * k;
+  for (int64_t i = t; i < total; i += (int64_t)gridDim.x * blockDim.x) {
+    const int64_t r = i / k, c = i - r * k;
+    out[i] = __hip_atomic_load(&shard[idx[r] * k + c], __ATOMIC_RELAXED,
+                               __HIP_MEMORY_SCOPE_SYSTEM);
+  }
+}
+
+__global__ void scatter_add_rows_i32_kernel(int* __restrict__ shard,
+                                            const int64_t* __restrict__ idx,
+                                            const int* __restrict__ delta,
+                                            int n, int k) {
+  const int64_t t = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t total = (int64_t)n * k;
+  for (int64_t i = t; i < total; i += (int64_t)gridDim.x * blockDim.x) {
+    const int64_t r = i / k, c = i - r * k;
+    const int d = delta[i];
+    if (d != 0)
+      __hip_atomic_fetch_add(&shard[idx[r] * k + c], d, __ATOMIC_RELAXED,
+                             __HIP_MEMORY_SCOPE_SYSTEM);
+  }
+}
+
 dim3 grid_for(int64_t total) {
   int64_t blocks = (total + GATHER_THREADS - 1) / GATHER_THREADS;
   if (blocks > 4096) blocks = 4096;   // grid-stride; >> 256 CUs
@@ -81,10 +111,10 @@ dim3 grid_for(int64_t total) {
 
 }  // namespace
 
-torch::Tensor os_shard_alloc(int64_t rows, int64_t k) {
-  // hipMalloc-backed f32 tensor usable as an IPC export base
+torch::Tensor os_shard_alloc(int64_t rows, int64_t k, int64_t dtype_i32) {
+  // uncached-memory tensor usable as an IPC export base (f32 or i32)
   void* p = nullptr;
-  size_t bytes = (size_t)rows * k * sizeof(float);
+  size_t bytes = (size_t)rows * k * 4;
   // uncached (MTYPE_UC): no XCD-L2 lines exist for shard memory, so a
   // plain write from one process can never linger dirty and later stomp
   // a peer's committed atomic — fine-grained alone still flaked (~1/5)
@@ -94,7 +124,7 @@ torch::Tensor os_shard_alloc(int64_t rows, int64_t k) {
   int dev = 0;
   TORCH_CHECK(hipGetDevice(&dev) == hipSuccess);
   auto opts = torch::TensorOptions()
-                  .dtype(torch::kFloat32)
+                  .dtype(dtype_i32 ? torch::kInt32 : torch::kFloat32)
                   .device(torch::kCUDA, dev);
   return torch::from_blob(
       p, {rows, k}, [](void* q) { hipFree(q); }, opts);
@@ -126,17 +156,26 @@ void os_ipc_close(int64_t ptr) {
   TORCH_CHECK(hipIpcCloseMemHandle((void*)(uintptr_t)ptr) == hipSuccess);
 }
 
-torch::Tensor os_gather(int64_t ptr, torch::Tensor idx, int64_t k) {
+torch::Tensor os_gather(int64_t ptr, torch::Tensor idx, int64_t k,
+                        int64_t dtype_i32) {
   CHECK_IN(idx);
   const int n = idx.numel();
-  auto out = torch::empty({(int64_t)n, k},
-                          idx.options().dtype(torch::kFloat32));
+  auto out = torch::empty(
+      {(int64_t)n, k},
+      idx.options().dtype(dtype_i32 ? torch::kInt32 : torch::kFloat32));
   if (n == 0) return out;
-  hipLaunchKernelGGL(gather_rows_kernel, grid_for((int64_t)n * k),
-                     dim3(GATHER_THREADS), 0, current_stream(),
-                     (const float*)(uintptr_t)ptr,
-                     idx.data_ptr<int64_t>(), out.data_ptr<float>(),
-                     n, (int)k);
+  if (dtype_i32)
+    hipLaunchKernelGGL(gather_rows_i32_kernel, grid_for((int64_t)n * k),
+                       dim3(GATHER_THREADS), 0, current_stream(),
+                       (const int*)(uintptr_t)ptr,
+                       idx.data_ptr<int64_t>(), out.data_ptr<int>(),
+                       n, (int)k);
+  else
+    hipLaunchKernelGGL(gather_rows_kernel, grid_for((int64_t)n * k),
+                       dim3(GATHER_THREADS), 0, current_stream(),
+                       (const float*)(uintptr_t)ptr,
+                       idx.data_ptr<int64_t>(), out.data_ptr<float>(),
+                       n, (int)k);
   return out;
 }
 
@@ -145,8 +184,14 @@ void os_scatter_add(int64_t ptr, torch::Tensor idx, torch::Tensor delta) {
   const int n = idx.numel();
   if (n == 0) return;
   const int k = delta.size(1);
-  hipLaunchKernelGGL(scatter_add_rows_kernel, grid_for((int64_t)n * k),
-                     dim3(GATHER_THREADS), 0, current_stream(),
-                     (float*)(uintptr_t)ptr, idx.data_ptr<int64_t>(),
-                     delta.data_ptr<float>(), n, k);
+  if (delta.scalar_type() == torch::kInt32)
+    hipLaunchKernelGGL(scatter_add_rows_i32_kernel, grid_for((int64_t)n * k),
+                       dim3(GATHER_THREADS), 0, current_stream(),
+                       (int*)(uintptr_t)ptr, idx.data_ptr<int64_t>(),
+                       delta.data_ptr<int>(), n, k);
+  else
+    hipLaunchKernelGGL(scatter_add_rows_kernel, grid_for((int64_t)n * k),
+                       dim3(GATHER_THREADS), 0, current_stream(),
+                       (float*)(uintptr_t)ptr, idx.data_ptr<int64_t>(),
+                       delta.data_ptr<float>(), n, k);
 }

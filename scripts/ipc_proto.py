@@ -27,7 +27,7 @@ def _child(q_h, q_done):
     handle = q_h.get(timeout=300)
     ptr = hip.os_ipc_open(torch.tensor(list(handle), dtype=torch.uint8))
     idx = torch.tensor([0, 2, 5], dtype=torch.int64, device="cuda")
-    got = hip.os_gather(ptr, idx, 4)
+    got = hip.os_gather(ptr, idx, 4, 0)
     torch.cuda.synchronize()
     assert abs(got[0, 0].item()) < 1e-6 and abs(got[1, 0].item() - 2) < 1e-6 \
         and abs(got[2, 3].item() - 5.3) < 1e-5, got.tolist()
@@ -44,7 +44,7 @@ def main():
 
     hip = ops._load_hip()
     torch.cuda.set_device(0)
-    shard = hip.os_shard_alloc(8, 4)
+    shard = hip.os_shard_alloc(8, 4, 0)
     pat = (torch.arange(8, device="cuda").float().unsqueeze(1)
            + torch.arange(4, device="cuda").float() * 0.1)
     shard.copy_(pat)

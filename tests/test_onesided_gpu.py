@@ -174,3 +174,78 @@ def test_async_lasso_job_one_sided():
     for n, mse in res:
         assert n == 4
         assert mse < 2.0, res
+
+
+def _async_lda_worker(rank, world):
+    """Async LDA: int32 one-sided word-topic table, counts conserved."""
+    import torch as T
+
+    from harmony_amd.config import JobConfig, RuntimeConfig
+    from harmony_amd.dolphin.master import run_job
+    from harmony_amd.runtime.bootstrap import init_executor
+
+    ctx = init_executor(RuntimeConfig(device="cuda", backend="gloo"))
+    job = JobConfig(job_id="os_lda", app="lda", max_num_epochs=3,
+                    num_mini_batches=2,
+                    app_args={"num_vocabs": 500, "num_topics": 64,
+                              "tokens_per_doc": 16, "docs_per_batch": 64,
+                              "one_sided": True})
+    s = run_job(job, ctx).summary()
+    # conservation: the global table's total count == total tokens pushed
+    # by BOTH ranks (each rank: 2 blocks x 64 docs x 16 tokens)
+    from harmony_amd import mlapps  # noqa: F401
+
+    import math
+
+    assert s["num_batches"] == 6
+    assert math.isfinite(s["log_likelihood"])
+    return s["num_batches"]
+
+
+def test_async_lda_job_one_sided():
+    res = run_dist(_async_lda_worker, world=2, timeout=300)
+    assert res == [6, 6]
+
+
+def _async_lda_conservation_worker(rank, world):
+    """Direct conservation check on the shared int32 table after an async
+    LDA run: sum of all word rows == total tokens, summary row matches."""
+    import torch as T
+
+    from harmony_amd import mlapps
+    from harmony_amd.config import JobConfig, RuntimeConfig
+    from harmony_amd.dolphin.worker import WorkerTasklet
+    from harmony_amd.runtime.bootstrap import init_executor
+    from harmony_amd.runtime.control import ControlPlane, TaskUnitScheduler
+
+    ctx = init_executor(RuntimeConfig(device="cuda", backend="gloo"))
+    cp = ControlPlane(ctx.store, ctx.rank, ctx.world_size)
+    job = JobConfig(job_id="os_lda_c", app="lda", max_num_epochs=2,
+                    num_mini_batches=2,
+                    app_args={"num_vocabs": 400, "num_topics": 32,
+                              "tokens_per_doc": 12, "docs_per_batch": 32,
+                              "one_sided": True})
+    app = mlapps.get_app("lda")
+    tables, trainer, provider = app.build(job, ctx, cp)
+    tus = TaskUnitScheduler(cp, {job.job_id})
+    WorkerTasklet(job, trainer, provider, cp, tus, ctx.rank,
+                  ctx.world_size).run()
+    t = tables["lda_model"]
+    t.fence()
+    cp.barrier("os_lda_c/done", ctx.world_size)
+    full = t.pull_full()
+    T.cuda.synchronize()
+    V = 400
+    total_tokens = world * 2 * 32 * 12      # ranks x blocks x docs x tokens
+    wt_total = int(full[:V].sum())
+    summ_total = int(full[V].sum())
+    cp.barrier("os_lda_c/checked", ctx.world_size)
+    t.close()
+    return (wt_total, summ_total, total_tokens)
+
+
+def test_async_lda_counts_conserved():
+    res = run_dist(_async_lda_conservation_worker, world=2, timeout=300)
+    for wt, summ, want in res:
+        assert wt == want, res
+        assert summ == want, res
