@@ -117,7 +117,7 @@ class OneSidedTable(Table):
     # ------------------------------------------------------------ async ops
 
     def _owner_of(self, blocks: torch.Tensor) -> torch.Tensor:
-        if not hasattr(self, "_owner_dev") or                 self._owner_dev.device != self.device:
+        if getattr(self, "_owner_dev", None) is None:
             self._owner_dev = self.ownership.owner.to(self.device,
                                                       torch.int64)
         return self._owner_dev[blocks]
