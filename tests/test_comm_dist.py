@@ -107,3 +107,44 @@ def test_mlr_two_ranks():
     for nb, acc in res:
         assert nb == 4
         assert acc > 0.4
+
+
+def _edge_cases_worker(rank, world):
+    # degenerate shapes the all-to-all-v plumbing must survive:
+    # empty key sets on one rank, single-key, all-keys-one-owner,
+    # duplicated keys through the merge path, then remove/put round-trips
+    ctx, table = _mk(rank, world)
+    out = []
+
+    # 1. one rank pulls nothing while the other pulls (collective pair)
+    keys = (torch.tensor([], dtype=torch.int64) if rank == 0
+            else torch.tensor([3]))
+    vals = table.get(keys)
+    out.append(vals.shape[0] == keys.shape[0])
+
+    # 2. all keys owned by ONE rank (0 owns blocks 0-3 = keys 0-31)
+    keys = torch.tensor([0, 1, 2, 30, 31])
+    vals = table.get(keys)
+    out.append(vals.shape == (5, 4))
+
+    # 3. duplicated keys in an update: merge path sums per key first
+    keys = torch.tensor([10, 10, 40, 10])
+    deltas = torch.ones(4, 4)
+    table.update(keys, deltas)
+    got = table.get(torch.tensor([10, 40]))
+    # both ranks pushed: key 10 got 3 per rank, key 40 got 1 per rank
+    out.append(bool(torch.allclose(
+        got, torch.tensor([[6.0], [2.0]]).repeat(1, 4))))
+
+    # 4. put then remove restores deterministic init (zeros)
+    table.put(torch.tensor([7]), torch.full((1, 4), 9.0))
+    out.append(float(table.get(torch.tensor([7]))[0, 0]) == 9.0)
+    table.remove(torch.tensor([7]))
+    out.append(float(table.get(torch.tensor([7]))[0, 0]) == 0.0)
+    return out
+
+
+def test_comm_edge_cases_two_ranks():
+    res = run_dist(_edge_cases_worker, world=2)
+    for r in res:
+        assert all(r), res
