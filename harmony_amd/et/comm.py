@@ -49,6 +49,12 @@ class DataPlane:
         """All-to-all with per-rank element counts on dim 0."""
         recv_shape = (sum(recv_splits),) + tuple(send.shape[1:])
         recv = torch.empty(recv_shape, dtype=send.dtype, device=send.device)
+        if send.shape[0] == 0 and recv.shape[0] == 0:
+            # globally-empty exchange (e.g. an LDA batch where no token
+            # changed topic): skip the collective — RCCL with all-zero
+            # splits is undefined-to-flaky, and every rank knows both
+            # split vectors here so the skip is symmetric
+            return recv
         if self._supports_a2a():
             dist.all_to_all_single(recv, send.contiguous(),
                                    output_split_sizes=recv_splits,
