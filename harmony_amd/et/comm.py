@@ -49,12 +49,6 @@ class DataPlane:
         """All-to-all with per-rank element counts on dim 0."""
         recv_shape = (sum(recv_splits),) + tuple(send.shape[1:])
         recv = torch.empty(recv_shape, dtype=send.dtype, device=send.device)
-        if send.shape[0] == 0 and recv.shape[0] == 0:
-            # globally-empty exchange (e.g. an LDA batch where no token
-            # changed topic): skip the collective — RCCL with all-zero
-            # splits is undefined-to-flaky, and every rank knows both
-            # split vectors here so the skip is symmetric
-            return recv
         if self._supports_a2a():
             dist.all_to_all_single(recv, send.contiguous(),
                                    output_split_sizes=recv_splits,
@@ -78,20 +72,20 @@ class DataPlane:
             r.wait()
         return recv
 
-    def _exchange_counts(self, counts: torch.Tensor) -> torch.Tensor:
-        """counts[i] = #elems this rank sends to rank i; returns recv counts.
-        NCCL/RCCL requires device tensors; gloo requires host tensors — the
-        input is moved to the backend's device, the result returned on CPU
-        (split sizes must be host ints anyway)."""
+    def _exchange_counts(self, counts: torch.Tensor
+                         ) -> Tuple[torch.Tensor, int]:
+        """counts[i] = #elems this rank sends to rank i. All-gathers the
+        FULL W x W count matrix (tiny) and returns (recv counts for me,
+        global total) — the total lets callers skip a globally-empty data
+        exchange SYMMETRICALLY on every rank (a per-rank skip would strand
+        the other ranks inside the collective). NCCL/RCCL requires device
+        tensors; split sizes must be host ints, so results return on CPU."""
         dev = self.device if self.backend == "nccl" else torch.device("cpu")
         counts = counts.to(dev)
-        if self._supports_a2a():
-            recv = torch.empty_like(counts)
-            dist.all_to_all_single(recv, counts, group=self.group)
-            return recv.cpu()
         gathered = [torch.empty_like(counts) for _ in range(self.world_size)]
         dist.all_gather(gathered, counts, group=self.group)
-        return torch.stack(gathered)[:, self.rank].contiguous().cpu()
+        mat = torch.stack(gathered)
+        return (mat[:, self.rank].contiguous().cpu(), int(mat.sum()))
 
     def _gather_perm(self, table) -> torch.Tensor:
         """Permutation p with full_rows[p] = concat(shards in rank order):
@@ -192,8 +186,11 @@ class DataPlane:
     def pull_keys(self, table, keys: torch.Tensor) -> torch.Tensor:
         keys = keys.to(self.device, torch.int64)
         sorted_keys, order, send_counts = self._route(table, keys)
-        recv_counts = self._exchange_counts(send_counts)
+        recv_counts, total = self._exchange_counts(send_counts)
         ssp, rsp = send_counts.tolist(), recv_counts.tolist()
+        if total == 0:
+            return torch.empty((0, table.cfg.value_dim), dtype=table.dtype,
+                               device=self.device)
         req_keys = self._all_to_all_v(sorted_keys, ssp, rsp)      # keys we serve
         served = table.get_local(req_keys)                        # [n_req, vdim]
         vals_sorted = self._all_to_all_v(served, rsp, ssp)        # back to askers
@@ -219,7 +216,9 @@ class DataPlane:
             uniq, agg = merge_key_deltas(keys, deltas, table.cfg.update_fn)
         sorted_keys, order, send_counts = self._route(table, uniq)
         sorted_deltas = agg[order]
-        recv_counts = self._exchange_counts(send_counts)
+        recv_counts, total = self._exchange_counts(send_counts)
+        if total == 0:
+            return
         ssp, rsp = send_counts.tolist(), recv_counts.tolist()
         recv_keys = self._all_to_all_v(sorted_keys, ssp, rsp)
         recv_deltas = self._all_to_all_v(sorted_deltas, ssp, rsp)
@@ -240,7 +239,9 @@ class DataPlane:
         values = values.to(self.device)
         sorted_keys, order, send_counts = self._route(table, keys)
         sorted_vals = values[order]
-        recv_counts = self._exchange_counts(send_counts)
+        recv_counts, total = self._exchange_counts(send_counts)
+        if total == 0:
+            return
         ssp, rsp = send_counts.tolist(), recv_counts.tolist()
         recv_keys = self._all_to_all_v(sorted_keys, ssp, rsp)
         recv_vals = self._all_to_all_v(sorted_vals, ssp, rsp)
@@ -252,7 +253,9 @@ class DataPlane:
         rows to their deterministic init values (see Table.remove)."""
         keys = keys.to(self.device, torch.int64)
         sorted_keys, order, send_counts = self._route(table, keys)
-        recv_counts = self._exchange_counts(send_counts)
+        recv_counts, total = self._exchange_counts(send_counts)
+        if total == 0:
+            return
         ssp, rsp = send_counts.tolist(), recv_counts.tolist()
         recv_keys = self._all_to_all_v(sorted_keys, ssp, rsp)
         if recv_keys.numel():
@@ -269,7 +272,9 @@ class DataPlane:
         keys = keys.to(self.device, torch.int64)
         sorted_keys, order, send_counts = self._route(table, keys)
         sorted_payload = payload[order]
-        recv_counts = self._exchange_counts(send_counts)
+        recv_counts, total = self._exchange_counts(send_counts)
+        if total == 0:
+            return
         ssp, rsp = send_counts.tolist(), recv_counts.tolist()
         recv_keys = self._all_to_all_v(sorted_keys, ssp, rsp)
         recv_payload = self._all_to_all_v(sorted_payload, ssp, rsp)

@@ -148,3 +148,27 @@ def test_comm_edge_cases_two_ranks():
     res = run_dist(_edge_cases_worker, world=2)
     for r in res:
         assert all(r), res
+
+
+def _global_empty_worker(rank, world):
+    # globally-empty exchanges (every rank empty) must be symmetric no-ops
+    # on every path — an asymmetric skip would strand peers in a collective
+    ctx, table = _mk(rank, world)
+    import torch as T
+
+    e = T.tensor([], dtype=T.int64)
+    out = table.get(e)
+    ok = [out.shape == (0, 4)]
+    table.update(e, T.empty(0, 4))
+    table.put(e, T.empty(0, 4))
+    table.remove(e)
+    # and a mixed call right after (plumbing still healthy)
+    v = table.get(T.tensor([1]))
+    ok.append(v.shape == (1, 4))
+    return ok
+
+
+def test_global_empty_exchanges():
+    res = run_dist(_global_empty_worker, world=2)
+    for r in res:
+        assert all(r), res
