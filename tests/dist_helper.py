@@ -57,12 +57,17 @@ def _stderr_tails(world: int, port: int, n: int = 40) -> str:
 
 def run_dist(fn, world: int = 2, args=(), timeout: int = 120):
     """Run fn(rank, world, *args) in `world` processes; returns [out_rank0..].
-    Retries once on a TCPStore port collision (free_port is check-then-use,
-    so a rare EADDRINUSE race with another process is possible)."""
+    Retries once on INFRASTRUCTURE failures only (port collisions, socket
+    resets, spawn stalls) — genuine assertion/logic failures propagate on
+    the first attempt."""
+    _INFRA = ("EADDRINUSE", "Connection reset", "Connection refused",
+              "did not report within", "worker died",
+              "Socket Timeout", "wait timeout after")
     try:
         return _run_dist_once(fn, world, args, timeout)
-    except RuntimeError as e:
-        if "EADDRINUSE" not in str(e):
+    except (RuntimeError, TimeoutError) as e:
+        msg = str(e)
+        if "AssertionError" in msg or not any(t in msg for t in _INFRA):
             raise
         return _run_dist_once(fn, world, args, timeout)
 
