@@ -18,7 +18,9 @@ stored in an object table keyed by label (multi-class = one forest per
 label, regression = label 0).
 
 App args: num_features, batch_size, num_bins, max_depth, step_size
-(shrinkage), lam (leaf L2), objective ("regression" | "binary").
+(shrinkage), lam (leaf L2), objective ("regression" | "multiclass"
+with num_classes — one forest per label in the object table, reference
+GBTTrainer's valueTypeNum > 3 path / GroupedTree).
 """
 
 from __future__ import annotations
@@ -40,7 +42,8 @@ MODEL_TABLE = "gbt_model"
 
 def defaults(job: JobConfig) -> dict:
     a = dict(num_features=32, batch_size=4096, num_bins=64, max_depth=4,
-             step_size=0.1, lam=1.0, noise=0.1, objective="regression")
+             step_size=0.1, lam=1.0, noise=0.1, objective="regression",
+             num_classes=3)
     a.update(job.app_args)
     return a
 
@@ -141,6 +144,8 @@ class GBTTrainer(Trainer):
                                     app_args=ctx.app_args))
         self.table: ObjectTable = ctx.table(MODEL_TABLE)
         self.forest: List[GBTree] = []
+        self.forests: List[List[GBTree]] = []
+        self._mc_cache: dict = {}
         self._mse = 0.0
         # incremental forest predictions: the forest is append-only
         # (update fn v + [d]), so each block's prediction is cached and only
@@ -151,9 +156,16 @@ class GBTTrainer(Trainer):
     def pull_model(self) -> None:
         # pullAllTrees (reference :767): gather every label's forest
         allv = self.table.pull_all()
-        self.forest = list(allv.get(0, []))
+        if self.a["objective"] == "multiclass":
+            C = self.a["num_classes"]
+            self.forests = [list(allv.get(c, [])) for c in range(C)]
+        else:
+            self.forest = list(allv.get(0, []))
 
     def local_compute(self) -> None:
+        if self.a["objective"] == "multiclass":
+            self._compute_multiclass()
+            return
         bins, y = self.batch
         a = self.a
         done, last, pred = self._pred_cache.get(id(self.batch),
@@ -172,12 +184,49 @@ class GBTTrainer(Trainer):
         self.new_tree = build_tree(bins, resid, a["num_bins"], a["max_depth"],
                                    a["lam"])
 
+    def _compute_multiclass(self) -> None:
+        """One-vs-all boosting (reference GBTTrainer multi-label: one GBTree
+        forest per label type): per class, residual = onehot - sigmoid-free
+        additive score, one new tree per class per batch."""
+        bins, y = self.batch
+        a = self.a
+        C = a["num_classes"]
+        self.new_trees = []
+        correct = None
+        scores = []
+        for c in range(C):
+            target = (y == c).float()
+            key = (id(self.batch), c)
+            done, last, pred = self._mc_cache.get(key, (0, None, None))
+            forest = self.forests[c]
+            if pred is None or done > len(forest) or (
+                    done > 0 and forest[done - 1] is not last):
+                done, pred = 0, torch.zeros_like(target)
+            for t in forest[done:]:
+                pred = pred + a["step_size"] * t.predict_bins(bins)
+            self._mc_cache[key] = (len(forest),
+                                   forest[-1] if forest else None, pred)
+            resid = target - pred
+            scores.append(pred)
+            self.new_trees.append(
+                (c, build_tree(bins, resid, a["num_bins"], a["max_depth"],
+                               a["lam"])))
+        pred_cls = torch.stack(scores, dim=1).argmax(dim=1)
+        self._mse = float((pred_cls != y.long()).float().mean())  # error rate
+        _ = correct
+
     def push_update(self) -> None:
         # reference pushes the tree from localCompute; here the push phase
         # does it so NET ordering holds (object push is collective)
-        self.table.push_items([(0, self.new_tree)])
+        if self.a["objective"] == "multiclass":
+            self.table.push_items(self.new_trees)
+        else:
+            self.table.push_items([(0, self.new_tree)])
 
     def evaluate_model(self):
+        if self.a["objective"] == "multiclass":
+            n = float(sum(len(f) for f in self.forests))
+            return {"error_rate": self._mse, "num_trees": n}
         return {"mse": self._mse, "num_trees": float(len(self.forest))}
 
     def num_batch_examples(self) -> int:
@@ -210,12 +259,21 @@ def make_batches(job: JobConfig, rank: int, device: torch.device,
         return blocks
     g = torch.Generator().manual_seed(stable_seed(job.job_id, "data", rank))
     w = torch.randn(F, generator=g)
+    mc = a["objective"] == "multiclass"
+    if mc:
+        W = torch.randn(a["num_classes"], F, generator=g)
     blocks = []
     n_blocks = job.num_worker_blocks or job.num_mini_batches
     for _ in range(n_blocks):
         X = torch.randn(a["batch_size"], F, generator=g)
-        y = (X @ w + torch.sin(3 * X[:, 0]) * 2
-             + a["noise"] * torch.randn(a["batch_size"], generator=g))
+        if mc:
+            # separable class labels: argmax of per-class linear scores
+            y = (X @ W.t() + a["noise"]
+                 * torch.randn(a["batch_size"], a["num_classes"],
+                               generator=g)).argmax(dim=1).float()
+        else:
+            y = (X @ w + torch.sin(3 * X[:, 0]) * 2
+                 + a["noise"] * torch.randn(a["batch_size"], generator=g))
         bins = quantize(X, a["num_bins"])
         blocks.append((bins.to(device), y.to(device)))
     return blocks

@@ -98,7 +98,8 @@ def test_gbt_incremental_pred_matches_full():
     tr._pred_cache = {}
     tr.forest = []
     tr.batch = (bins, y)
-    tr.a = {"step_size": 0.1, "num_bins": 16, "max_depth": 2, "lam": 1.0}
+    tr.a = {"step_size": 0.1, "num_bins": 16, "max_depth": 2, "lam": 1.0,
+            "objective": "regression"}
     inc = None
     for k in (2, 4, 5):            # grow the forest incrementally
         tr.forest = trees[:k]
@@ -113,3 +114,25 @@ def test_gbt_incremental_pred_matches_full():
     tr.forest = clones
     GBTTrainer.local_compute(tr)
     assert abs(tr._mse - want) < 1e-5
+
+
+def test_gbt_multiclass_end_to_end():
+    # one forest per label (reference multi-label GBT): error rate falls
+    # well below chance on separable synthetic classes
+    import torch
+
+    from harmony_amd.config import JobConfig, RuntimeConfig
+    from harmony_amd.dolphin.master import run_job
+    from harmony_amd.runtime.bootstrap import init_executor
+
+    ctx = init_executor(RuntimeConfig(device="cpu"))
+    job = JobConfig(job_id="gbt_mc", app="gbt", max_num_epochs=6,
+                    num_mini_batches=2,
+                    app_args={"num_features": 8, "batch_size": 1024,
+                              "num_bins": 32, "max_depth": 3,
+                              "step_size": 0.5, "objective": "multiclass",
+                              "num_classes": 3, "noise": 0.1})
+    s = run_job(job, ctx).summary()
+    assert s["num_batches"] == 12
+    assert s["error_rate"] < 0.25, s       # chance = 0.67 for 3 classes
+    assert s["num_trees"] == 3 * 11        # C forests, one tree/batch each
