@@ -192,7 +192,6 @@ class GBTTrainer(Trainer):
         a = self.a
         C = a["num_classes"]
         self.new_trees = []
-        correct = None
         scores = []
         for c in range(C):
             target = (y == c).float()
@@ -213,7 +212,6 @@ class GBTTrainer(Trainer):
                                a["lam"])))
         pred_cls = torch.stack(scores, dim=1).argmax(dim=1)
         self._mse = float((pred_cls != y.long()).float().mean())  # error rate
-        _ = correct
 
     def push_update(self) -> None:
         # reference pushes the tree from localCompute; here the push phase
