@@ -302,3 +302,52 @@ def test_one_sided_requires_gpu_clear_error():
                       update_fn="add", init_fn="zeros")
     with _pt.raises(RuntimeError, match="HIP extension"):
         OneSidedTable(cfg, 0, 1, torch.device("cpu"))
+
+
+def test_ssp_clock_bounds_skew_and_stops():
+    # reference MiniBatchController: a fast worker blocks when more than
+    # `slack` batches ahead of the slowest; stop_at halts deterministically
+    import threading
+    import time
+
+    from harmony_amd.runtime.bootstrap import LocalStore
+    from harmony_amd.runtime.control import ControlPlane, SSPClock
+
+    store = LocalStore()
+    cp = ControlPlane(store, 0, 1)
+    clock = SSPClock(cp, "ssp_t", num_workers=2, slack=1)
+    progress = {0: 0, 1: 0}
+    max_skew = [0]
+    done = threading.Event()
+
+    def fast():
+        for _ in range(20):
+            assert clock.tick_and_wait(0)
+            progress[0] += 1
+            max_skew[0] = max(max_skew[0], progress[0] - progress[1])
+        done.set()
+
+    def slow():
+        for _ in range(20):
+            time.sleep(0.002)
+            assert clock.tick_and_wait(1)
+            progress[1] += 1
+
+    ts = [threading.Thread(target=fast), threading.Thread(target=slow)]
+    for t in ts:
+        t.start()
+    for t in ts:
+        t.join(timeout=30)
+    assert done.is_set()
+    # skew observed between a tick and the peer's next tick can be at most
+    # slack+1 (the fast worker passes the gate, increments, THEN we sample)
+    assert max_skew[0] <= 2, max_skew[0]
+
+    # deterministic stop: both workers stop after exactly 3 more batches
+    clock2 = SSPClock(cp, "ssp_s", num_workers=2, slack=-1)
+    clock2.request_stop_at(3)
+    for r in (0, 1):
+        ticks = 0
+        while clock2.tick_and_wait(r):
+            ticks += 1
+        assert ticks == 3, ticks
