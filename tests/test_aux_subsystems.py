@@ -351,3 +351,43 @@ def test_ssp_clock_bounds_skew_and_stops():
         while clock2.tick_and_wait(r):
             ticks += 1
         assert ticks == 3, ticks
+
+
+def test_task_unit_scheduler_orders_interleaved_jobs():
+    # direct unit test of the ticket protocol (integration tests cover the
+    # full stack): two "jobs" on two threads of one rank must execute NET
+    # phases in global-ticket order even when requested out of order
+    import threading
+    import time
+
+    from harmony_amd.runtime.bootstrap import LocalStore
+    from harmony_amd.runtime.control import ControlPlane, TaskUnitScheduler
+
+    store = LocalStore()
+    cp = ControlPlane(store, 0, 1)
+    tus = TaskUnitScheduler(cp, {"jA", "jB"}, multi_job=True)
+    log = []
+    lock = threading.Lock()
+
+    def run_job(jid, phases, delay):
+        for p in range(phases):
+            time.sleep(delay)
+            with tus.net(jid, p):
+                with lock:
+                    log.append((jid, p))
+                time.sleep(0.001)
+
+    ts = [threading.Thread(target=run_job, args=("jA", 6, 0.0)),
+          threading.Thread(target=run_job, args=("jB", 6, 0.003))]
+    for t in ts:
+        t.start()
+    for t in ts:
+        t.join(timeout=30)
+    assert len(log) == 12
+    # reconstruct each phase's global ticket and check the log is sorted
+    tickets = [int(store.get(f"tu/seq_of/{j}/{p}/v")) for j, p in log]
+    assert tickets == sorted(tickets), list(zip(log, tickets))
+    # per-job phases in order
+    for jid in ("jA", "jB"):
+        ph = [p for j, p in log if j == jid]
+        assert ph == sorted(ph)
