@@ -105,14 +105,14 @@ def _async_mlr_worker(rank, world):
     X = torch.randn(512, F, generator=g).cuda()
     y = (X @ W_true.t().cuda()).argmax(dim=1)
     keys = torch.arange(C, device="cuda")
-    steps = 30 if rank == 0 else 18    # uneven pace
+    steps = 80 if rank == 0 else 48    # uneven pace
     for _ in range(steps):
         W = t.pull_full()              # async full pull
         logits = X @ W.t()
         p = torch.softmax(logits, dim=1)
         p[torch.arange(512, device="cuda"), y] -= 1
         grad = p.t() @ X / 512
-        t.push(keys, -0.5 * grad)      # async add push
+        t.push(keys, -0.3 * grad)      # async add push
     t.fence()
     _barrier(store, "done", rank, world)
     W = t.pull_full()
