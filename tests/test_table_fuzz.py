@@ -61,3 +61,30 @@ def test_table_matches_dict_model(ops_list):
     got = t.get(torch.arange(NUM_KEYS))
     want = torch.tensor([ref[k] for k in range(NUM_KEYS)])
     assert torch.allclose(got, want), (got, want)
+
+
+def mk_min_table():
+    cfg = TableConfig(table_id="fuzzmin", num_keys=NUM_KEYS, value_dim=VDIM,
+                      num_blocks=6, init_fn="zeros", update_fn="min",
+                      init_args={})
+    return Table(cfg, 0, 1, torch.device("cpu"))
+
+
+@settings(max_examples=40, deadline=None)
+@given(st.lists(st.tuples(st.lists(st.integers(0, NUM_KEYS - 1),
+                                   min_size=1, max_size=8),
+                          st.integers(-9, 9)),
+                min_size=1, max_size=16))
+def test_min_update_fn_matches_model(updates):
+    # the "min" merge algebra (SSSP-style): duplicates within one push and
+    # across pushes must both take elementwise minimums
+    t = mk_min_table()
+    ref = {k: [0.0] * VDIM for k in range(NUM_KEYS)}
+    for keys, val in updates:
+        kt = torch.tensor(keys)
+        t.update(kt, torch.full((kt.numel(), VDIM), float(val)))
+        for k in keys:
+            ref[k] = [min(x, float(val)) for x in ref[k]]
+    got = t.get(torch.arange(NUM_KEYS))
+    want = torch.tensor([ref[k] for k in range(NUM_KEYS)])
+    assert torch.allclose(got, want), (got, want)
