@@ -236,3 +236,20 @@ def test_jobserver_concurrent_jobs_two_ranks():
 
     res = run_dist(_jobserver_concurrent_worker, world=2, timeout=240)
     assert res == [True, True]
+
+
+def test_client_flag_parsing():
+    from harmony_amd.jobserver.client import _parse_flags
+
+    job_kw, app_args, wait = _parse_flags(
+        ["-app", "mlr", "-job_id", "j1", "-max_num_epochs", "3",
+         "-num_mini_batches", "2", "-step_size", "0.5",
+         "-one_sided", "true", "-restore_chkp", "src/epoch2",
+         "-model_chkp_per_epoch", "true", "--wait"])
+    assert wait is True
+    assert job_kw["app"] == "mlr" and job_kw["job_id"] == "j1"
+    assert job_kw["max_num_epochs"] == 3          # typed job field
+    assert job_kw["restore_chkp"] == "src/epoch2"
+    assert job_kw["model_chkp_per_epoch"] is True
+    assert app_args["step_size"] == 0.5           # numeric app arg
+    assert app_args["one_sided"] is True          # bool-coerced app arg
