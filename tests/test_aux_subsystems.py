@@ -391,3 +391,28 @@ def test_task_unit_scheduler_orders_interleaved_jobs():
     for jid in ("jA", "jB"):
         ph = [p for j, p in log if j == jid]
         assert ph == sorted(ph)
+
+
+def test_heartbeat_failfast_logic():
+    # failure detection (reference fail-fast handlers): a stale heartbeat
+    # flips the failed+shutdown flags; fresh heartbeats do not
+    import time
+
+    from harmony_amd.jobserver.server import JobServerDriver
+    from harmony_amd.runtime.bootstrap import LocalStore
+    from harmony_amd.runtime.control import ControlPlane
+
+    class _Ctx:
+        rank = 0
+        world_size = 2
+        is_master = True
+        store = LocalStore()
+
+    drv = JobServerDriver.__new__(JobServerDriver)
+    drv.ctx = _Ctx()
+    drv.cp = ControlPlane(_Ctx.store, 0, 2)
+    # rank 1's heartbeat is ancient -> the loop must fail fast and return
+    _Ctx.store.set("js/hb/1", str(time.time() - 120))
+    drv._heartbeat_loop()
+    assert drv.cp.flag_set("js/failed")
+    assert drv.cp.flag_set("js/shutdown")
