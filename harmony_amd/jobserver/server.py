@@ -205,6 +205,10 @@ class JobServerDriver:
                 rec = json.loads(self.ctx.store.get(f"js/job/{next_idx}"))
                 job = JobConfig(**rec["job"])
                 ranks = rec["ranks"]
+                from harmony_amd.utils.joblog import job_logger
+
+                job_logger(job.job_id, self.ctx.rank).info(
+                    "scheduled on executors %s", ranks)
                 if self.ctx.rank in ranks:
                     my_jobs.add(job.job_id)
                     self.tus.set_jobs(my_jobs)
@@ -213,6 +217,10 @@ class JobServerDriver:
                     group = (dist.new_group(ranks)
                              if dist.is_initialized() and
                              len(ranks) < self.ctx.world_size else None)
+                from harmony_amd.utils.joblog import job_logger
+
+                job_logger(job.job_id, self.ctx.rank).info(
+                    "scheduled on executors %s", ranks)
                 if self.ctx.rank in ranks:
                     view = JobView(rank=ranks.index(self.ctx.rank),
                                    world_size=len(ranks),

@@ -416,3 +416,17 @@ def test_heartbeat_failfast_logic():
     drv._heartbeat_loop()
     assert drv.cp.flag_set("js/failed")
     assert drv.cp.flag_set("js/shutdown")
+
+
+def test_job_logger_prefix(capsys):
+    import logging
+
+    from harmony_amd.utils.joblog import job_logger
+
+    lg = job_logger("jx", 2)
+    assert isinstance(lg, logging.Logger)
+    lg.info("hello %d", 7)
+    err = capsys.readouterr().err
+    assert "[jx r2] INFO hello 7" in err
+    # same logger instance on repeat (no duplicate handlers)
+    assert job_logger("jx", 2) is lg and len(lg.handlers) == 1
