@@ -76,6 +76,38 @@ class DropTableOp:
         return {"op": "drop", "table_id": self.table_id}
 
 
+def compile_switch(table_id: str, stop_ranks, start, moves) -> "Plan":
+    """Role-switch plan compiler (reference PlanCompiler.translateToSwitch,
+    dolphin/plan/PlanCompiler.java:66-90): ranks leaving the worker role
+    STOP first, then their mini-batch load and/or table blocks MOVE, then
+    joining ranks START — the dependency DAG encodes that order so the
+    executor can never start a worker before its data exists.
+
+    stop_ranks: ranks whose worker role ends (become pure servers)
+    start:      {rank: num_batches} ranks (re)starting as workers
+    moves:      {block_id: dst_rank} table blocks to migrate
+    """
+    p = Plan()
+    stop_idx = []
+    for r in sorted(stop_ranks):
+        stop_idx.append(len(p.ops))
+        p.ops.append(StopWorkerOp(r))
+    move_idx = None
+    if moves:
+        move_idx = len(p.ops)
+        p.ops.append(MoveOp(table_id, tuple(sorted(moves.items()))))
+        for i in stop_idx:
+            p.deps.append((i, move_idx))
+    for r, n in sorted((start or {}).items()):
+        i = len(p.ops)
+        p.ops.append(StartWorkerOp(r, n))
+        p.deps.append((move_idx if move_idx is not None
+                       else (stop_idx[-1] if stop_idx else i), i))
+        if move_idx is None and not stop_idx:
+            p.deps.pop()          # nothing to depend on
+    return p
+
+
 def op_from_json(d: dict):
     if d["op"] == "move":
         return MoveOp(d["table_id"], tuple(tuple(m) for m in d["moves"]))

@@ -201,3 +201,30 @@ def run_dist_opt(fn, world, timeout):
     from tests.dist_helper import run_dist
 
     return run_dist(fn, world=world, timeout=timeout)
+
+
+def test_compile_switch_dag_order():
+    # PlanCompiler.translateToSwitch analogue: Stop -> Move -> Start order
+    # is encoded as dependencies and honored by the executor
+    from harmony_amd.optimizer.plan import (MoveOp, PlanExecutor,
+                                            StartWorkerOp, StopWorkerOp,
+                                            compile_switch)
+
+    p = compile_switch("t", stop_ranks=[1], start={0: 6},
+                       moves={3: 0, 4: 0})
+    kinds = [type(o).__name__ for o in p.ops]
+    assert kinds == ["StopWorkerOp", "MoveOp", "StartWorkerOp"]
+    assert (0, 1) in p.deps and (1, 2) in p.deps
+    # round-trips through json
+    from harmony_amd.optimizer.plan import Plan
+
+    p2 = Plan.from_json(p.to_json())
+    assert [type(o).__name__ for o in p2.ops] == kinds
+    # executor consumes it (no tables -> moves no-op) and records shares
+    ex = PlanExecutor({}, rank=0, world_size=2)
+    ex.execute(p2)
+    assert ex.batch_shares == {1: 0, 0: 6}
+
+    # degenerate: only starts, no deps needed
+    p3 = compile_switch("t", stop_ranks=[], start={2: 4}, moves={})
+    assert len(p3.ops) == 1 and p3.deps == []
