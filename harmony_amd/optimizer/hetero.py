@@ -110,7 +110,12 @@ class HeterogeneousOptimizer(Optimizer):
             moves = moves_to_targets(ol, target)
             if moves:
                 plan.ops.append(MoveOp(tid, tuple(sorted(moves.items()))))
+        # shares apply only AFTER the block moves: a rank whose share grows
+        # must already own its new blocks (PlanCompiler's switch ordering)
+        share_idx = len(plan.ops)
         plan.ops.append(SetBatchShareOp(
             tuple((i, max(1, int(round(di)))) for i, di in enumerate(d))))
+        for i in range(share_idx):
+            plan.deps.append((i, share_idx))
         plan.estimated_benefit = (cur - T) / cur
         return plan
