@@ -39,6 +39,51 @@ def _make_optimizer(name: str):
     raise KeyError(f"unknown optimizer '{name}'")
 
 
+def _restore_tables(job: JobConfig, tables: dict) -> None:
+    """Start from a model snapshot (reference ETMaster.createTable(chkpId):
+    block files are rank-independent, so the restore re-partitions for
+    whatever executor set this job runs on).
+
+    `restore_chkp` forms, tried in order per table:
+      "<chkp_id>"            — this job's own checkpoint
+      "<src_job>/<chkp_id>"  — another job's (e.g. ModelChkpManager's
+                               per-epoch snapshots); the per-table subdir is
+                               keyed by the SOURCE job's table id, so the
+                               lookup rewrites this job's prefix to the
+                               source's.
+    Raises if a table restores zero blocks — a silent no-op restore would
+    masquerade as training-from-scratch."""
+    from harmony_amd.dolphin.model_eval import _safe
+    from harmony_amd.et.checkpoint import CheckpointManager
+
+    cm = CheckpointManager(temp_root=job.chkp_path)
+    for t in tables.values():
+        if not hasattr(t, "cfg"):
+            continue
+        short = t.cfg.table_id.split("/", 1)[-1]
+        src, _, rest = job.restore_chkp.partition("/")
+        candidates = [
+            (job.job_id, f"{job.restore_chkp}/{_safe(t.cfg.table_id)}"),
+            (job.job_id, job.restore_chkp),
+        ]
+        if rest:
+            candidates += [
+                (src, f"{rest}/{_safe(src + '/' + short)}"),
+                (src, rest),
+            ]
+        loaded = -1
+        for app_id, cid in candidates:
+            if cm.exists(app_id, cid):
+                loaded = cm.load_into(t, app_id, cid)
+                if loaded > 0:
+                    break
+        if loaded <= 0:
+            raise FileNotFoundError(
+                f"restore_chkp {job.restore_chkp!r}: no usable checkpoint "
+                f"for table {t.cfg.table_id} under {job.chkp_path} "
+                f"(tried {candidates}; loaded={loaded})")
+
+
 def run_job(job: JobConfig, ctx: ExecutorContext,
             cp: Optional[ControlPlane] = None,
             tus: Optional[TaskUnitScheduler] = None,
@@ -54,32 +99,7 @@ def run_job(job: JobConfig, ctx: ExecutorContext,
     app = mlapps.get_app(job.app)
     tables, trainer, provider = app.build(job, ctx, cp)
     if job.restore_chkp:
-        # start from a model snapshot (reference ETMaster.createTable(chkpId):
-        # block files are rank-independent, so the restore re-partitions for
-        # whatever executor set this job runs on). Accepts either a plain
-        # checkpoint id or a ModelChkpManager epoch id (per-table subdirs).
-        from harmony_amd.dolphin.model_eval import _safe
-        from harmony_amd.et.checkpoint import CheckpointManager
-
-        cm = CheckpointManager(temp_root=job.chkp_path)
-        for t in tables.values():
-            if not hasattr(t, "cfg"):
-                continue
-            cid = f"{job.restore_chkp}/{_safe(t.cfg.table_id)}"
-            app_id, use = ((job.job_id, cid) if cm.exists(job.job_id, cid)
-                           else (None, None))
-            if app_id is None:
-                # cross-job restore: "<src_job>/<chkp_id>" syntax
-                src, _, rest = job.restore_chkp.partition("/")
-                for c in (f"{rest}/{_safe(t.cfg.table_id)}", rest):
-                    if rest and cm.exists(src, c):
-                        app_id, use = src, c
-                        break
-            if app_id is None:
-                raise FileNotFoundError(
-                    f"restore_chkp {job.restore_chkp!r}: no checkpoint for "
-                    f"table {t.cfg.table_id} under {job.chkp_path}")
-            cm.load_into(t, app_id, use)
+        _restore_tables(job, tables)
     orch = None
     if optimizer is None and job.optimizer:
         optimizer = _make_optimizer(job.optimizer)

@@ -85,12 +85,15 @@ class CheckpointManager:
 
     # --------------------------------------------------------------- restore
 
-    def load_into(self, table, app_id: str, chkp_id: str) -> None:
+    def load_into(self, table, app_id: str, chkp_id: str) -> int:
         """Load blocks this rank owns NOW (restore re-partitions freely —
-        reference ChkpLoadMsg.blockOwners planning). Call on every rank."""
+        reference ChkpLoadMsg.blockOwners planning). Call on every rank.
+        Returns the number of blocks loaded (0 = nothing matched — callers
+        that require a real restore must check)."""
         d = self.exists(app_id, chkp_id)
         if d is None:
             raise FileNotFoundError(f"no checkpoint {app_id}/{chkp_id}")
+        n = 0
         for b in table.owned_blocks:
             f = d / str(b)
             if not f.exists():
@@ -102,6 +105,8 @@ class CheckpointManager:
             else:
                 view[payload["rows"].to(view.device)] = \
                     payload["values"].to(view.device, view.dtype)
+            n += 1
+        return n
 
     def saved_table_config(self, app_id: str, chkp_id: str) -> TableConfig:
         d = self.exists(app_id, chkp_id)

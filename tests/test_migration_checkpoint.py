@@ -173,14 +173,35 @@ def test_job_restore_chkp_resumes_training(tmp_path):
     m1 = run_job(j1, ctx)
     assert m1.summary()["accuracy"] > 0.6
 
+    # exact restore check: build a NEW job's tables, restore from the
+    # trained job's last snapshot, and compare table contents against the
+    # source's final shard (block files are content-addressed by block id)
+    import torch
+
+    from harmony_amd import mlapps
+    from harmony_amd.dolphin.master import _restore_tables
+    from harmony_amd.runtime.control import ControlPlane
+
     j2 = JobConfig(job_id="rc_resume", app="mlr", max_num_epochs=1,
                    num_mini_batches=4, app_args=app_args,
                    chkp_path=str(tmp_path),
                    restore_chkp="rc_train/epoch4")
+    cp = ControlPlane(ctx.store, 0, 1)
+    tables2, _, _ = mlapps.get_app("mlr").build(j2, ctx, cp)
+    t2 = tables2["mlr_model"]
+    before = t2.pull_all().clone()
+    _restore_tables(j2, tables2)
+    after = t2.pull_all()
+    assert not torch.equal(before, after)      # something actually loaded
+    from harmony_amd.et.checkpoint import CheckpointManager
+
+    src = CheckpointManager(temp_root=str(tmp_path))
+    d = src.exists("rc_train", "epoch4/rc_train_mlr_model")
+    blk0 = torch.load(d / "0", weights_only=True)["values"]
+    assert torch.allclose(t2.local_block_view(0).cpu(), blk0)
+    # and a full resumed run still works end-to-end
     m2 = run_job(j2, ctx)
-    # fresh-start accuracy over the first epoch is near chance (~0.2);
-    # restored start must be far above it
-    assert m2.summary()["accuracy"] > 0.6
+    assert m2.summary()["num_batches"] == 4
 
     j3 = JobConfig(job_id="rc_missing", app="mlr", max_num_epochs=1,
                    num_mini_batches=2, app_args=app_args,
