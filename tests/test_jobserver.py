@@ -253,3 +253,41 @@ def test_client_flag_parsing():
     assert job_kw["model_chkp_per_epoch"] is True
     assert app_args["step_size"] == 0.5           # numeric app arg
     assert app_args["one_sided"] is True          # bool-coerced app arg
+
+
+def test_jobserver_restore_flow(tmp_path):
+    # the showcase path: train with per-epoch snapshots through the server,
+    # then submit a second job restoring from the first's snapshot —
+    # _restore_tables now raises on a silent no-op, so "done" means the
+    # blocks really loaded
+    from harmony_amd.config import JobConfig, RuntimeConfig
+    from harmony_amd.jobserver import client
+    from harmony_amd.jobserver.server import JobServerDriver
+    from harmony_amd.runtime.bootstrap import init_executor
+
+    port = free_port()
+    ctx = init_executor(RuntimeConfig(device="cpu"))
+    driver = JobServerDriver(ctx, scheduler="default", port=port)
+    t = threading.Thread(target=driver.run, daemon=True)
+    t.start()
+    time.sleep(0.3)
+    args = {"num_classes": 3, "num_features": 32,
+            "num_parts_per_class": 2, "batch_size": 64, "step_size": 0.5}
+    j1 = JobConfig(job_id="rs1", app="mlr", max_num_epochs=2,
+                   num_mini_batches=2, app_args=args,
+                   chkp_path=str(tmp_path), model_chkp_per_epoch=True)
+    r1 = client.submit(j1, port=port, wait=True, timeout=60)
+    assert r1["status"] == "done", r1
+    j2 = JobConfig(job_id="rs2", app="mlr", max_num_epochs=1,
+                   num_mini_batches=2, app_args=args,
+                   chkp_path=str(tmp_path), restore_chkp="rs1/epoch1")
+    r2 = client.submit(j2, port=port, wait=True, timeout=60)
+    assert r2["status"] == "done", r2
+    # bad restore id fails the job, not the server
+    j3 = JobConfig(job_id="rs3", app="mlr", max_num_epochs=1,
+                   num_mini_batches=2, app_args=args,
+                   chkp_path=str(tmp_path), restore_chkp="rs1/epoch99")
+    r3 = client.submit(j3, port=port, wait=True, timeout=60)
+    assert r3["status"] == "failed", r3
+    client.shutdown(port=port)
+    t.join(timeout=30)
