@@ -144,6 +144,9 @@ def test_offline_model_eval(tmp_path):
     assert len(accs) == 3
     assert accs[-1] >= accs[0] - 0.05
     assert accs[-1] > 0.5
+    # non-vacuity: a no-op snapshot load would evaluate the SAME live model
+    # three times — the epoch-0 snapshot must differ from the final one
+    assert accs[0] != accs[-1], accs
 
 
 def test_hetero_milp_optimizer():
