@@ -184,6 +184,8 @@ def test_offline_eval_via_job_flags(tmp_path):
     s = run_job(job, ctx).summary()
     assert "offline/epoch0/accuracy" in s
     assert "offline/epoch1/accuracy" in s
+    # non-vacuity: distinct snapshots -> distinct replayed accuracies
+    assert s["offline/epoch0/accuracy"] != s["offline/epoch1/accuracy"]
 
 
 def test_sanitizer_validate_unit():
