@@ -181,3 +181,25 @@ def parse_lda_split(path: str, rank: int, world_size: int):
         buf = _read_split_bytes(path, split)
         return tuple(nat.parse_lda_bytes(buf.decode("utf-8", "replace")))
     return parse_lda(read_split(path, split))
+
+
+def load_keyless_split(path: str, rank: int, world_size: int,
+                       parse_line=None):
+    """Keyless bulk load (reference NoneKeyBulkDataLoader + its local
+    key generator, et/evaluator/impl/NoneKeyBulkDataLoader): records with
+    no key column get GLOBALLY UNIQUE sequential keys — each rank counts
+    every split's records (cheap line scan) so its keys start after all
+    earlier splits' records, with no communication.
+
+    Returns (keys int64 tensor, values list) for this rank's split;
+    parse_line maps a stripped line to a value (default: the line itself).
+    """
+    parse_line = parse_line or (lambda s: s)
+    splits = compute_splits(path, world_size)
+    start = 0
+    for r in range(rank):
+        start += sum(1 for ln in read_split(path, splits[r]) if ln.strip())
+    values = [parse_line(ln.strip()) for ln in read_split(path, splits[rank])
+              if ln.strip()]
+    keys = torch.arange(start, start + len(values), dtype=torch.int64)
+    return keys, values

@@ -97,3 +97,23 @@ def test_nmf_job_from_file(tmp_path):
     s = run_job(job, ctx).summary()
     assert s["num_batches"] == 4
     assert s["total_examples"] == 2 * 64
+
+
+def test_keyless_bulk_loader_global_keys(tmp_path):
+    # NoneKeyBulkDataLoader parity: keyless records get globally unique
+    # sequential keys; ranks' ranges tile with no gaps or overlaps
+    p = tmp_path / "raw.txt"
+    p.write_text("".join(f"row{i}\n" for i in range(23)))
+    import torch
+
+    from harmony_amd.dataloader import load_keyless_split
+
+    allk, allv = [], []
+    for r in range(3):
+        k, v = load_keyless_split(str(p), r, 3)
+        assert k.numel() == len(v)
+        allk.append(k)
+        allv.extend(v)
+    cat = torch.cat(allk)
+    assert torch.equal(cat, torch.arange(23))
+    assert allv == [f"row{i}" for i in range(23)]
