@@ -291,3 +291,27 @@ def test_jobserver_restore_flow(tmp_path):
     assert r3["status"] == "failed", r3
     client.shutdown(port=port)
     t.join(timeout=30)
+
+
+def test_jobserver_pregel_submit():
+    # pregel jobs route through the jobserver dispatcher too
+    from harmony_amd.config import JobConfig, RuntimeConfig
+    from harmony_amd.jobserver import client
+    from harmony_amd.jobserver.server import JobServerDriver
+    from harmony_amd.runtime.bootstrap import init_executor
+
+    port = free_port()
+    ctx = init_executor(RuntimeConfig(device="cpu"))
+    driver = JobServerDriver(ctx, scheduler="default", port=port)
+    t = threading.Thread(target=driver.run, daemon=True)
+    t.start()
+    time.sleep(0.3)
+    job = JobConfig(job_id="pr1", app="pagerank", max_num_epochs=1,
+                    num_mini_batches=1,
+                    app_args={"num_vertices": 500, "out_degree": 4,
+                              "num_iters": 5})
+    r = client.submit(job, port=port, wait=True, timeout=60)
+    assert r["status"] == "done", r
+    assert r["per_rank"][0]["supersteps"] >= 5
+    client.shutdown(port=port)
+    t.join(timeout=30)
