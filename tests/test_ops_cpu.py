@@ -229,3 +229,32 @@ def test_gbt_hist_ref_shapes_and_totals():
         want = resid[node == nd].sum()
         got = s[nd, 0].sum()
         assert torch.allclose(got, want, atol=1e-4)
+
+
+def test_ops_refs_edge_shapes():
+    # torch reference implementations on degenerate shapes: single-row CSR,
+    # rank-1 factors, single-topic rows — guards refactors of the oracles
+    import torch
+
+    from harmony_amd import ops
+
+    # NMF: 1 row, 1 nonzero, rank 1
+    L = torch.tensor([[2.0]])
+    R = torch.tensor([[3.0]])
+    lg, rg, sq = ops.nmf_grad(L, R, torch.tensor([0, 1]),
+                              torch.tensor([0]), torch.tensor([5.0]), 0.0)
+    e = 2.0 * 3.0 - 5.0
+    assert abs(lg[0, 0] - 2 * e * 3.0) < 1e-6
+    assert abs(rg[0, 0] - 2 * e * 2.0) < 1e-6
+    assert abs(sq - e * e) < 1e-6
+
+    # softmax_grad_ce: 1 sample, 2 classes
+    p, loss, correct = ops.softmax_grad_ce(torch.tensor([[10.0, -10.0]]),
+                                           torch.tensor([0]))
+    assert int(correct) == 1 and float(loss) < 1e-3
+    assert abs(float(p[0, 0])) < 1e-3 and abs(float(p[0, 1])) < 1e-3
+
+    # segment_sum with all-duplicate keys
+    u, agg = ops.segment_sum(torch.tensor([7, 7, 7]),
+                             torch.ones(3, 2))
+    assert u.tolist() == [7] and agg.tolist() == [[3.0, 3.0]]
