@@ -27,7 +27,7 @@ def report(paths):
         ds.sort()
         n = len(ds)
         rows.append((job, name, rank, n, sum(ds),
-                     sum(ds) / n, ds[int(0.95 * (n - 1))]))
+                     sum(ds) / n, ds[min(n - 1, int(round(0.95 * n)))]))
     w = max((len(f"{j}/{nm}") for j, nm, *_ in rows), default=10)
     print(f"{'job/span':<{w}}  rank    n   total ms    mean ms     p95 ms")
     for job, name, rank, n, tot, mean, p95 in rows:
