@@ -100,6 +100,9 @@ class JobConfig:
     model_chkp_per_epoch: bool = False # snapshot model tables every epoch
     offline_model_eval: bool = False   # replay epoch snapshots after training
     chkp_path: str = "/tmp/harmony_chkp_temp"  # reference ChkpTempPath
+    chkp_commit_path: str = "/tmp/harmony_chkp_commit"  # ChkpCommitPath:
+                                       # temp checkpoints move here when the
+                                       # executor closes (two-phase commit)
     restore_chkp: Optional[str] = None # start with model tables restored from
                                        # this checkpoint id (reference
                                        # ETMaster.createTable(chkpId, ...))

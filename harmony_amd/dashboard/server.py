@@ -34,10 +34,15 @@ _PAGE = """<!doctype html><html><head><title>harmony_amd dashboard</title>
 <div id="root">loading...</div>
 <script>
 const sel=new URLSearchParams(location.search).get('job');
+// job ids arrive via unauthenticated POST /metrics: escape before any
+// innerHTML/href interpolation (stored-XSS hardening)
+const esc=s=>String(s).replace(/[&<>"']/g,c=>({'&':'&amp;','<':'&lt;',
+  '>':'&gt;','"':'&quot;',"'":'&#39;'}[c]));
 fetch('/jobs').then(r=>r.json()).then(jobs=>{
   document.getElementById('jobs').innerHTML='jobs: '+
     ['<a href="/">all</a>'].concat(jobs.map(j=>
-      `<a href="/?job=${j.job_id}">${j.job_id}</a> (${j.reports} reports, `+
+      `<a href="/?job=${encodeURIComponent(j.job_id)}">${esc(j.job_id)}</a>`+
+      ` (${j.reports|0} reports, `+
       `${new Date(j.t0*1000).toISOString().slice(0,19)})`)).join(' | ');
 });
 fetch('/data'+(sel?'?job='+sel:'')).then(r=>r.json()).then(rows=>{
@@ -53,7 +58,7 @@ fetch('/data'+(sel?'?job='+sel:'')).then(r=>r.json()).then(rows=>{
   for (const [job,rs] of Object.entries(byJob)) {
     const get=k=>rs.map(r=>{const p=JSON.parse(r.payload);return p[k]||0;});
     const rate=get('data_processing_rate'), et=get('epoch_time_sec');
-    html+=`<div class=m><b>${job}</b> (${rs.length} reports)<br>
+    html+=`<div class=m><b>${esc(job)}</b> (${rs.length} reports)<br>
       rate ex/s: ${chart(rate,'steelblue')}<br>`+
       (et.some(v=>v>0)?`epoch sec: ${chart(et,'indianred')}`:'')+`</div>`;
   }

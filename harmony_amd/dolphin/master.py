@@ -56,7 +56,8 @@ def _restore_tables(job: JobConfig, tables: dict) -> None:
     from harmony_amd.dolphin.model_eval import _safe
     from harmony_amd.et.checkpoint import CheckpointManager
 
-    cm = CheckpointManager(temp_root=job.chkp_path)
+    cm = CheckpointManager(temp_root=job.chkp_path,
+                           commit_root=job.chkp_commit_path)
     for t in tables.values():
         if not hasattr(t, "cfg"):
             continue
@@ -132,7 +133,8 @@ def run_job(job: JobConfig, ctx: ExecutorContext,
         from harmony_amd.dolphin.model_eval import ModelChkpManager
         from harmony_amd.et.checkpoint import CheckpointManager
 
-        cm = CheckpointManager(temp_root=job.chkp_path)
+        cm = CheckpointManager(temp_root=job.chkp_path,
+                               commit_root=job.chkp_commit_path)
         chkp_mgr = ModelChkpManager(cm, job.job_id, tables)
         orig_hook = trainer.on_epoch_finished
 

@@ -26,10 +26,14 @@ class ModelChkpManager:
         self.chkp_ids: List[str] = []
 
     def on_epoch_finished(self, epoch: int) -> None:
+        from harmony_amd.et.checkpoint import register_pending_commit
+
         cid = f"epoch{epoch}"
         for t in self.tables.values():
-            self.cm.checkpoint(t, self.app_id, f"{cid}/{_safe(t.cfg.table_id)}",
-                               ratio=self.ratio)
+            sub = f"{cid}/{_safe(t.cfg.table_id)}"
+            self.cm.checkpoint(t, self.app_id, sub, ratio=self.ratio)
+            # committed on executor close (reference ChkpManagerSlave:226)
+            register_pending_commit(self.cm, self.app_id, sub)
         if cid not in self.chkp_ids:
             self.chkp_ids.append(cid)
 

@@ -28,6 +28,32 @@ import torch
 from harmony_amd.config import TableConfig
 
 
+# checkpoints written during a server session, committed on executor close
+# (reference ChkpManagerSlave.commitAllLocalChkps, ChkpManagerSlave.java:226):
+# (temp_root, commit_root, app_id, chkp_id) tuples, process-local
+_pending_commits: "set[tuple]" = set()
+
+
+def register_pending_commit(mgr: "CheckpointManager", app_id: str,
+                            chkp_id: str) -> None:
+    _pending_commits.add((str(mgr.temp_root), str(mgr.commit_root),
+                          app_id, chkp_id))
+
+
+def commit_all_pending() -> int:
+    """Two-phase temp->commit of every registered checkpoint; returns how
+    many were committed. Call once per node on executor close (one mover:
+    commit is a directory move on the shared FS)."""
+    n = 0
+    for temp, commit, app_id, chkp_id in sorted(_pending_commits):
+        mgr = CheckpointManager(temp_root=temp, commit_root=commit)
+        if (mgr.temp_root / app_id / chkp_id).exists():
+            mgr.commit(app_id, chkp_id)
+            n += 1
+    _pending_commits.clear()
+    return n
+
+
 class CheckpointManager:
     def __init__(self, temp_root: str = "/tmp/harmony_chkp_temp",
                  commit_root: str = "/tmp/harmony_chkp_commit"):

@@ -19,10 +19,15 @@ isend/irecv, which gloo supports.
 
 from __future__ import annotations
 
+import itertools
 from typing import Dict, List, Optional, Tuple
 
 import torch
 import torch.distributed as dist
+
+# identity for route-cacheable key tensors: a fresh id per marked tensor —
+# data_ptr() could alias a freed tensor's reallocated address (advisor r01)
+_route_ids = itertools.count(1)
 
 
 class DataPlane:
@@ -165,8 +170,13 @@ class DataPlane:
         (`keys._harmony_static = True`, set by batch constructors): caching by
         data_ptr alone would alias freed/reallocated tensors."""
         cacheable = getattr(keys, "_harmony_static", False)
-        ck = (table.cfg.table_id, keys.data_ptr(), keys.numel(),
-              table.ownership.version) if cacheable else None
+        ck = None
+        if cacheable:
+            rid = getattr(keys, "_harmony_route_id", None)
+            if rid is None:
+                rid = next(_route_ids)
+                keys._harmony_route_id = rid
+            ck = (table.cfg.table_id, rid, table.ownership.version)
         if ck is not None:
             hit = self._route_cache.get(ck)
             if hit is not None:

@@ -217,10 +217,6 @@ class JobServerDriver:
                     group = (dist.new_group(ranks)
                              if dist.is_initialized() and
                              len(ranks) < self.ctx.world_size else None)
-                from harmony_amd.utils.joblog import job_logger
-
-                job_logger(job.job_id, self.ctx.rank).info(
-                    "scheduled on executors %s", ranks)
                 if self.ctx.rank in ranks:
                     view = JobView(rank=ranks.index(self.ctx.rank),
                                    world_size=len(ranks),
@@ -235,8 +231,12 @@ class JobServerDriver:
             if self.cp.flag_set("js/shutdown") and next_idx >= self.cp.read("js/njobs"):
                 break
             time.sleep(0.02)
+        # fail-fast: cancelled tasklets unwind via JobCancelled raised from
+        # control-plane waits, so a short join suffices; a clean shutdown
+        # waits for running jobs to finish
+        failed = self.cp.flag_set("js/failed")
         for t in self._threads:
-            t.join(timeout=600)
+            t.join(timeout=15 if failed else 600)
 
     # ------------------------------------------------- failure detection
 
@@ -273,6 +273,16 @@ class JobServerDriver:
         self.dispatch_loop()
         if dist.is_initialized():
             dist.barrier()
+        # executor close: commit temp checkpoints (reference ChkpManagerSlave
+        # commitAllLocalChkps on close, ChkpManagerSlave.java:226). After the
+        # barrier every rank's block files are on disk; rank 0 moves the dirs.
+        if self.ctx.is_master and not self.cp.flag_set("js/failed"):
+            from harmony_amd.et.checkpoint import commit_all_pending
+
+            n = commit_all_pending()
+            if n:
+                print(f"[jobserver] committed {n} checkpoint(s) on close",
+                      flush=True)
 
 
 def main() -> None:

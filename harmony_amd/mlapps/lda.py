@@ -187,6 +187,15 @@ class LDATrainer(Trainer):
 
     def pull_model(self) -> None:
         pulled = self.accessor.pull(self.batch.pull_keys)
+        from harmony_amd.dolphin.model_accessor import OneSidedAccessor
+
+        if isinstance(self.accessor, OneSidedAccessor):
+            # async pushes are per-CELL atomic, not per-row: a concurrent
+            # pull can see a row where the -1 landed but the +1 hasn't,
+            # i.e. transiently negative counts -> negative pmf mass in the
+            # samplers. Clamp at the consumer (reference's defensive clamp,
+            # LDAETModelUpdateFunction.java:43-64; advisor r01).
+            pulled = pulled.clamp_min_(0)
         self.word_topic = pulled[:-1]          # [n_uniq_words, K]
         self.topic_sum = pulled[-1]            # [K]
         if self.a["sampler"] == "alias":

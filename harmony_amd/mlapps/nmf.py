@@ -141,6 +141,7 @@ class NMFTrainer(Trainer):
         self.L = (torch.rand(num_local_rows, self.a["rank"], generator=g)
                   .to(ctx.device))
         self.step_size = self.a["step_size"]
+        self.use_pull_all = False   # set job-wide by build() (see pull_model)
         # device-resident loss accumulator: a per-batch float() would force
         # a host sync and serialize the async step pipeline
         self._sq_err = torch.zeros((), device=ctx.device)
@@ -164,9 +165,13 @@ class NMFTrainer(Trainer):
 
     def pull_model(self) -> None:
         b = self.batch
-        # when a batch touches most of the column space (dense-ish batches),
-        # one all-gather beats the two-sided all-to-all key exchange
-        if b.uniq_cols.shape[0] * 2 > self.a["num_cols"]:
+        # when batches touch most of the column space (dense-ish data), one
+        # all-gather beats the two-sided all-to-all key exchange. The choice
+        # MUST be identical on every rank (both branches are collectives; a
+        # per-rank data-dependent branch would issue mismatched RCCL
+        # collectives inside one NET phase and deadlock) — build() agrees on
+        # it once via the control store and sets self.use_pull_all.
+        if self.use_pull_all:
             pulled = self.accessor.pull_all()[b.uniq_cols]
         else:
             pulled = self.accessor.pull(b.uniq_cols)
@@ -232,6 +237,14 @@ def build(job: JobConfig, ctx, cp):
                           world_size=ctx.world_size, device=ctx.device,
                           tables={MODEL_TABLE: table}, app_args=job.app_args)
     trainer = NMFTrainer(tctx, rows_local)
+    # Agree job-wide (store-based, not a collective) whether batches are
+    # dense enough for the all-gather pull: max uniq-column count over every
+    # rank's blocks, decided ONCE — a per-rank/per-batch decision could put
+    # ranks of one job on different collectives (advisor r01, high).
+    local_max = max((b.uniq_cols.shape[0] for b in blocks), default=0)
+    agreed = cp.agree_max(f"{job.job_id}/nmf_uniqmax", int(local_max),
+                          n=ctx.world_size) if cp is not None else local_max
+    trainer.use_pull_all = agreed * 2 > defaults(job)["num_cols"]
     def _reslice(b, frac):
         # row-prefix re-slice for SetBatchShareOp: rebuild the static
         # precomputes (uniq_cols, col_sorted) for the smaller batch once
