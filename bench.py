@@ -129,7 +129,9 @@ class JobBench:
             batch = self.blocks[self._i % len(self.blocks)]
             self._i += 1
             self.trainer.set_batch_data(batch)
-            with self.tus.net(self.job.job_id, self._next_phase()):
+            # PULL draws PUSH's ticket too (one store round-trip per step)
+            with self.tus.net(self.job.job_id, self._next_phase(),
+                              lookahead=1):
                 self.trainer.pull_model()
             self.trainer.local_compute()
             with self.tus.net(self.job.job_id, self._next_phase()):
@@ -151,6 +153,8 @@ def main():
     multi = len(jobs) > 1
     tus = TaskUnitScheduler(cp, {j.job_id for j in jobs.values()},
                             multi_job=multi)
+    for j in jobs.values():
+        tus.set_drawer(j.job_id, rank == 0)
     benches = [JobBench(j, ctx, cp, tus, use_stream=dev_cuda)
                for j in jobs.values()]
     for b in benches:

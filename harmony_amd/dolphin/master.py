@@ -151,8 +151,17 @@ def run_job(job: JobConfig, ctx: ExecutorContext,
     # NET ticket or two jobs' post-run evals can interleave on the wire.
     # All ranks reach here after the job's cleanup barrier, so a fixed
     # out-of-band phase index is requested in the same order everywhere.
+    import contextlib
+
     _EVAL_PHASE = 1 << 30
-    with tasklet.tus.net(job.job_id, _EVAL_PHASE):
+
+    def _eval_net(idx):
+        # async (one-sided) jobs issue kernels, not collectives: no ticket
+        if tasklet.is_async:
+            return contextlib.nullcontext()
+        return tasklet.tus.net(job.job_id, idx)
+
+    with _eval_net(_EVAL_PHASE):
         ev = trainer.evaluate_model()
     for k, v in (ev or {}).items():
         metrics.add_custom(k, float(v))
@@ -161,7 +170,7 @@ def run_job(job: JobConfig, ctx: ExecutorContext,
 
         evaluator = ModelEvaluator(chkp_mgr.cm, job.job_id, tables, trainer,
                                    provider)
-        with tasklet.tus.net(job.job_id, _EVAL_PHASE + 1):
+        with _eval_net(_EVAL_PHASE + 1):
             offline = evaluator.evaluate_all(chkp_mgr.chkp_ids)
         for cid, res in offline.items():
             for k, v in (res or {}).items():

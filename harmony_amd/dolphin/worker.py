@@ -48,6 +48,23 @@ class WorkerTasklet:
         self.metrics = MetricCollector(job.job_id, rank)
         self._phase = 0
         self.ssp = SSPClock(cp, job.job_id, world_size, job.clock_slack)
+        # Async (one-sided) jobs: pulls/pushes are xGMI kernels, not
+        # collectives -> no NET tickets needed, and the SSP slack wait may
+        # genuinely block (bounded-async). Collective jobs: ranks are
+        # lockstepped by the collectives themselves (skew <= 1 batch), so
+        # the SSP wait is vacuous AND, if allowed to block, can deadlock
+        # against the global ticket order (see SSPClock.tick_and_wait).
+        from harmony_amd.dolphin.model_accessor import OneSidedAccessor
+
+        self.is_async = isinstance(getattr(trainer, "accessor", None),
+                                   OneSidedAccessor)
+
+    def _net(self, jid: str, lookahead: int = 0):
+        """lookahead=1 on PULL also draws PUSH's ticket in the same store
+        round-trip (both phases are guaranteed to run for every batch)."""
+        if self.is_async:
+            return contextlib.nullcontext()
+        return self.tus.net(jid, self._next_phase(), lookahead)
 
     def _next_phase(self) -> int:
         self._phase += 1
@@ -75,7 +92,7 @@ class WorkerTasklet:
         with stream_ctx:
             # initialize() may issue collectives (e.g. LDA's initial count
             # push) — serialize it like any NET phase.
-            with self.tus.net(jid, self._next_phase()):
+            with self._net(jid):
                 self.trainer.initialize()
             # INIT -> RUN global barrier (reference WorkerGlobalBarrier)
             self.cp.barrier(f"{jid}/run", self.world_size)
@@ -87,7 +104,8 @@ class WorkerTasklet:
                 ep_examples = 0
                 for bidx, batch in enumerate(self.provider.epoch_iter(epoch)):
                     # SYNC: SSP clock (reference MiniBatchBarrier -> master)
-                    if not self.ssp.tick_and_wait(self.rank):
+                    if not self.ssp.tick_and_wait(self.rank,
+                                                  wait=self.is_async):
                         stopped = True
                         break
                     b_t0 = time.perf_counter()
@@ -95,7 +113,7 @@ class WorkerTasklet:
                     # PULL
                     t0 = time.perf_counter()
                     with self.tracer.span("pull"):
-                        with self.tus.net(jid, self._next_phase()):
+                        with self._net(jid, lookahead=1):
                             self.trainer.pull_model()
                     pull_t = time.perf_counter() - t0
                     # COMP
@@ -106,7 +124,7 @@ class WorkerTasklet:
                     # PUSH
                     t0 = time.perf_counter()
                     with self.tracer.span("push"):
-                        with self.tus.net(jid, self._next_phase()):
+                        with self._net(jid):
                             self.trainer.push_update()
                     push_t = time.perf_counter() - t0
                     n = self.trainer.num_batch_examples()
@@ -124,7 +142,7 @@ class WorkerTasklet:
                         self.orch.report_batch(bt, comp_t, pull_t, push_t, n)
                         plan = self.orch.boundary_plan()
                         if plan is not None:
-                            with self.tus.net(jid, self._next_phase()):
+                            with self._net(jid):
                                 self.orch.apply(plan)
                             self._consume_shares()
                 self.trainer.on_epoch_finished(epoch)

@@ -66,6 +66,7 @@ class JobServerDriver:
         self.ctx = ctx
         self.cp = ControlPlane(ctx.store, ctx.rank, ctx.world_size)
         self.tus = TaskUnitScheduler(self.cp, {ADMIN_JOB}, multi_job=True)
+        self.tus.set_drawer(ADMIN_JOB, ctx.rank == 0)
         self.pool = ResourcePool(ctx.world_size)
         self.scheduler = load_scheduler(scheduler)
         self.port = port
@@ -212,6 +213,9 @@ class JobServerDriver:
                 if self.ctx.rank in ranks:
                     my_jobs.add(job.job_id)
                     self.tus.set_jobs(my_jobs)
+                    # job-local rank 0 draws this job's NET tickets
+                    self.tus.set_drawer(job.job_id,
+                                        self.ctx.rank == ranks[0])
                 # ordered group creation: an admin NET phase (all ranks)
                 with self.tus.net(ADMIN_JOB, next_idx + 1):
                     group = (dist.new_group(ranks)

@@ -165,6 +165,12 @@ class ThreadLocalTCPStore:
     def delete_key(self, key):
         return self._client().delete_key(key)
 
+    def multi_get(self, keys):
+        return self._client().multi_get(keys)
+
+    def multi_set(self, keys, values):
+        return self._client().multi_set(keys, values)
+
 
 class LocalStore:
     """In-process stand-in for TCPStore in single-process mode (tests, local
@@ -224,6 +230,15 @@ class LocalStore:
     def delete_key(self, key: str) -> bool:
         with self._cv:
             return self._d.pop(key, None) is not None
+
+    def multi_get(self, keys):
+        return [self.get(k) for k in keys]
+
+    def multi_set(self, keys, values) -> None:
+        with self._cv:
+            for k, v in zip(keys, values):
+                self._d[k] = _to_bytes(v)
+            self._cv.notify_all()
 
 
 def _to_bytes(v) -> bytes:
