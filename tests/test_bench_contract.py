@@ -32,3 +32,23 @@ def test_bench_json_contract():
     # value is the WHOLE-JOB aggregate: examples per step x steps / elapsed
     assert abs(d["value"] * d["makespan_sec"]
                - cfg["global_batch"] * d["steps"]) < 1e-3 * d["value"]
+
+
+def test_bench_runtime_mode_contract():
+    """--mode runtime drives the same jobs through run_job/WorkerTasklet
+    (benchmark through the real runtime)."""
+    out = subprocess.run(
+        [sys.executable, "bench.py", "--mode", "runtime", "--steps", "2",
+         "--nmf-cols", "512", "--nmf-rows-per-batch", "64",
+         "--nmf-nnz-per-row", "8", "--mlr-features", "64",
+         "--mlr-batch", "64", "--lda-vocab", "500",
+         "--lda-docs-per-batch", "32", "--lda-tokens-per-doc", "8"],
+        capture_output=True, text=True, timeout=300)
+    assert out.returncode == 0, out.stderr[-2000:]
+    line = out.stdout.strip().splitlines()[-1]
+    d = json.loads(line)
+    assert d["metric"] == "aggregate_examples_per_sec_3job_runtime"
+    assert d["value"] > 0
+    for job in ("nmf", "mlr", "lda"):
+        pj = d["per_job"][job]
+        assert pj["examples_per_sec"] > 0 and pj["timed_epochs"] >= 1
