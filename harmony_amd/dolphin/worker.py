@@ -146,6 +146,12 @@ class WorkerTasklet:
                                 self.orch.apply(plan)
                             self._consume_shares()
                 self.trainer.on_epoch_finished(epoch)
+                # Per-batch spans above are ENQUEUE-side on GPU (streams
+                # pipeline); the epoch metric must reflect COMPLETED work
+                # (reference dataProcessingRate is wall-clock per processed
+                # items), so drain this job's stream before timing.
+                if self.stream is not None:
+                    self.stream.synchronize()
                 ep_dt = time.perf_counter() - ep_t0
                 self.metrics.add_epoch(EpochMetrics(
                     epoch_idx=epoch, num_examples=ep_examples,

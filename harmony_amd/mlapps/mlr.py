@@ -129,6 +129,14 @@ class MLRTrainer(Trainer):
         C, F = self.a["num_classes"], self.a["num_features"]
         return self.W.view(C, F)
 
+    def _wt_buf(self) -> torch.Tensor:
+        # persistent padded W^T staging buffer for the MFMA step kernel
+        if getattr(self, "_wt", None) is None:
+            F = self.a["num_features"]
+            self._wt = torch.zeros((F, 16), dtype=torch.float32,
+                                   device=self.ctx.device)
+        return self._wt
+
     def pull_model(self) -> None:
         C, P = self.a["num_classes"], self.a["num_parts_per_class"]
         pulled = self.accessor.pull_all()[:C * P]
@@ -144,14 +152,32 @@ class MLRTrainer(Trainer):
         body = self._bodies.get(key)
         if body is None:
             W = self._w_matrix()                   # [C, F] view of W_buf
+            import os as _os
 
-            def body(x=x, y=y, W=W):
-                # forward GEMM + fused softmax + grad GEMM (one graph launch)
-                p, loss, correct = ops.mlr_forward(x, W, y)
-                g = ops.mlr_grad_gemm(p, x)        # [C, F]
-                self.grad_raw = g / x.shape[0] + self.a["lam"] * W
-                self._loss_sum += loss
-                self._correct += correct
+            use_mfma = (x.device.type == "cuda"
+                        and ops.mlr_step_ok(x, self.a["num_classes"])
+                        and _os.environ.get("HARMONY_MLR_MFMA", "1") != "0")
+            rb = int(_os.environ.get("HARMONY_MLR_RB", "2048"))
+            if rb and (x.shape[0] % rb or rb % 64):
+                rb = 0
+
+            if use_mfma:
+                # fused fwd+softmax+grad on f32 MFMA, L3 row-blocked
+                # (K4-MFMA, ops/csrc/mlr_mfma.hip; A/B scripts/mlr_mfma_ab.py)
+                def body(x=x, y=y, W=W, rb=rb):
+                    g, loss, correct = ops.mlr_step_mfma(
+                        x, W, y, row_block=rb, Wt_buf=self._wt_buf())
+                    self.grad_raw = g / x.shape[0] + self.a["lam"] * W
+                    self._loss_sum += loss
+                    self._correct += correct
+            else:
+                def body(x=x, y=y, W=W):
+                    # forward GEMM + fused softmax + grad GEMM
+                    p, loss, correct = ops.mlr_forward(x, W, y)
+                    g = ops.mlr_grad_gemm(p, x)        # [C, F]
+                    self.grad_raw = g / x.shape[0] + self.a["lam"] * W
+                    self._loss_sum += loss
+                    self._correct += correct
 
             self._bodies[key] = body
         self._graphs.run(key, body, state=(self._loss_sum, self._correct))

@@ -94,6 +94,56 @@ def mlr_grad_gemm(P: torch.Tensor, X: torch.Tensor) -> torch.Tensor:
     return P.t() @ X
 
 
+def mlr_step_ok(X: torch.Tensor, C: int) -> bool:
+    """Shapes the fused MFMA step kernel supports (see mlr_mfma.hip)."""
+    B, F = X.shape
+    return (C <= 16 and F % 64 == 0 and B % 64 == 0
+            and X.dtype == torch.float32)
+
+
+def _pow2_div(limit: int, n: int) -> int:
+    """Largest power of two <= limit that divides n."""
+    p = 1
+    while p * 2 <= limit and n % (p * 2) == 0:
+        p *= 2
+    return p
+
+
+def mlr_step_mfma(X: torch.Tensor, W: torch.Tensor, labels: torch.Tensor,
+                  row_block: int = 0, splitf: Optional[int] = None,
+                  splitb: Optional[int] = None,
+                  Wt_buf: Optional[torch.Tensor] = None
+                  ) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
+    """One fused MLR step on the gfx950 matrix cores (K4-MFMA,
+    ops/csrc/mlr_mfma.hip): logits = X @ W^T via v_mfma_f32_16x16x4_f32,
+    in-kernel softmax - onehot, grad = P^T @ X via MFMA with split-B
+    atomics. With row_block > 0 the fwd/grad pair runs per row block so the
+    grad pass re-reads X from the Infinity Cache (the step is X-bandwidth
+    bound). Returns (grad [C,F], loss_sum, n_correct) — exact-f32 MFMA
+    numerics (a k-ordered fmaf chain).
+
+    Replaces reference MLRTrainer.java:374-398 + :475-489 + the rocBLAS
+    GEMM pair (measured A/B: scripts/mlr_mfma_ab.py)."""
+    C, F = W.shape
+    if not _use_hip(X):                # CPU oracle path
+        p, loss, correct = mlr_forward(X, W, labels)
+        return mlr_grad_gemm(p, X), loss, correct
+    B = X.shape[0]
+    rows = row_block if row_block > 0 else B
+    if splitf is None:
+        # target ~512 workgroups per launch (2 blocks/CU on 256 CUs)
+        splitf = _pow2_div(max(1, 512 // (rows // 64)), F // 64)
+    if splitb is None:
+        splitb = _pow2_div(max(1, 512 // (F // 64)) * 2, rows // 4)
+    if Wt_buf is None:
+        Wt_buf = torch.zeros((F, 16), dtype=X.dtype, device=X.device)
+    Wt_buf[:, :C] = W.t()
+    gradT, loss, correct = _hip.mlr_step_mfma(
+        X.contiguous(), Wt_buf, labels.contiguous(), row_block, C,
+        splitf, splitb)
+    return gradT[:C], loss[0], correct[0].to(torch.int64)
+
+
 def softmax_grad_ce(logits: torch.Tensor, labels: torch.Tensor
                     ) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
     """Row softmax with log-sum-exp guard; returns (p - onehot(label),

@@ -16,6 +16,10 @@ std::vector<torch::Tensor> nmf_grad_twopass(
 std::vector<torch::Tensor> mlr_fwd(torch::Tensor X, torch::Tensor W,
                                    torch::Tensor labels);
 torch::Tensor mlr_grad(torch::Tensor P, torch::Tensor X);
+std::vector<torch::Tensor> mlr_step_mfma(torch::Tensor X, torch::Tensor Wt,
+                                         torch::Tensor labels,
+                                         int64_t row_block, int64_t C,
+                                         int64_t splitf, int64_t splitb);
 torch::Tensor lda_gibbs(torch::Tensor doc_topic, torch::Tensor word_topic,
                         torch::Tensor topic_sum, torch::Tensor doc_offsets,
                         torch::Tensor word_ids, torch::Tensor assignments,
@@ -66,6 +70,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mlr_fwd", &mlr_fwd,
         "fused MLR forward: X@W^T + softmax + grad + CE/acc (K4)");
   m.def("mlr_grad", &mlr_grad, "skinny-C gradient GEMM P^T @ X (K5)");
+  m.def("mlr_step_mfma", &mlr_step_mfma,
+        "fused MLR step on f32 MFMA: fwd+softmax+grad, optional L3 "
+        "row-blocking (K4-MFMA)");
   m.def("lda_gibbs", &lda_gibbs, "LDA collapsed Gibbs sweep (K7)");
   m.def("lda_apply_pairs", &lda_apply_pairs,
         "apply TopicChanges +/-1 pairs to the word-topic shard (K9 sparse)");
