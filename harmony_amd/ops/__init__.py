@@ -144,6 +144,76 @@ def mlr_step_mfma(X: torch.Tensor, W: torch.Tensor, labels: torch.Tensor,
     return gradT[:C], loss[0], correct[0].to(torch.int64)
 
 
+_mlr_aux_streams = {}
+
+
+def mlr_step_mfma_pipelined(X: torch.Tensor, W: torch.Tensor,
+                            labels: torch.Tensor, row_block: int = 2048,
+                            splitf: Optional[int] = None,
+                            splitb: Optional[int] = None,
+                            Wt_buf: Optional[torch.Tensor] = None
+                            ) -> Tuple[torch.Tensor, torch.Tensor,
+                                       torch.Tensor]:
+    """Row-blocked K4-MFMA step with a two-stream software pipeline:
+    fwd+softmax of block i+1 (stream A) overlaps grad of block i (stream
+    B). The grad pass then re-reads its X block from the 256 MiB Infinity
+    Cache while fwd streams the next block from HBM — the step's HBM
+    traffic drops from 2 passes over X toward 1 (the step is X-bandwidth
+    bound). Falls back to the single-stream path when shapes don't block
+    evenly."""
+    C, F = W.shape
+    if not _use_hip(X):
+        return mlr_step_mfma(X, W, labels)
+    B = X.shape[0]
+    if row_block <= 0 or B % row_block or row_block % 64 or B == row_block:
+        return mlr_step_mfma(X, W, labels, row_block=0, splitf=splitf,
+                             splitb=splitb, Wt_buf=Wt_buf)
+    if splitf is None:
+        splitf = _pow2_div(max(1, 512 // (row_block // 64)), F // 64)
+    if splitb is None:
+        splitb = _pow2_div(max(2, 1024 // (F // 64)), row_block // 64)
+    if Wt_buf is None:
+        Wt_buf = torch.zeros((F, 16), dtype=X.dtype, device=X.device)
+    Wt_buf[:, :C] = W.t()
+    dev = X.device
+    key = dev.index
+    aux = _mlr_aux_streams.get(key)
+    if aux is None:
+        aux = (torch.cuda.Stream(device=dev), torch.cuda.Stream(device=dev))
+        _mlr_aux_streams[key] = aux
+    sA, sB = aux
+    X = X.contiguous()
+    labels = labels.contiguous()
+    P = torch.zeros((B, 16), dtype=X.dtype, device=dev)
+    gradT = torch.zeros((16, F), dtype=X.dtype, device=dev)
+    loss = torch.zeros((1,), dtype=X.dtype, device=dev)
+    correct = torch.zeros((1,), dtype=torch.int32, device=dev)
+    cur = torch.cuda.current_stream(dev)
+    start = torch.cuda.Event()
+    start.record(cur)
+    sA.wait_event(start)
+    sB.wait_event(start)
+    for off in range(0, B, row_block):
+        ev = torch.cuda.Event()
+        with torch.cuda.stream(sA):
+            _hip.mlr_fwd_mfma_part(X, Wt_buf, P, off, row_block, splitf)
+            _hip.mlr_softmax_part(P, labels, loss, correct, off, row_block,
+                                  C)
+            ev.record(sA)
+        sB.wait_event(ev)
+        with torch.cuda.stream(sB):
+            _hip.mlr_grad_mfma_part(P, X, gradT, off, row_block, splitb)
+    endA, endB = torch.cuda.Event(), torch.cuda.Event()
+    endA.record(sA)
+    endB.record(sB)
+    cur.wait_event(endA)
+    cur.wait_event(endB)
+    for t in (X, Wt_buf, P, gradT, loss, correct, labels):
+        t.record_stream(sA)
+        t.record_stream(sB)
+    return gradT[:C], loss[0], correct[0].to(torch.int64)
+
+
 def softmax_grad_ce(logits: torch.Tensor, labels: torch.Tensor
                     ) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
     """Row softmax with log-sum-exp guard; returns (p - onehot(label),

@@ -20,6 +20,14 @@ std::vector<torch::Tensor> mlr_step_mfma(torch::Tensor X, torch::Tensor Wt,
                                          torch::Tensor labels,
                                          int64_t row_block, int64_t C,
                                          int64_t splitf, int64_t splitb);
+void mlr_fwd_mfma_part(torch::Tensor X, torch::Tensor Wt, torch::Tensor P,
+                       int64_t off, int64_t n, int64_t splitf);
+void mlr_softmax_part(torch::Tensor P, torch::Tensor labels,
+                      torch::Tensor loss, torch::Tensor correct,
+                      int64_t off, int64_t n, int64_t C);
+void mlr_grad_mfma_part(torch::Tensor P, torch::Tensor X,
+                        torch::Tensor gradT, int64_t off, int64_t n,
+                        int64_t splitb);
 torch::Tensor lda_gibbs(torch::Tensor doc_topic, torch::Tensor word_topic,
                         torch::Tensor topic_sum, torch::Tensor doc_offsets,
                         torch::Tensor word_ids, torch::Tensor assignments,
@@ -73,6 +81,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mlr_step_mfma", &mlr_step_mfma,
         "fused MLR step on f32 MFMA: fwd+softmax+grad, optional L3 "
         "row-blocking (K4-MFMA)");
+  m.def("mlr_fwd_mfma_part", &mlr_fwd_mfma_part, "K4-MFMA fwd piece");
+  m.def("mlr_softmax_part", &mlr_softmax_part, "K4-MFMA softmax piece");
+  m.def("mlr_grad_mfma_part", &mlr_grad_mfma_part, "K4-MFMA grad piece");
   m.def("lda_gibbs", &lda_gibbs, "LDA collapsed Gibbs sweep (K7)");
   m.def("lda_apply_pairs", &lda_apply_pairs,
         "apply TopicChanges +/-1 pairs to the word-topic shard (K9 sparse)");

@@ -37,7 +37,11 @@ def main():
     ap.add_argument("--B", type=int, default=16384)
     ap.add_argument("--F", type=int, default=16384)
     ap.add_argument("--C", type=int, default=10)
+    ap.add_argument("--sweep", action="store_true")
     args = ap.parse_args()
+    if args.sweep:
+        sweep()
+        return
     torch.manual_seed(0)
     dev = "cuda"
     X = torch.randn(args.B, args.F, device=dev)
@@ -64,11 +68,48 @@ def main():
                                                    Wt_buf=wt), args.iters)
         out[f"mfma_rb{rb}_ms"] = ms
         out[f"mfma_rb{rb}_grad_rel"] = rel
+    for rb in (1024, 2048, 4096):
+        if args.B % rb:
+            continue
+        g1, l1, c1 = ops.mlr_step_mfma_pipelined(X, W, y, row_block=rb,
+                                                 Wt_buf=wt)
+        rel = ((g1 - g0).abs().max() / g0.abs().max()).item()
+        assert rel < 1e-3, f"pipe grad mismatch rb={rb}: {rel}"
+        assert int(c1) == int(c0)
+        out[f"mfma_pipe_rb{rb}_ms"] = bench(
+            lambda rb=rb: ops.mlr_step_mfma_pipelined(X, W, y, row_block=rb,
+                                                      Wt_buf=wt), args.iters)
     xbytes = args.B * args.F * 4
     out["hbm_floor_ms_2pass"] = 2 * xbytes / 6.3e12 * 1e3
     out["hbm_floor_ms_1pass"] = xbytes / 6.3e12 * 1e3
     for k, v in out.items():
         print(f"{k:26s} {v:.4f}" if isinstance(v, float) else f"{k}: {v}")
+
+
+def sweep():
+    """Split/occupancy sweep + per-kernel timing (run with --sweep)."""
+    import itertools
+    torch.manual_seed(0)
+    dev = "cuda"
+    B = F = 16384
+    X = torch.randn(B, F, device=dev)
+    W = torch.randn(10, F, device=dev) * 0.01
+    y = torch.randint(0, 10, (B,), device=dev)
+    wt = torch.zeros((F, 16), device=dev)
+    for rb in (0, 1024, 2048):
+        for sf, sb in itertools.product((1, 2, 4, 8, 16, 32),
+                                        (2, 4, 8, 16, 32)):
+            if rb == 0 and (sf > 8 or sb > 16):
+                continue
+            if rb and rb // 64 < sb:
+                continue
+            try:
+                ms = bench(lambda: ops.mlr_step_mfma(
+                    X, W, y, row_block=rb, splitf=sf, splitb=sb, Wt_buf=wt),
+                    12)
+                print(f"rb={rb:5d} splitf={sf} splitb={sb:2d}  {ms:.4f} ms")
+            except Exception as e:
+                print(f"rb={rb} sf={sf} sb={sb}: {str(e)[:60]}")
 
 
 if __name__ == "__main__":
