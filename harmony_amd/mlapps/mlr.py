@@ -154,10 +154,14 @@ class MLRTrainer(Trainer):
             W = self._w_matrix()                   # [C, F] view of W_buf
             import os as _os
 
+            # default: rocBLAS pair (measured at the fetch roofline and
+            # 4% faster than the in-tree MFMA kernels both isolated and
+            # in-bench — profiles/r02_mlr_mfma.md); HARMONY_MLR_MFMA=1
+            # switches to the hand-written K4-MFMA path
             use_mfma = (x.device.type == "cuda"
                         and ops.mlr_step_ok(x, self.a["num_classes"])
-                        and _os.environ.get("HARMONY_MLR_MFMA", "1") != "0")
-            rb = int(_os.environ.get("HARMONY_MLR_RB", "2048"))
+                        and _os.environ.get("HARMONY_MLR_MFMA", "0") == "1")
+            rb = int(_os.environ.get("HARMONY_MLR_RB", "0"))
             if rb and (x.shape[0] % rb or rb % 64):
                 rb = 0
 

@@ -131,10 +131,11 @@ def mlr_step_mfma(X: torch.Tensor, W: torch.Tensor, labels: torch.Tensor,
     B = X.shape[0]
     rows = row_block if row_block > 0 else B
     if splitf is None:
-        # target ~512 workgroups per launch (2 blocks/CU on 256 CUs)
+        # target ~512 workgroups per launch (2 blocks/CU on 256 CUs) —
+        # the sweep's optimum on the bench shape (scripts/mlr_mfma_ab.py)
         splitf = _pow2_div(max(1, 512 // (rows // 64)), F // 64)
     if splitb is None:
-        splitb = _pow2_div(max(1, 512 // (F // 64)) * 2, rows // 4)
+        splitb = _pow2_div(max(1, 512 // (F // 64)), rows // 64)
     if Wt_buf is None:
         Wt_buf = torch.zeros((F, 16), dtype=X.dtype, device=X.device)
     Wt_buf[:, :C] = W.t()
