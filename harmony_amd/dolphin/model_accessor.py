@@ -112,6 +112,13 @@ class OneSidedAccessor:
     def pull(self, keys):
         return self.table.pull(keys)
 
+    def drain(self):
+        # owner-side apply-queue drain (one-sided v2 rings; no-op for
+        # add-algebra tables) — called once per batch after push
+        if hasattr(self.table, "drain"):
+            return self.table.drain()
+        return 0
+
     def push_dense(self, grad_full):
         if self._keys is None or self._keys.shape[0] != grad_full.shape[0]:
             import torch as _t

@@ -42,20 +42,46 @@ class OneSidedTable(Table):
     """Async-access dense table. Construct on every rank, then `connect()`
     collectively (exchanges IPC handles through the control store)."""
 
+    # update fns applied by direct system-scope atomics (v1 path):
+    #   add        — atomic float/int add IS the update
+    #   lda_counts — ±1 count deltas; the reference's clamp>=0 is a
+    #                defensive no-op (counts conserved by construction,
+    #                docs/ROADMAP.md), so atomic int add is exact
+    _ATOMIC_FNS = ("add", "lda_counts")
+
     def __init__(self, cfg: TableConfig, rank: int, world_size: int,
-                 device: torch.device, store=None):
-        # update fns with pure-add merge algebra work one-sided:
-        #   add        — atomic float/int add IS the update
-        #   lda_counts — ±1 count deltas; the reference's clamp>=0 is a
-        #                defensive no-op (counts conserved by construction,
-        #                docs/ROADMAP.md), so atomic int add is exact
-        assert cfg.update_fn in ("add", "lda_counts"), \
-            "one-sided push is atomic add: update_fn must be add-algebra"
-        assert cfg.dtype in ("float32", "int32")
+                 device: torch.device, store=None,
+                 ring_capacity: int = 8192):
+        from harmony_amd import ops as _ops
+
+        # NON-add update fns (NMF's clamp(old - step*delta), assign) go
+        # through owner-side apply-queue RINGS (v2, ops/csrc/os_ring.hip):
+        # writers enqueue (key, delta) asynchronously; the owner alone
+        # drains and applies — the reference's per-block op-queue write
+        # serialization (CommManager.java:36-155).
+        self._ring_mode = cfg.update_fn not in self._ATOMIC_FNS
+        if self._ring_mode:
+            assert _ops.fused_apply_supported(cfg.update_fn), \
+                f"one-sided v2 needs a device apply mode for {cfg.update_fn}"
+            assert cfg.dtype == "float32", "ring payloads are f32"
+        else:
+            assert cfg.dtype in ("float32", "int32")
         self._hip = _require_hip()
         super().__init__(cfg, rank, world_size, device)
         self.store = store
+        self._ring_cap = int(ring_capacity)
         self._peer_ptr = {}          # rank -> mapped device pointer (int)
+        self._ring_peer = {}         # rank -> mapped RING pointer (int)
+        if self._ring_mode:
+            vd = cfg.value_dim
+            nwords = (self._hip.os_ring_bytes(world_size, self._ring_cap,
+                                              vd) + 3) // 4
+            self._ring_buf = self._hip.os_shard_alloc(nwords, 1, 0)
+            # host-side exact backpressure state (we are the only writer
+            # of our slot in each peer's ring)
+            self._pushed = {r: 0 for r in range(world_size)}
+            self._head_cache = {r: 0 for r in range(world_size)}
+            self._scratch = torch.zeros(2, dtype=torch.int64, device=device)
         # per-table op stats (reference RemoteAccessOpStat): pulled rows /
         # pushed rows / remote bytes moved over xGMI
         self.stats = {"pull_rows": 0, "push_rows": 0, "remote_bytes": 0}
@@ -94,13 +120,16 @@ class OneSidedTable(Table):
         self._block_slot = slot.to(self.device)
 
     def connect(self, store=None) -> None:
-        """Collective: export my shard, map every peer's (via the store)."""
+        """Collective: export my shard (+ ring), map every peer's."""
         store = store or self.store
         assert store is not None
         torch.cuda.synchronize()
-        h = self._hip.os_ipc_handle(self.shard)
         key = f"os/{self.cfg.table_id}"
+        h = self._hip.os_ipc_handle(self.shard)
         store.set(f"{key}/{self.rank}", bytes(h.tolist()).hex())
+        if self._ring_mode:
+            hr = self._hip.os_ipc_handle(self._ring_buf)
+            store.set(f"{key}/ring/{self.rank}", bytes(hr.tolist()).hex())
         for r in range(self.world_size):
             if r == self.rank:
                 continue
@@ -108,11 +137,19 @@ class OneSidedTable(Table):
             hb = torch.tensor(list(bytes.fromhex(raw.decode())),
                               dtype=torch.uint8)
             self._peer_ptr[r] = self._hip.os_ipc_open(hb)
+            if self._ring_mode:
+                raw = store.get(f"{key}/ring/{r}")
+                hb = torch.tensor(list(bytes.fromhex(raw.decode())),
+                                  dtype=torch.uint8)
+                self._ring_peer[r] = self._hip.os_ipc_open(hb)
 
     def close(self) -> None:
         for p in self._peer_ptr.values():
             self._hip.os_ipc_close(p)
         self._peer_ptr.clear()
+        for p in self._ring_peer.values():
+            self._hip.os_ipc_close(p)
+        self._ring_peer.clear()
 
     # ------------------------------------------------------------ async ops
 
@@ -153,7 +190,9 @@ class OneSidedTable(Table):
         return out
 
     def push(self, keys: torch.Tensor, deltas: torch.Tensor) -> None:
-        """Async push: atomicAdd scatter into each owner's HBM."""
+        """Async push. add-algebra fns: atomicAdd scatter into each
+        owner's HBM (v1). Other fns: enqueue into each owner's apply-queue
+        ring (v2) — the owner applies on its next drain()."""
         keys = keys.to(self.device, torch.int64)
         self.stats["push_rows"] += keys.shape[0]
         deltas = deltas.to(self.device, self.dtype).contiguous()
@@ -162,18 +201,112 @@ class OneSidedTable(Table):
             sel = owner == r
             if not bool(sel.any()):
                 continue
-            rows = self._rows_on(r, keys[sel])
+            k = keys[sel]
             d = deltas[sel].contiguous()
+            if r != self.rank:
+                self.stats["remote_bytes"] += (k.shape[0]
+                                               * self.cfg.value_dim * 4)
+            if self._ring_mode:
+                if r == self.rank:
+                    # owner-local fast path: the owner is the ONLY applier,
+                    # and this thread is the owner's applier thread, so a
+                    # direct apply preserves the per-block serialization
+                    self._apply_local(k, d)
+                else:
+                    self._ring_push_to(r, k, d)
+                continue
+            rows = self._rows_on(r, k)
             # local pushes use the SAME system-scope atomic kernel as
             # remote ones: torch index_add_ is a plain read-modify-write,
             # and racing it against another process's atomics on the same
             # cells can drop updates
-            if r != self.rank:
-                self.stats["remote_bytes"] += (rows.shape[0]
-                                               * self.cfg.value_dim * 4)
             ptr = (self.shard.data_ptr() if r == self.rank
                    else self._peer_ptr[r])
             self._hip.os_scatter_add(ptr, rows, d)
+
+    # --------------------------------------------------- v2 ring plumbing
+
+    def _apply_local(self, keys: torch.Tensor, deltas: torch.Tensor) -> None:
+        """Apply update_fn(shard[key], delta) IN ORDER. Duplicate keys in
+        one batch are applied in occurrence rounds (scatter_apply assumes
+        unique rows; a concurrent read-modify-write on duplicates would
+        lose updates) — f(f(v,d1),d2), the reference's sequential op-queue
+        semantics."""
+        from harmony_amd import ops as _ops
+
+        ua = self.cfg.update_args or {}
+        kw = dict(step_size=float(ua.get("step_size", 0.0)),
+                  max_val=float(ua.get("max_val", 0.0)))
+        rows = self._rows_on(self.rank, keys)
+        order = torch.argsort(rows, stable=True)
+        sr = rows[order]
+        # occurrence index of each item within its key group
+        uniq, inv, counts = torch.unique_consecutive(
+            sr, return_inverse=True, return_counts=True)
+        starts = torch.zeros_like(counts)
+        starts[1:] = counts.cumsum(0)[:-1]
+        occ = torch.arange(sr.numel(), device=sr.device) - starts[inv]
+        max_occ = int(counts.max()) if counts.numel() else 0
+        if max_occ <= 1:
+            _ops.scatter_apply(self.shard, rows, deltas,
+                               self.cfg.update_fn, **kw)
+            return
+        sd = deltas[order]
+        for o in range(max_occ):
+            sel = occ == o
+            _ops.scatter_apply(self.shard, sr[sel], sd[sel],
+                               self.cfg.update_fn, **kw)
+
+    def _ring_push_to(self, r: int, keys: torch.Tensor,
+                      deltas: torch.Tensor) -> None:
+        import time
+
+        n = int(keys.shape[0])
+        cap, vd = self._ring_cap, self.cfg.value_dim
+        assert n <= cap, f"single push of {n} items exceeds ring cap {cap}"
+        # exact host-side backpressure: we are the only writer of our slot
+        deadline = time.monotonic() + 30.0
+        while self._pushed[r] + n - self._head_cache[r] > cap:
+            # refresh the owner's head (remote read), bounded wait
+            self._hip.os_ring_read_head(self._ring_peer[r], self.world_size,
+                                        cap, vd, self.rank, self._scratch)
+            torch.cuda.synchronize()
+            self._head_cache[r] = int(self._scratch[0])
+            if self._pushed[r] + n - self._head_cache[r] <= cap:
+                break
+            if time.monotonic() > deadline:
+                raise RuntimeError(
+                    f"one-sided ring to rank {r} full for 30s "
+                    "(owner not draining?)")
+            time.sleep(0.001)
+        self._hip.os_ring_reserve(self._ring_peer[r], self.world_size, cap,
+                                  vd, self.rank, n, self._scratch[1:])
+        self._hip.os_ring_push(self._ring_peer[r], self.world_size, cap, vd,
+                               self.rank, self._scratch[1:],
+                               keys.contiguous(), deltas)
+        self._pushed[r] += n
+
+    def drain(self, max_per: int = 0) -> int:
+        """OWNER-side: apply queued remote pushes in per-writer order with
+        this table's update function (call between batches; the reference's
+        per-block op queue drains continuously on comm threads — here the
+        batch boundary is the natural quiesce point). Returns items applied."""
+        if not self._ring_mode:
+            return 0
+        cap, vd = self._ring_cap, self.cfg.value_dim
+        max_per = max_per or cap
+        keys, deltas, counts = self._hip.os_ring_drain(
+            self._ring_buf.data_ptr(), self.world_size, cap, vd, max_per)
+        counts = counts.cpu()
+        total = 0
+        for w in range(self.world_size):
+            c = int(counts[w])
+            if not c:
+                continue
+            o = w * max_per
+            self._apply_local(keys[o:o + c], deltas[o:o + c])
+            total += c
+        return total
 
     def pull_full(self) -> torch.Tensor:
         """Async full-table pull (dense apps: MLR/Lasso pull everything)."""
@@ -187,8 +320,7 @@ class OneSidedTable(Table):
     # -------------------------------------------- inherited-API guard rails
 
     def update(self, keys, deltas, assume_unique: bool = False) -> None:
-        # 'add' algebra == async push; other update fns were rejected at
-        # construction, so this is always safe
+        # add-algebra fns -> atomic push; other fns -> ring enqueue (v2)
         self.push(keys, deltas)
 
     def push_dense(self, grad_full) -> None:

@@ -134,7 +134,19 @@ class NMFTrainer(Trainer):
         super().__init__(ctx)
         self.a = defaults(JobConfig(job_id=ctx.job_id, app="nmf",
                                     app_args=ctx.app_args))
-        self.accessor = ETModelAccessor(ctx.table(MODEL_TABLE))
+        table = ctx.table(MODEL_TABLE)
+        from harmony_amd.et.onesided import OneSidedTable
+
+        if isinstance(table, OneSidedTable):
+            # async mode (one_sided=true): pulls are xGMI gather kernels;
+            # pushes go through the owner-side apply-queue RINGS (v2) since
+            # nmf_sgd is not add-algebra — reference per-block op-queue
+            # semantics (CommManager.java:36-155)
+            from harmony_amd.dolphin.model_accessor import OneSidedAccessor
+
+            self.accessor = OneSidedAccessor(table)
+        else:
+            self.accessor = ETModelAccessor(table)
         # Worker-local model table (reference: local table rowKey -> L row,
         # DolphinJobEntity.java:100-110): device-resident, never leaves HBM.
         g = torch.Generator().manual_seed(stable_seed(ctx.job_id, "L", ctx.rank))
@@ -210,8 +222,14 @@ class NMFTrainer(Trainer):
 
     def push_update(self) -> None:
         # uniq_cols are unique and rgrad is already per-key aggregated by K1
-        self.accessor.push(self.batch.uniq_cols, self.rgrad,
-                           assume_unique=True)
+        from harmony_amd.dolphin.model_accessor import OneSidedAccessor
+
+        if isinstance(self.accessor, OneSidedAccessor):
+            self.accessor.push(self.batch.uniq_cols, self.rgrad)
+            self.accessor.drain()      # apply peers' queued pushes (owner)
+        else:
+            self.accessor.push(self.batch.uniq_cols, self.rgrad,
+                               assume_unique=True)
 
     def on_epoch_finished(self, epoch: int) -> None:
         if (epoch + 1) % self.a["decay_period"] == 0:
@@ -229,8 +247,18 @@ class NMFTrainer(Trainer):
 
 def build(job: JobConfig, ctx, cp):
     cfg = model_table_cfg(job, ctx.world_size)
-    comm = ctx.new_data_plane()
-    table = Table(cfg, ctx.rank, ctx.world_size, ctx.device, comm=comm)
+    a = defaults(job)
+    if str(a.get("one_sided", "")).lower() in ("true", "1"):
+        from harmony_amd.et.onesided import OneSidedTable
+
+        table = OneSidedTable(cfg, ctx.rank, ctx.world_size, ctx.device,
+                              store=ctx.store)
+        cp.barrier(f"{job.job_id}/os_alloc", ctx.world_size)
+        table.connect()
+        cp.barrier(f"{job.job_id}/os_conn", ctx.world_size)
+    else:
+        comm = ctx.new_data_plane()
+        table = Table(cfg, ctx.rank, ctx.world_size, ctx.device, comm=comm)
     blocks, rows_local = make_batches(job, ctx.rank, ctx.device,
                                       ctx.world_size)
     tctx = TrainerContext(job_id=job.job_id, rank=ctx.rank,

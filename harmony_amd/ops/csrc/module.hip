@@ -57,6 +57,17 @@ void os_ipc_close(int64_t ptr);
 torch::Tensor os_gather(int64_t ptr, torch::Tensor idx, int64_t k,
                         int64_t dtype_i32);
 void os_scatter_add(int64_t ptr, torch::Tensor idx, torch::Tensor delta);
+int64_t os_ring_bytes(int64_t W, int64_t cap, int64_t vd);
+void os_ring_reserve(int64_t base, int64_t W, int64_t cap, int64_t vd,
+                     int64_t writer, int64_t n, torch::Tensor scratch);
+void os_ring_read_head(int64_t base, int64_t W, int64_t cap, int64_t vd,
+                       int64_t writer, torch::Tensor scratch);
+void os_ring_push(int64_t base, int64_t W, int64_t cap, int64_t vd,
+                  int64_t writer, torch::Tensor seq_base,
+                  torch::Tensor keys, torch::Tensor deltas);
+std::vector<torch::Tensor> os_ring_drain(int64_t base, int64_t W,
+                                         int64_t cap, int64_t vd,
+                                         int64_t max_per);
 void gbt_hist(torch::Tensor bins, torch::Tensor resid, torch::Tensor node,
               torch::Tensor cnt, torch::Tensor sum);
 void scatter_apply(torch::Tensor shard, torch::Tensor rows,
@@ -102,6 +113,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("os_gather", &os_gather, "gather rows from a mapped shard (K12)");
   m.def("os_scatter_add", &os_scatter_add,
         "atomic scatter-add into a mapped shard (K12)");
+  m.def("os_ring_bytes", &os_ring_bytes, "ring buffer size (K12b)");
+  m.def("os_ring_reserve", &os_ring_reserve, "reserve seq range in a peer ring");
+  m.def("os_ring_read_head", &os_ring_read_head, "read peer ring head");
+  m.def("os_ring_push", &os_ring_push, "enqueue items into a peer ring");
+  m.def("os_ring_drain", &os_ring_drain, "owner-side in-order ring drain");
   m.def("scatter_apply", &scatter_apply, "owner-side sparse update (K9)");
   m.def("dense_apply", &dense_apply, "owner-side dense update (K3)");
   m.def("parse_nmf_bytes", &parse_nmf_bytes, "native NMF text parser");

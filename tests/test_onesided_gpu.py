@@ -251,3 +251,136 @@ def test_async_lda_counts_conserved():
     for wt, summ, want in res:
         assert wt == want, res
         assert summ == want, res
+
+
+def _ring_worker(rank, world):
+    """One-sided v2 apply-queue rings: exact totals under concurrent
+    remote pushes, single-writer FIFO order through clamp nonlinearity,
+    forced wraparound (cap=32 << items pushed) and backpressure."""
+    import time
+
+    from harmony_amd.config import TableConfig
+    from harmony_amd.et.onesided import OneSidedTable
+
+    torch.cuda.set_device(0)
+    store = _store(rank, world)
+    step, maxv = 0.1, 1e6
+    cfg = TableConfig(table_id="os_ring", num_keys=32, value_dim=8,
+                      num_blocks=8, update_fn="nmf_sgd",
+                      update_args={"step_size": step, "max_val": maxv},
+                      init_fn="zeros")
+    t = OneSidedTable(cfg, rank, world, torch.device("cuda"), store=store,
+                      ring_capacity=32)
+    assert t._ring_mode
+    _barrier(store, "alloc", rank, world)
+    t.connect()
+    _barrier(store, "conn", rank, world)
+    other = 1 - rank
+    # keys owned by the OTHER rank (pure ring path)
+    all_keys = torch.arange(32, device="cuda")
+    owners = t._owner_of(t.part.block_of(all_keys))
+    rkeys = all_keys[owners == other]
+    n = rkeys.numel()
+    assert n == 16
+    # phase 1: uneven rounds of remote pushes; drain own ring as we go.
+    # delta -d => value += step*d per push (linear region, no clamp)
+    rounds = 40 if rank == 0 else 17
+    d = torch.full((n, 8), -(0.01 * (rank + 1)), device="cuda")
+    for _ in range(rounds):
+        t.push(rkeys, d)
+        t.drain()
+    store.add("pushed_done", 1)
+    while int(store.add("pushed_done", 0)) < world:
+        t.drain()
+        time.sleep(0.002)
+    # drain until empty twice in a row (peer's last items may be in flight)
+    empties = 0
+    while empties < 3:
+        empties = empties + 1 if t.drain() == 0 else 0
+        time.sleep(0.002)
+    t.fence()
+    _barrier(store, "ph1", rank, world)
+    # my rows got `other_rounds` pushes of delta -(0.01*(other+1))
+    other_rounds = 40 if other == 0 else 17
+    mine = all_keys[owners == rank]
+    got = t.pull(mine)
+    exp = torch.full_like(got, step * 0.01 * (other + 1) * other_rounds)
+    assert torch.allclose(got, exp, atol=1e-4), (rank, got[0, 0].item(),
+                                                 exp[0, 0].item())
+    _barrier(store, "ph1ok", rank, world)
+
+    # phase 2: single-writer FIFO order through the clamp: rank 1 pushes
+    # an order-sensitive sequence to ONE rank-0 key; rank 0 drains then
+    # both check against the in-order oracle.
+    key0 = all_keys[owners == 0][:1]
+    seq = [-30.0, +25.0, -1.0]   # clamp at 0 hits iff applied in order
+    if rank == 1:
+        for v in seq:
+            t.push(key0, torch.full((1, 8), v, device="cuda"))
+        t.fence()
+    store.add("ph2_pushed", 1)
+    if rank == 0:
+        while int(store.add("ph2_pushed", 0)) < 1:
+            time.sleep(0.002)
+        while t.drain() == 0:
+            time.sleep(0.002)
+        empties = 0
+        while empties < 3:
+            empties = empties + 1 if t.drain() == 0 else 0
+            time.sleep(0.002)
+        v = float(t.pull(key0)[0, 0])
+        # oracle: v0 = phase-1 value; then clamp(v - 0.1*d) in order
+        v0 = step * 0.01 * 2 * 17   # rank1 pushed 17 rounds of -0.02
+        for dd in seq:
+            v0 = min(max(v0 - step * dd, 0.0), maxv)
+        assert abs(v - v0) < 1e-4, (v, v0)
+    _barrier(store, "ph2ok", rank, world)
+
+    # phase 3: backpressure — rank 1 pushes 4x capacity while rank 0
+    # delays draining; the writer must block (not corrupt) and finish.
+    if rank == 1:
+        for i in range(8):
+            t.push(key0.repeat(16), torch.full((16, 8), -0.5, device="cuda"))
+        t.fence()
+    else:
+        time.sleep(0.4)
+        applied = 0
+        while applied < 8 * 16:
+            applied += t.drain()
+            time.sleep(0.01)
+    _barrier(store, "ph3", rank, world)
+    if rank == 0:
+        v = float(t.pull(key0)[0, 0])
+        # 128 pushes of -0.5 after phase 2: linear adds of +0.05 each
+        assert v > 0.05 * 127, v
+    _barrier(store, "done", rank, world)
+    t.close()
+    return True
+
+
+def test_onesided_ring_apply_queue():
+    assert all(run_dist(_ring_worker, world=2, timeout=300))
+
+
+def _async_nmf_worker(rank, world):
+    """Full run_job NMF in one-sided mode: nmf_sgd pushes go through the
+    v2 apply-queue rings (non-add update fn), training converges."""
+    from harmony_amd.config import JobConfig, RuntimeConfig
+    from harmony_amd.dolphin.master import run_job
+    from harmony_amd.runtime.bootstrap import init_executor
+
+    ctx = init_executor(RuntimeConfig(device="cuda", backend="gloo"))
+    job = JobConfig(job_id="os_nmf", app="nmf", max_num_epochs=4,
+                    num_mini_batches=2, clock_slack=2,
+                    app_args={"num_cols": 256, "rank": 16, "nnz_per_row": 8,
+                              "rows_per_batch": 256, "step_size": 0.05,
+                              "one_sided": True})
+    s = run_job(job, ctx).summary()
+    return (s["num_batches"], s.get("sq_err", None))
+
+
+def test_async_nmf_job_one_sided_rings():
+    res = run_dist(_async_nmf_worker, world=2, timeout=300)
+    for n, sq in res:
+        assert n == 8
+        assert sq is not None
