@@ -123,6 +123,8 @@ class Table:
         after multiUpdate merging (TableImpl.java:460, BlockImpl.update:71).
         """
         rows = self.local_rows_of(keys)
+        # incremental touched-row tracking (Pregel has_msg; reset by callers)
+        self.last_touched_keys = keys
         if self.device.type == "cuda" and ops.fused_apply_supported(self.cfg.update_fn):
             # fused gather-apply-scatter kernel (K3/K9)
             ops.scatter_apply(self.shard, rows, deltas.to(self.dtype),
@@ -191,18 +193,23 @@ class Table:
             self.shard[rows[sel]] = fresh[offs]
 
     def update(self, keys: torch.Tensor, deltas: torch.Tensor,
-               assume_unique: bool = False) -> None:
+               assume_unique: bool = False, piggyback=None):
         """multiUpdate (reference TableImpl.java:460): route deltas to owner
         ranks; the owner applies the table's update function.
         assume_unique: keys are already unique + per-key aggregated (skips a
-        device sort/unique on the hot path)."""
+        device sort/unique on the hot path). piggyback: small int vector
+        summed globally inside the push's count exchange (see
+        DataPlane.push_keys); returned when given."""
         if self._local_only():
-            if not assume_unique:
-                keys, deltas = uf.merge_key_deltas(keys, deltas,
-                                                   self.cfg.update_fn)
-            self.update_local(keys, deltas)
-            return
-        self.comm.push_keys(self, keys, deltas, assume_unique=assume_unique)
+            if keys.numel():
+                if not assume_unique:
+                    keys, deltas = uf.merge_key_deltas(keys, deltas,
+                                                       self.cfg.update_fn)
+                self.update_local(keys, deltas)
+            return piggyback
+        return self.comm.push_keys(self, keys, deltas,
+                                   assume_unique=assume_unique,
+                                   piggyback=piggyback)
 
     def pull_all(self) -> torch.Tensor:
         """Gather the whole table (dense apps pull every partition each batch,
@@ -270,6 +277,10 @@ class ObjectTable:
         self.update_value = update_value
         self.blocks: Dict[int, Dict[int, Any]] = {
             b: {} for b in self.ownership.owned_blocks(rank)}
+        # bumped by out-of-band content changes (checkpoint reload,
+        # migration, puts/removes) so replicated caches built from the
+        # append-only push stream know to rebuild (gbt.py forest replica)
+        self.content_epoch = 0
 
     def get_or_init_local(self, key: int) -> Any:
         b = self.part.block_of_int(key)
@@ -285,9 +296,11 @@ class ObjectTable:
         return blk[key]
 
     def put_local(self, key: int, value: Any) -> None:
+        self.content_epoch += 1
         self.blocks[self.part.block_of_int(key)][key] = value
 
     def remove_local(self, key: int) -> Any:
+        self.content_epoch += 1
         return self.blocks[self.part.block_of_int(key)].pop(key, None)
 
     def get(self, key: int) -> Any:
@@ -317,9 +330,11 @@ class ObjectTable:
 
     # migration support
     def drop_blocks(self, block_ids: List[int]) -> Dict[int, Dict[int, Any]]:
+        self.content_epoch += 1
         return {b: self.blocks.pop(b) for b in block_ids}
 
     def adopt_blocks(self, blocks: Dict[int, Dict[int, Any]]) -> None:
+        self.content_epoch += 1
         self.blocks.update(blocks)
 
 

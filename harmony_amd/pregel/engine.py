@@ -132,32 +132,42 @@ class PregelEngine:
         has_msg = torch.zeros(n_local, dtype=torch.bool, device=dev)
         active = torch.ones(n_local, dtype=torch.bool, device=dev)
         jid = self.job.job_id
+        zero_keys = torch.empty(0, dtype=torch.int64, device=dev)
+        zero_msgs = torch.empty((0, comp.msg_dim), dtype=torch.float32,
+                                device=dev)
+        lo, _hi = self.local_vertex_range()
         for step in range(self.max_supersteps):
             self.supersteps_run = step + 1
             # COMP: local vertex update over combined incoming messages
             incoming = self.msg[self.cur].shard[:n_local]
-            run_mask = active | has_msg
             values, edge_msgs, active = comp.compute(
                 step, values, incoming, has_msg, g)
             # clear the consumed buffer for reuse (flip semantics,
             # reference MessageManager.flip:72-74)
             self._clear(self.msg[self.cur], ident)
-            # SEND: combining scatter into the other message table
+            # SEND + SYNC fused: ONE collective per superstep. Every rank
+            # enters the push even with zero messages (a per-rank skip
+            # would strand peers inside the collective); the halt vote
+            # (allVerticesHalt AND noOngoingMsgs, PregelMaster.java:48-56)
+            # rides the push's count exchange as a piggyback sum, and the
+            # data legs are skipped symmetrically when no rank sent.
             nxt = self.msg[1 - self.cur]
-            sent = 0
-            if edge_msgs is not None and g.edge_dst.numel() > 0:
-                with self.tus.net(jid, self._next_phase()):
-                    nxt.update(g.edge_dst, edge_msgs)
-                sent = int(g.edge_dst.numel())
-            # SYNC: halt vote (allVerticesHalt AND noOngoingMsgs)
-            votes = torch.tensor([float(active.sum()), float(sent)],
-                                 device=dev)
-            if dist.is_initialized():
-                with self.tus.net(jid, self._next_phase()):
-                    dist.all_reduce(votes, group=getattr(self.ctx, "group", None))
+            keys = (g.edge_dst if edge_msgs is not None else zero_keys)
+            msgs = (edge_msgs if edge_msgs is not None else zero_msgs)
+            sent = int(keys.numel())
+            nxt.last_touched_keys = None
+            vote = torch.tensor([int(active.sum()), sent])
+            with self.tus.net(jid, self._next_phase()):
+                sums = nxt.update(keys, msgs, piggyback=vote)
             self.cur = 1 - self.cur
-            has_msg = (self.msg[self.cur].shard[:n_local] != ident).any(dim=1)
-            if float(votes[0]) == 0 and float(votes[1]) == 0:
+            # incremental has-message mask from the owner-apply's touched
+            # keys (replaces the full-shard != identity scan)
+            has_msg = torch.zeros(n_local, dtype=torch.bool, device=dev)
+            touched = getattr(self.msg[self.cur], "last_touched_keys", None)
+            if touched is not None and touched.numel():
+                rows = self.msg[self.cur].local_rows_of(touched.to(dev))
+                has_msg[rows[rows < n_local]] = True
+            if sums is not None and int(sums[0]) == 0 and int(sums[1]) == 0:
                 break
         self.vertex_table.shard[:n_local] = values
         return values

@@ -30,7 +30,9 @@ def test_gbt_fits_nonlinear_signal():
     m = run_job(job, ctx)
     s = m.summary()
     assert s["num_batches"] == 16
-    assert s["num_trees"] == 15          # forest pulled before last push
+    # replicated forest includes the last push (r02 tensorized sync
+    # appends at push time; the r01 pull-lag reported 15)
+    assert s["num_trees"] == 16
     # boosting reduced mse well below the signal variance (~9)
     assert s["mse"] < 1.5, s
 
@@ -67,7 +69,9 @@ def test_gbt_two_ranks_forest_shared():
     res = run_dist(_gbt_2rank_worker, world=2, timeout=120)
     for nb, nt in res:
         assert nb == 4
-        assert nt == 6   # before the 4th batch's push: 3 batches * 2 ranks
+        # r02 replicated forest includes the final collective push:
+        # 4 batches * 2 ranks (r01's pull-lag view saw 6)
+        assert nt == 8
 
 
 def test_gbt_incremental_pred_matches_full():
@@ -135,4 +139,5 @@ def test_gbt_multiclass_end_to_end():
     s = run_job(job, ctx).summary()
     assert s["num_batches"] == 12
     assert s["error_rate"] < 0.25, s       # chance = 0.67 for 3 classes
-    assert s["num_trees"] == 3 * 11        # C forests, one tree/batch each
+    # C forests, one tree per batch each; replica includes the last push
+    assert s["num_trees"] == 3 * 12
