@@ -62,8 +62,10 @@ class JobView:
 
 class JobServerDriver:
     def __init__(self, ctx: ExecutorContext, scheduler: str = "default",
-                 port: int = DEFAULT_JOBSERVER_PORT):
+                 port: int = DEFAULT_JOBSERVER_PORT,
+                 hb_period: float = 2.0):
         self.ctx = ctx
+        self.hb_period = hb_period
         self.cp = ControlPlane(ctx.store, ctx.rank, ctx.world_size)
         self.tus = TaskUnitScheduler(self.cp, {ADMIN_JOB}, multi_job=True)
         self.tus.set_drawer(ADMIN_JOB, ctx.rank == 0)
@@ -249,7 +251,7 @@ class JobServerDriver:
         in the reference (JobServerDriver failed-evaluator handlers throw —
         TODO #677 'no recovery'): a dead executor marks the server failed
         and shuts it down; restart + checkpoint-restore is the story."""
-        period = 2.0
+        period = self.hb_period
         while not self.cp.flag_set("js/shutdown"):
             self.cp.store.set(f"js/hb/{self.ctx.rank}", str(time.time()))
             if self.ctx.is_master:
@@ -275,7 +277,8 @@ class JobServerDriver:
         hb = threading.Thread(target=self._heartbeat_loop, daemon=True)
         hb.start()
         self.dispatch_loop()
-        if dist.is_initialized():
+        # on fail-fast the dead executor can never reach a barrier
+        if dist.is_initialized() and not self.cp.flag_set("js/failed"):
             dist.barrier()
         # executor close: commit temp checkpoints (reference ChkpManagerSlave
         # commitAllLocalChkps on close, ChkpManagerSlave.java:226). After the
