@@ -31,8 +31,12 @@ class TrainingDataProvider:
         self._sliced: dict = {}
 
     def set_share(self, frac: float) -> None:
-        """frac in (0, 1]: serve that fraction of each block's examples."""
-        frac = min(1.0, max(frac, 0.01))
+        """frac in [0, 1]: serve that fraction of each block's examples.
+        frac == 0 (StopWorkerOp) serves truly EMPTY batches — the stopped
+        worker keeps entering collectives (it still serves its table
+        blocks, the reference's worker->server switch) but contributes
+        zero keys, shedding its pull/push fan-in."""
+        frac = min(1.0, max(frac, 0.0))
         if frac != self.share:
             self.share = frac
             self._sliced.clear()

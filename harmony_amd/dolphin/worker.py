@@ -77,12 +77,17 @@ class WorkerTasklet:
         rank serves a prefix slice of each block instead)."""
         shares = getattr(self.orch.executor, "batch_shares", None)
         if not shares or self.rank not in shares:
+            return   # ranks absent from the plan keep their current share
+        if not hasattr(self.provider, "set_share"):
+            return
+        if shares[self.rank] <= 0:
+            # StopWorkerOp: truly EMPTY batches — the rank keeps serving
+            # its table blocks (pure server) but sheds all pull/push keys
+            self.provider.set_share(0.0)
             return
         vals = [v for v in shares.values() if v > 0]
-        if not vals:
-            return
         mean = sum(vals) / len(vals)
-        if mean > 0 and hasattr(self.provider, "set_share"):
+        if mean > 0:
             self.provider.set_share(shares[self.rank] / mean)
 
     def run(self) -> MetricCollector:

@@ -49,6 +49,15 @@ class DataPlane:
         # it removes one all-gather + host sync per phase
         self._counts_cache: Dict[tuple, tuple] = {}
 
+    def invalidate_routes(self) -> None:
+        """Drop route + recv-count caches. MUST be called collectively
+        whenever any rank's key sets change out-of-band (batch reslicing
+        by an elasticity plan): a per-rank cache hit/miss divergence would
+        put ranks in different stages of the same push (observed hang),
+        and cached recv counts would be stale even when consistent."""
+        self._route_cache.clear()
+        self._counts_cache.clear()
+
     # ------------------------------------------------------------------ utils
 
     def _supports_a2a(self) -> bool:

@@ -207,7 +207,23 @@ class GBTTrainer(Trainer):
         else:
             self.forest = allv.setdefault(0, [])
 
+    def _trivial_tree(self):
+        # a stopped worker still contributes to the equal-count tree
+        # all-gather: a pass-through tree predicting 0 (harmless append)
+        d = self.a["max_depth"]
+        return GBTree(depth=d, feature=[0] * ((1 << d) - 1),
+                      threshold=[-1] * ((1 << d) - 1),
+                      leaf_value=[0.0] * (1 << d))
+
     def local_compute(self) -> None:
+        if self.batch[0].shape[0] == 0:
+            if self.a["objective"] == "multiclass":
+                self.new_trees = [(c, self._trivial_tree())
+                                  for c in range(self.a["num_classes"])]
+            else:
+                self.new_tree = self._trivial_tree()
+            self._mse = 0.0
+            return
         if self.a["objective"] == "multiclass":
             self._compute_multiclass()
             return
@@ -353,7 +369,9 @@ def build(job: JobConfig, ctx, cp):
                           tables={MODEL_TABLE: table}, app_args=job.app_args)
     trainer = GBTTrainer(tctx)
     def _reslice(b, frac):
-        n = max(1, int(b[0].shape[0] * frac))
+        # frac<=0 -> EMPTY batch: a stopped worker (StopWorkerOp) does
+        # zero work and its sparse pulls/pushes carry zero keys
+        n = 0 if frac <= 0 else max(1, int(b[0].shape[0] * frac))
         return (b[0][:n], b[1][:n])
 
     provider = TrainingDataProvider(reslice=_reslice, local_blocks=

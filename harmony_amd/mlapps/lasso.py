@@ -101,6 +101,10 @@ class LassoTrainer(Trainer):
 
     def local_compute(self) -> None:
         X, y = self.batch
+        if X.shape[0] == 0:      # stopped worker: no work, zero delta
+            self.delta = torch.zeros_like(self.w)
+            self._loss = torch.zeros(())
+            return
         F = X.shape[1]
         lam_n = self.a["lam"] * X.shape[0]
         r = y - X @ self.w                     # residual
@@ -146,7 +150,9 @@ def build(job: JobConfig, ctx, cp):
                           tables={MODEL_TABLE: table}, app_args=job.app_args)
     trainer = LassoTrainer(tctx)
     def _reslice(b, frac):
-        n = max(1, int(b[0].shape[0] * frac))
+        # frac<=0 -> EMPTY batch: a stopped worker (StopWorkerOp) does
+        # zero work and its sparse pulls/pushes carry zero keys
+        n = 0 if frac <= 0 else max(1, int(b[0].shape[0] * frac))
         return (b[0][:n], b[1][:n])
 
     provider = TrainingDataProvider(reslice=_reslice, local_blocks=blocks)

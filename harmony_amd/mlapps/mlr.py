@@ -148,6 +148,9 @@ class MLRTrainer(Trainer):
 
     def local_compute(self) -> None:
         x, y = self.batch
+        if x.shape[0] == 0:       # stopped worker: zero grad, no forward
+            self.grad_raw = torch.zeros_like(self._w_matrix())
+            return
         key = id(self.batch)
         body = self._bodies.get(key)
         if body is None:
@@ -233,7 +236,9 @@ def build(job: JobConfig, ctx, cp):
                           tables={MODEL_TABLE: table}, app_args=job.app_args)
     trainer = MLRTrainer(tctx)
     def _reslice(b, frac):
-        n = max(1, int(b[0].shape[0] * frac))
+        # frac<=0 -> EMPTY batch: a stopped worker (StopWorkerOp) does
+        # zero work and its sparse pulls/pushes carry zero keys
+        n = 0 if frac <= 0 else max(1, int(b[0].shape[0] * frac))
         return (b[0][:n], b[1][:n])
 
     provider = TrainingDataProvider(reslice=_reslice, local_blocks=

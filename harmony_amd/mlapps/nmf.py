@@ -200,6 +200,10 @@ class NMFTrainer(Trainer):
 
     def local_compute(self) -> None:
         b = self.batch
+        if b.num_examples == 0:   # stopped worker: empty push, no kernel
+            self.rgrad = torch.zeros((0, self.a["rank"]),
+                                     device=self.ctx.device)
+            return
         key = id(b)
         body = self._bodies.get(key)
         if body is None:
@@ -276,7 +280,7 @@ def build(job: JobConfig, ctx, cp):
     def _reslice(b, frac):
         # row-prefix re-slice for SetBatchShareOp: rebuild the static
         # precomputes (uniq_cols, col_sorted) for the smaller batch once
-        n = max(1, int(b.l_rows.shape[0] * frac))
+        n = 0 if frac <= 0 else max(1, int(b.l_rows.shape[0] * frac))
         nnz = int(b.row_ptr[n])
         nb = NMFBatch(b.l_rows[:n], b.row_ptr[:n + 1].contiguous(),
                       b.col_idx[:nnz], b.vals[:nnz])

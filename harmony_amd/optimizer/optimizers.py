@@ -170,6 +170,29 @@ class SampleOptimizers:
         return _Sh()
 
     @staticmethod
+    def stop_then_start(rank: int, num_batches: int = 1,
+                        stop_at_call: int = 1, start_at_call: int = 3):
+        """Scripted worker role switch (reference DeleteOneWorker +
+        AddOneWorker): STOP `rank` at optimization round `stop_at_call`
+        (its batches become EMPTY — zero examples, zero sparse wire),
+        re-START it with `num_batches` share at `start_at_call`."""
+        from harmony_amd.optimizer.plan import compile_switch
+
+        class _SS(Optimizer):
+            def __init__(self):
+                self.calls = 0
+
+            def optimize(self, metrics, owners, world_size):
+                self.calls += 1
+                if self.calls == stop_at_call:
+                    return compile_switch("", [rank], {}, {})
+                if self.calls == start_at_call:
+                    return compile_switch("", [], {rank: num_batches}, {})
+                return Plan()
+
+        return _SS()
+
+    @staticmethod
     def even_rebalance(table_id: str):
         class _Even(Optimizer):
             def optimize(self, metrics, owners, world_size):

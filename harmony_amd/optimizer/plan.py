@@ -191,3 +191,11 @@ class PlanExecutor:
                 if table is not None and hasattr(table, "drop_blocks"):
                     table.drop_blocks(list(table.owned_blocks))
             ready = sorted(ready + dag.on_complete(i))
+        # plans change key routing (migration moved blocks; share changes
+        # resliced batches) — drop route/count caches on EVERY rank at the
+        # same quiesced point, or cache hit/miss diverges across ranks and
+        # desynchronizes the next push's collectives
+        for t in self.tables.values():
+            c = getattr(t, "comm", None)
+            if c is not None and hasattr(c, "invalidate_routes"):
+                c.invalidate_routes()

@@ -21,6 +21,12 @@ def _entry(fn, rank, world, port, q, args):
     # tee stderr to a per-rank file so run_dist can show it on failure
     err_path = f"/tmp/dist_helper_r{rank}_p{port}.err"
     sys.stderr = open(err_path, "w", buffering=1)
+    if os.environ.get("DIST_HELPER_DUMP"):
+        import faulthandler
+
+        faulthandler.dump_traceback_later(
+            int(os.environ["DIST_HELPER_DUMP"]), exit=False,
+            file=sys.stderr, repeat=False)
     try:
         os.environ["RANK"] = str(rank)
         os.environ["LOCAL_RANK"] = str(rank)

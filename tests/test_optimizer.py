@@ -228,3 +228,33 @@ def test_compile_switch_dag_order():
     # degenerate: only starts, no deps needed
     p3 = compile_switch("t", stop_ranks=[], start={2: 4}, moves={})
     assert len(p3.ops) == 1 and p3.deps == []
+
+
+def _stop_start_worker(rank, world):
+    """True worker stop: after StopWorkerOp the rank's batches are EMPTY
+    (zero examples -> zero sparse fan-in), StartWorkerOp restores work
+    (VERDICT r01 item 4 / missing #3)."""
+    from harmony_amd.config import JobConfig, RuntimeConfig
+    from harmony_amd.dolphin.master import run_job
+    from harmony_amd.optimizer.optimizers import SampleOptimizers
+    from harmony_amd.runtime.bootstrap import init_executor
+
+    ctx = init_executor(RuntimeConfig(device="cpu"))
+    job = JobConfig(job_id="ss1", app="nmf", max_num_epochs=8,
+                    num_mini_batches=2, optimizer_period=2,
+                    app_args={"num_cols": 128, "rank": 8, "nnz_per_row": 4,
+                              "rows_per_batch": 64})
+    opt = SampleOptimizers.stop_then_start(rank=1, num_batches=1,
+                                           stop_at_call=1, start_at_call=3)
+    m = run_job(job, ctx, optimizer=opt)
+    return [b.num_examples for b in m.batches]
+
+
+def test_stop_start_worker_sheds_work():
+    res = run_dist(_stop_start_worker, world=2, timeout=180)
+    ex0, ex1 = res
+    assert all(n == 64 for n in ex0), ex0          # rank 0 never stopped
+    assert 0 in ex1, ex1                           # rank 1 truly stopped
+    # stopped in the middle, working at both ends
+    assert ex1[0] == 64 and ex1[-1] > 0, ex1
+    assert len(ex0) == len(ex1)                    # same batch COUNT (collective)
