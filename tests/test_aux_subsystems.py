@@ -415,6 +415,7 @@ def test_heartbeat_failfast_logic():
 
     drv = JobServerDriver.__new__(JobServerDriver)
     drv.ctx = _Ctx()
+    drv.hb_period = 0.2
     drv.cp = ControlPlane(_Ctx.store, 0, 2)
     # rank 1's heartbeat is ancient -> the loop must fail fast and return
     _Ctx.store.set("js/hb/1", str(time.time() - 120))
