@@ -92,6 +92,64 @@ class CachedModelAccessor(ETModelAccessor):
                                    **self.table.cfg.update_args)
 
 
+class CachedOneSidedAccessor:
+    """Cached accessor with a BACKGROUND refresh thread (reference
+    CachedModelAccessor.java:40-130: a Guava cache re-pulled every
+    MODEL_REFRESH_SEC by a refresh thread). Background pulls are only safe
+    on the ASYNC plane — a one-sided table's pull is a gather kernel, so a
+    timer thread cannot desynchronize any collective order. Pushes go
+    through immediately and are applied write-through to the cache."""
+
+    def __init__(self, table, refresh_sec: float = 0.5):
+        import threading
+
+        self.table = table
+        self.refresh_sec = refresh_sec
+        self._cache: Optional[torch.Tensor] = None
+        self._lock = threading.Lock()
+        self._stop = threading.Event()
+        self.refreshes = 0
+        self._thread = threading.Thread(target=self._refresh_loop,
+                                        daemon=True)
+        self._thread.start()
+
+    def _refresh_loop(self) -> None:
+        while not self._stop.wait(self.refresh_sec):
+            fresh = self.table.pull_full()
+            with self._lock:
+                self._cache = fresh
+                self.refreshes += 1
+
+    def close(self) -> None:
+        self._stop.set()
+        self._thread.join(timeout=5)
+
+    def pull_all(self) -> torch.Tensor:
+        with self._lock:
+            if self._cache is None:
+                self._cache = self.table.pull_full()
+            return self._cache
+
+    def pull(self, keys: torch.Tensor) -> torch.Tensor:
+        return self.pull_all()[keys]
+
+    def push(self, keys, deltas, assume_unique: bool = False) -> None:
+        self.table.push(keys, deltas)
+        from harmony_amd.et import update_functions as uf
+
+        with self._lock:
+            if self._cache is not None:
+                fn = uf.update_fn(self.table.cfg.update_fn)
+                rows = self._cache[keys]
+                self._cache[keys] = fn(rows, deltas.to(rows.dtype),
+                                       **self.table.cfg.update_args)
+
+    def drain(self):
+        if hasattr(self.table, "drain"):
+            return self.table.drain()
+        return 0
+
+
 class OneSidedAccessor:
     """ETModelAccessor-shaped facade over an et.onesided.OneSidedTable:
     pulls and pushes are direct xGMI kernels, never collectives, so a

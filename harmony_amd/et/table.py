@@ -75,6 +75,7 @@ class Table:
         else:
             self.shard = torch.empty((0, self.cfg.value_dim), dtype=self.dtype,
                                      device=self.device)
+        self._buf = self.shard     # capacity buffer (see _rebuild)
         slot = torch.full((self.cfg.num_blocks,), -1, dtype=torch.int64)
         for i, b in enumerate(owned):
             slot[b] = i
@@ -239,22 +240,61 @@ class Table:
         """Insert received blocks into the local shard."""
         self._rebuild(sorted(set(self._local_blocks) | set(blocks)), carry=blocks)
 
-    def _rebuild(self, new_block_list: List[int], carry: Dict[int, torch.Tensor]) -> None:
-        rows = []
-        for b in sorted(new_block_list):
-            if b in carry:
-                rows.append(carry[b].to(self.device, self.dtype))
-            else:
-                rows.append(self.local_block_view(b))
-        new_shard = (torch.cat(rows, dim=0) if rows else
-                     torch.empty((0, self.cfg.value_dim), dtype=self.dtype,
-                                 device=self.device))
+    def _rebuild(self, new_block_list: List[int],
+                 carry: Dict[int, torch.Tensor]) -> None:
+        """Re-lay the shard for a new block list. IN PLACE whenever the
+        capacity buffer fits (transient memory = ONE block, not a full
+        shard copy — round 1 concatenated a second full shard, 2x peak at
+        100+ GB scale); a capacity miss reallocates once with headroom so
+        steady-state migration churn never reallocates again."""
+        bs, vd = self.block_size, self.cfg.value_dim
+        final = sorted(new_block_list)
+        need = len(final) * bs
+        buf = getattr(self, "_buf", None)
+        if buf is None or buf.shape[0] < need or buf.device != self.device:
+            # grow (or first migration): allocate with 25% block headroom
+            cap = (len(final) + max(1, len(final) // 4)) * bs
+            newbuf = torch.empty((cap, vd), dtype=self.dtype,
+                                 device=self.device)
+            for i, b in enumerate(final):
+                src = (carry[b].to(self.device, self.dtype) if b in carry
+                       else self.local_block_view(b))
+                newbuf[i * bs:(i + 1) * bs] = src
+            self._buf = newbuf
+        else:
+            # in-place: move kept blocks to their new slots. Downward moves
+            # (dst < src) in increasing dst order, upward in decreasing dst
+            # order — never clobbers an unmoved block; a one-block temp
+            # covers the overlapping-src/dst case.
+            moves = []
+            for i, b in enumerate(final):
+                if b in carry:
+                    continue
+                src = int(self._block_slot[b])
+                if src != i:
+                    moves.append((src, i))
+            for src, dst in sorted([m for m in moves if m[1] < m[0]],
+                                   key=lambda m: m[1]):
+                self._block_move(buf, src, dst)
+            for src, dst in sorted([m for m in moves if m[1] > m[0]],
+                                   key=lambda m: -m[1]):
+                self._block_move(buf, src, dst)
+            for i, b in enumerate(final):
+                if b in carry:
+                    buf[i * bs:(i + 1) * bs] = carry[b].to(self.device,
+                                                           self.dtype)
+        self.shard = self._buf.narrow(0, 0, need)
         slot = torch.full((self.cfg.num_blocks,), -1, dtype=torch.int64)
-        for i, b in enumerate(sorted(new_block_list)):
+        for i, b in enumerate(final):
             slot[b] = i
-        self.shard = new_shard
         self._block_slot = slot.to(self.device)
-        self._local_blocks = sorted(new_block_list)
+        self._local_blocks = final
+
+    def _block_move(self, buf: torch.Tensor, src: int, dst: int) -> None:
+        bs = self.block_size
+        s = buf[src * bs:(src + 1) * bs]
+        d = buf[dst * bs:(dst + 1) * bs]
+        d.copy_(s)          # block-granular slots never overlap
 
 
 class ObjectTable:

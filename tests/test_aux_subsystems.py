@@ -505,3 +505,45 @@ def test_agree_max_multiproc():
 
     outs = run_dist(_agree_max_worker, world=3)
     assert all(o == (12, 100) for o in outs)
+
+
+def test_cached_one_sided_accessor_background_refresh():
+    """Background timer refresh (reference CachedModelAccessor refresh
+    thread) against a fake async table — CPU-testable thread logic."""
+    import time as _time
+
+    from harmony_amd.dolphin.model_accessor import CachedOneSidedAccessor
+
+    class FakeCfg:
+        update_fn = "add"
+        update_args = {}
+
+    class FakeTable:
+        cfg = FakeCfg()
+
+        def __init__(self):
+            self.version = torch.zeros(4, 2)
+
+        def pull_full(self):
+            return self.version.clone()
+
+        def push(self, keys, deltas):
+            self.version[keys] += deltas
+
+    t = FakeTable()
+    acc = CachedOneSidedAccessor(t, refresh_sec=0.05)
+    try:
+        v0 = acc.pull_all()
+        assert float(v0.sum()) == 0.0
+        # a REMOTE writer changes the table; the refresh thread picks it up
+        t.version += 1.0
+        deadline = _time.monotonic() + 5
+        while float(acc.pull_all().sum()) == 0.0:
+            assert _time.monotonic() < deadline, "refresh thread never ran"
+            _time.sleep(0.02)
+        assert acc.refreshes >= 1
+        # write-through push visible immediately (no refresh wait)
+        acc.push(torch.tensor([0]), torch.ones(1, 2) * 5)
+        assert float(acc.pull_all()[0, 0]) >= 6.0
+    finally:
+        acc.close()
