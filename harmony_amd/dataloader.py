@@ -145,10 +145,16 @@ def _read_split_bytes(path: str, split: Tuple[int, int]) -> bytes:
         return f.read(max(0, end - start))
 
 
-def _native():
+def _native(world_size: int = 1):
     from harmony_amd import ops
 
-    return ops._load_hip()
+    nat = ops._load_hip()
+    if nat is not None and "HARMONY_PARSE_THREADS" not in os.environ:
+        # one loader process per GPU: cap scan threads to a fair core share
+        # (oversubscription measured 5x slower at 8 ranks — ingest_bench.py)
+        os.environ["HARMONY_PARSE_THREADS"] = str(
+            max(1, (os.cpu_count() or 1) // max(1, world_size)))
+    return nat
 
 
 def parse_nmf_split(path: str, rank: int, world_size: int):
@@ -156,30 +162,30 @@ def parse_nmf_split(path: str, rank: int, world_size: int):
     the extension is built — multi-threaded byte scan, measured 5-10x the
     Python line parser — else Python)."""
     split = compute_splits(path, world_size)[rank]
-    nat = _native()
+    nat = _native(world_size)
     if nat is not None:
         buf = _read_split_bytes(path, split)
-        return tuple(nat.parse_nmf_bytes(buf.decode("utf-8", "replace")))
+        return tuple(nat.parse_nmf_bytes(buf))
     return parse_nmf(read_split(path, split))
 
 
 def parse_libsvm_split(path: str, rank: int, world_size: int,
                        num_features: int):
     split = compute_splits(path, world_size)[rank]
-    nat = _native()
+    nat = _native(world_size)
     if nat is not None:
         buf = _read_split_bytes(path, split)
-        return tuple(nat.parse_libsvm_bytes(buf.decode("utf-8", "replace"),
+        return tuple(nat.parse_libsvm_bytes(buf,
                                             num_features))
     return parse_libsvm(read_split(path, split), num_features)
 
 
 def parse_lda_split(path: str, rank: int, world_size: int):
     split = compute_splits(path, world_size)[rank]
-    nat = _native()
+    nat = _native(world_size)
     if nat is not None:
         buf = _read_split_bytes(path, split)
-        return tuple(nat.parse_lda_bytes(buf.decode("utf-8", "replace")))
+        return tuple(nat.parse_lda_bytes(buf))
     return parse_lda(read_split(path, split))
 
 

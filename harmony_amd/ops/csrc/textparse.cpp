@@ -72,6 +72,14 @@ std::vector<std::pair<const char*, const char*>> line_chunks(
 }
 
 int nthreads() {
+  // HARMONY_PARSE_THREADS caps the per-process scan threads: with W
+  // loader processes per node (one per GPU), W * hw threads oversubscribe
+  // the cores 8x and parsing runs 5x SLOWER than linear (measured in
+  // scripts/ingest_bench.py) — the loader sets it to cores/world.
+  if (const char* e = std::getenv("HARMONY_PARSE_THREADS")) {
+    int v = std::atoi(e);
+    if (v > 0) return std::min(v, 64);
+  }
   unsigned hw = std::thread::hardware_concurrency();
   return std::max(1u, std::min(hw, 16u));
 }
