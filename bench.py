@@ -165,7 +165,9 @@ def run_runtime_mode(args, ctx, cp, jobs):
         nb = job.num_mini_batches
         epochs = 1 + max(1, -(-args.steps // nb))   # 1 warmup + timed
         shaped[name] = dataclasses.replace(
-            job, job_id=job.job_id + "_rt", max_num_epochs=epochs)
+            job, job_id=job.job_id + "_rt", max_num_epochs=epochs,
+            trace_path=(os.environ.get("HARMONY_BENCH_TRACE", "") + name
+                        if os.environ.get("HARMONY_BENCH_TRACE") else None))
     tus = TaskUnitScheduler(cp, {j.job_id for j in shaped.values()},
                             multi_job=len(shaped) > 1)
     for j in shaped.values():
@@ -200,6 +202,11 @@ def run_runtime_mode(args, ctx, cp, jobs):
         eps = [e for e in mc.epochs if e.epoch_idx >= 1]
         ex = sum(e.num_examples for e in eps)
         tsec = sum(e.epoch_time_sec for e in eps)
+        steady = [b for b in mc.batches if b.epoch_idx >= 1]
+        net_ms = (sum(b.net_wait_sec for b in steady)
+                  / max(1, len(steady)) * 1e3)
+        bt_ms = (sum(b.batch_time_sec for b in steady)
+                 / max(1, len(steady)) * 1e3)
         # MAX time over ranks, SUM examples over ranks
         if dist.is_initialized():
             tt = torch.tensor([tsec], dtype=torch.float64,
@@ -214,12 +221,26 @@ def run_runtime_mode(args, ctx, cp, jobs):
         nb = jobs[name].num_mini_batches
         per_job[name] = {"examples_per_sec": rate,
                          "timed_epochs": len(eps),
-                         "ms_per_batch": tsec / max(1, len(eps) * nb) * 1e3}
+                         "ms_per_batch": tsec / max(1, len(eps) * nb) * 1e3,
+                         # measured control-plane cost:
+                         # NET-ticket sequencer wait per batch and its
+                         # share of the batch wall time
+                         "net_wait_ms": net_ms,
+                         "net_wait_pct": (100.0 * net_ms / bt_ms
+                                          if bt_ms else 0.0)}
         total_rate += rate
         steps_total += len(eps) * nb
     if rank == 0:
+        ctl = {j: {"draw_s": round(tus.stat_draw_s.get(j, 0.0), 3),
+                   "order_s": round(tus.stat_order_s.get(j, 0.0), 3)}
+               for j in sorted(tus.stat_draw_s)}
+        if os.environ.get("HARMONY_DUMP_ORDER"):
+            seqs = sorted(tus._job_cache)
+            print("ORDER:", "".join(tus._job_cache[s][6] for s in seqs),
+                  file=__import__("sys").stderr)
         out = {
             "metric": "aggregate_examples_per_sec_3job_runtime",
+            "rank0_control_split": ctl,
             "value": total_rate,
             "unit": "examples/s",
             "n_gpus": world,
@@ -245,6 +266,15 @@ def run_runtime_mode(args, ctx, cp, jobs):
 
 
 def main():
+    if os.environ.get("HARMONY_DUMP_STACKS"):
+        import faulthandler
+
+        rank = os.environ.get("RANK", "0")
+        global _stackf
+        _stackf = open(f"/tmp/bench_stacks_r{rank}.txt", "w", buffering=1)
+        faulthandler.dump_traceback_later(
+            int(os.environ["HARMONY_DUMP_STACKS"]), exit=False,
+            file=_stackf, repeat=True)
     args = parse_args()
     from harmony_amd.config import RuntimeConfig
     from harmony_amd.runtime.bootstrap import init_executor

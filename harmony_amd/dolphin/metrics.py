@@ -21,6 +21,7 @@ class BatchMetrics:
     pull_time_sec: float
     comp_time_sec: float
     push_time_sec: float
+    net_wait_sec: float = 0.0   # NET-ticket sequencer wait (control plane)
 
     @property
     def data_processing_rate(self) -> float:

@@ -64,6 +64,10 @@ class WorkerTasklet:
         round-trip (both phases are guaranteed to run for every batch)."""
         if self.is_async:
             return contextlib.nullcontext()
+        import os as _os
+
+        if _os.environ.get("HARMONY_NO_LOOKAHEAD"):
+            lookahead = 0
         return self.tus.net(jid, self._next_phase(), lookahead)
 
     def _next_phase(self) -> int:
@@ -115,11 +119,17 @@ class WorkerTasklet:
                         break
                     b_t0 = time.perf_counter()
                     self.trainer.set_batch_data(batch)
-                    # PULL
+                    # PULL (net-ticket wait timed separately: the measured
+                    # control-plane cost of the global NET sequencer)
                     t0 = time.perf_counter()
                     with self.tracer.span("pull"):
-                        with self._net(jid, lookahead=1):
+                        cm = self._net(jid, lookahead=1)
+                        cm.__enter__()
+                        net_t = time.perf_counter() - t0
+                        try:
                             self.trainer.pull_model()
+                        finally:
+                            cm.__exit__(None, None, None)
                     pull_t = time.perf_counter() - t0
                     # COMP
                     t0 = time.perf_counter()
@@ -129,8 +139,14 @@ class WorkerTasklet:
                     # PUSH
                     t0 = time.perf_counter()
                     with self.tracer.span("push"):
-                        with self._net(jid):
+                        cm = self._net(jid)
+                        t1 = time.perf_counter()
+                        cm.__enter__()
+                        net_t += time.perf_counter() - t1
+                        try:
                             self.trainer.push_update()
+                        finally:
+                            cm.__exit__(None, None, None)
                     push_t = time.perf_counter() - t0
                     n = self.trainer.num_batch_examples()
                     ep_examples += n
@@ -139,7 +155,7 @@ class WorkerTasklet:
                         epoch_idx=epoch, batch_idx=bidx, num_examples=n,
                         batch_time_sec=bt,
                         pull_time_sec=pull_t, comp_time_sec=comp_t,
-                        push_time_sec=push_t))
+                        push_time_sec=push_t, net_wait_sec=net_t))
                     # optimization window (reference RUN<->OPTIMIZE state):
                     # a quiesced gap between batches where an elasticity
                     # plan (block migration) applies collectively

@@ -235,6 +235,10 @@ class TaskUnitScheduler:
         self._job_cache: Dict[int, str] = {}
         self._drawer: Dict[str, bool] = {}   # job -> am I its ticket drawer?
         self._pre: Dict[Tuple[str, int], int] = {}  # prefetched (job, phase)->seq
+        # cumulative control-cost split (diagnostics): ticket draw/lookup
+        # vs order wait, per job
+        self.stat_draw_s: Dict[str, float] = {}
+        self.stat_order_s: Dict[str, float] = {}
         self._lock = threading.Lock()
         self._cv = threading.Condition(self._lock)
 
@@ -339,7 +343,11 @@ class TaskUnitScheduler:
     def _enter(self, job_id: str, phase_idx: int, lookahead: int = 0) -> int:
         if not self.multi_job:
             return -1
+        t0 = time.perf_counter()
         seq = self._ticket(job_id, phase_idx, lookahead)
+        t1 = time.perf_counter()
+        self.stat_draw_s[job_id] = self.stat_draw_s.get(job_id, 0.0) \
+            + (t1 - t0)
         # Wait until all earlier tickets of my jobs completed locally.
         # Seqs of jobs this rank does not run are marked done immediately.
         # The blocker check + wait happen UNDER the lock (an _exit notify
@@ -374,6 +382,8 @@ class TaskUnitScheduler:
                     self._job_of(s)       # store I/O, populates _job_cache
             elif waits and waits % 20 == 0:
                 self.cp.check_failed()    # store I/O — outside the lock
+        self.stat_order_s[job_id] = self.stat_order_s.get(job_id, 0.0) \
+            + (time.perf_counter() - t1)
         # record in ISSUE order (post-wait): this is the order the
         # rank actually enqueues the phase's collectives
         if sanitize.enabled():
