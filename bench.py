@@ -58,6 +58,11 @@ def parse_args():
     p.add_argument("--lda-alias-refresh", type=int, default=4)
     p.add_argument("--lda-sampler", type=str, default="alias",
                    choices=["exact", "alias"])
+    p.add_argument("--elastic", action="store_true",
+                   help="runtime mode only: mid-run StopWorker + live "
+                        "block migration of the last rank, then StartWorker"
+                        " + migration back (BASELINE config #5); emits an "
+                        "elastic_timeline field")
     return p.parse_args()
 
 
@@ -166,6 +171,8 @@ def run_runtime_mode(args, ctx, cp, jobs):
         epochs = 1 + max(1, -(-args.steps // nb))   # 1 warmup + timed
         shaped[name] = dataclasses.replace(
             job, job_id=job.job_id + "_rt", max_num_epochs=epochs,
+            optimizer_period=(max(2, args.steps // 5)
+                              if args.elastic and name == "nmf" else 8),
             trace_path=(os.environ.get("HARMONY_BENCH_TRACE", "") + name
                         if os.environ.get("HARMONY_BENCH_TRACE") else None))
     tus = TaskUnitScheduler(cp, {j.job_id for j in shaped.values()},
@@ -176,7 +183,13 @@ def run_runtime_mode(args, ctx, cp, jobs):
 
     def runner(name, job):
         stream = torch.cuda.Stream() if dev_cuda else None
-        results[name] = run_job(job, ctx, cp=cp, tus=tus, stream=stream)
+        opt = None
+        if args.elastic and name == "nmf" and world > 1:
+            from harmony_amd.optimizer.optimizers import SampleOptimizers
+
+            opt = SampleOptimizers.elastic_showcase(world - 1)
+        results[name] = run_job(job, ctx, cp=cp, tus=tus, stream=stream,
+                                optimizer=opt)
 
     if dev_cuda:
         torch.cuda.synchronize()
@@ -230,6 +243,18 @@ def run_runtime_mode(args, ctx, cp, jobs):
                                           if bt_ms else 0.0)}
         total_rate += rate
         steps_total += len(eps) * nb
+    timeline = None
+    if args.elastic and "nmf" in results:
+        print(f"[elastic] rank{rank} applied_plans="
+              f"{getattr(results['nmf'], '_applied_plans', '?')}",
+              file=__import__('sys').stderr)
+    if args.elastic and "nmf" in results and dist.is_initialized():
+        mine = [(b.epoch_idx, b.batch_idx,
+                 round(b.batch_time_sec * 1e3, 2), b.num_examples)
+                for b in results["nmf"].batches]
+        allt = [None] * world
+        dist.all_gather_object(allt, mine)
+        timeline = allt
     if rank == 0:
         ctl = {j: {"draw_s": round(tus.stat_draw_s.get(j, 0.0), 3),
                    "order_s": round(tus.stat_order_s.get(j, 0.0), 3)}
@@ -241,6 +266,7 @@ def run_runtime_mode(args, ctx, cp, jobs):
         out = {
             "metric": "aggregate_examples_per_sec_3job_runtime",
             "rank0_control_split": ctl,
+            "elastic_timeline": timeline,
             "value": total_rate,
             "unit": "examples/s",
             "n_gpus": world,

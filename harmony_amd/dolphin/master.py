@@ -144,6 +144,8 @@ def run_job(job: JobConfig, ctx: ExecutorContext,
 
         trainer.on_epoch_finished = _hook
     metrics = tasklet.run()
+    if orch is not None:
+        metrics._applied_plans = orch.applied_plans
     if tracer is not None:
         tracer.flush()
     # evaluate_model may issue collectives (e.g. LDA's log-likelihood does

@@ -193,6 +193,45 @@ class SampleOptimizers:
         return _SS()
 
     @staticmethod
+    def elastic_showcase(rank: int, stop_at_call: int = 1,
+                         start_at_call: int = 2):
+        """The BASELINE config-#5 showcase: mid-run, STOP worker `rank`
+        AND live-migrate all its model blocks away (ownership-first); a
+        few windows later, START it again and migrate an even share back.
+        One plan per event, compiled with the reference's switch ordering
+        (stop -> move -> start, PlanCompiler.translateToSwitch)."""
+        from harmony_amd.optimizer.plan import compile_switch
+
+        class _E(Optimizer):
+            def __init__(self):
+                self.calls = 0
+
+            def optimize(self, metrics, owners, world_size):
+                self.calls += 1
+                tid = sorted(owners)[0]
+                ol = owners[tid]
+                if self.calls == stop_at_call:
+                    others = [r for r in range(world_size) if r != rank]
+                    moves = {b: others[i % len(others)] for i, b in
+                             enumerate(i for i, o in enumerate(ol)
+                                       if o == rank)}
+                    return compile_switch(tid, [rank], {}, moves)
+                if self.calls == start_at_call:
+                    total = len(ol)
+                    per = total // world_size
+                    mine = [i for i, o in enumerate(ol) if o == rank]
+                    need = per - len(mine)
+                    moves = {}
+                    if need > 0:
+                        donors = [i for i, o in enumerate(ol) if o != rank]
+                        for b in donors[:need]:
+                            moves[b] = rank
+                    return compile_switch(tid, [], {rank: 1}, moves)
+                return Plan()
+
+        return _E()
+
+    @staticmethod
     def even_rebalance(table_id: str):
         class _Even(Optimizer):
             def optimize(self, metrics, owners, world_size):

@@ -101,8 +101,13 @@ class OptimizationOrchestrator:
         if self._batches % self.period != 0:
             return None
         at = self._batches
-        if self.rank == 0 and self.optimizer is not None:
-            self._decide(at, plan_key_base=f"opt/{self.job_id}/plan")
+        if self.rank == 0 and self.optimizer is not None and at > self.period:
+            # decide on the PREVIOUS window's metrics: co-located ranks are
+            # within one batch of each other, so the current window's
+            # reports race rank 0's check (observed: perpetual skips when
+            # jobs co-run); the previous window is guaranteed complete
+            self._decide(at - self.period,
+                         plan_key_base=f"opt/{self.job_id}/plan")
         my_key = f"opt/{self.job_id}/plan/{at}"
         if self.cp.flag_set(my_key):
             plan = Plan.from_json(self.cp.store.get(my_key).decode())
@@ -126,5 +131,7 @@ class OptimizationOrchestrator:
                   for t in self.tables.values() if hasattr(t, "cfg")}
         plan = self.optimizer.optimize(metrics, owners, self.world_size)
         if plan is not None and not plan.empty():
-            self.cp.store.set(f"{plan_key_base}/{at + self.period}",
+            # applies two windows after `at` (one after the caller's
+            # current boundary — see boundary_plan's at-period shift)
+            self.cp.store.set(f"{plan_key_base}/{at + 2 * self.period}",
                               plan.to_json())
