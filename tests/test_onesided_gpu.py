@@ -28,6 +28,20 @@ def _store(rank, world):
                     wait_for_workers=False)
 
 
+def _finish(store, rank, world):
+    """End-of-worker sync: rank 0 hosts the TCPStore server, so it must
+    exit LAST — non-zero ranks signal and leave; rank 0 polls its own
+    server until everyone signalled (exiting earlier kills the server
+    under a peer still polling the final barrier — observed flake)."""
+    import time
+
+    if rank == 0:
+        while int(store.add("bye", 0)) < world - 1:
+            time.sleep(0.005)
+    else:
+        store.add("bye", 1)
+
+
 def _barrier(store, name, rank, world):
     store.add(f"b/{name}", 1)
     store.wait([f"b/{name}"])
@@ -73,6 +87,7 @@ def _async_worker(rank, world):
         diag.append((r_, c_, float(full[r_, c_]), owner))
     _barrier(store, "checked", rank, world)
     t.close()
+    _finish(store, rank, world)
     return (ok, full.unique().tolist(), diag)
 
 
@@ -119,6 +134,7 @@ def _async_mlr_worker(rank, world):
     acc = float(((X @ W.t()).argmax(dim=1) == y).float().mean())
     _barrier(store, "checked", rank, world)
     t.close()
+    _finish(store, rank, world)
     return acc
 
 
@@ -243,6 +259,7 @@ def _async_lda_conservation_worker(rank, world):
     summ_total = int(full[V].sum())
     cp.barrier("os_lda_c/checked", ctx.world_size)
     t.close()
+    _finish(ctx.store, ctx.rank, ctx.world_size)
     return (wt_total, summ_total, total_tokens)
 
 
@@ -289,7 +306,8 @@ def _ring_worker(rank, world):
     for _ in range(rounds):
         t.push(rkeys, d)
         t.drain()
-    store.add("pushed_done", 1)
+    t.fence()          # my enqueued remote pushes must be COMPLETE before
+    store.add("pushed_done", 1)   # the peer's final-drain loop can trust 0
     while int(store.add("pushed_done", 0)) < world:
         t.drain()
         time.sleep(0.002)
@@ -355,6 +373,7 @@ def _ring_worker(rank, world):
         assert v > 0.05 * 127, v
     _barrier(store, "done", rank, world)
     t.close()
+    _finish(store, rank, world)
     return True
 
 
