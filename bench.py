@@ -176,7 +176,7 @@ def run_runtime_mode(args, ctx, cp, jobs):
             trace_path=(os.environ.get("HARMONY_BENCH_TRACE", "") + name
                         if os.environ.get("HARMONY_BENCH_TRACE") else None))
     tus = TaskUnitScheduler(cp, {j.job_id for j in shaped.values()},
-                            multi_job=len(shaped) > 1)
+                            multi_job=len(shaped) > 1 and world > 1)
     for j in shaped.values():
         tus.set_drawer(j.job_id, rank == 0)
     results = {}

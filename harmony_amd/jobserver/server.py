@@ -67,7 +67,9 @@ class JobServerDriver:
         self.ctx = ctx
         self.hb_period = hb_period
         self.cp = ControlPlane(ctx.store, ctx.rank, ctx.world_size)
-        self.tus = TaskUnitScheduler(self.cp, {ADMIN_JOB}, multi_job=True)
+        # world 1: jobs use local tables (no collectives) — no ordering
+        self.tus = TaskUnitScheduler(self.cp, {ADMIN_JOB},
+                                     multi_job=ctx.world_size > 1)
         self.tus.set_drawer(ADMIN_JOB, ctx.rank == 0)
         self.pool = ResourcePool(ctx.world_size)
         self.scheduler = load_scheduler(scheduler)
