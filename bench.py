@@ -56,8 +56,8 @@ def parse_args():
     p.add_argument("--lda-docs-per-batch", type=int, default=16384)
     p.add_argument("--lda-tokens-per-doc", type=int, default=128)
     p.add_argument("--lda-alias-refresh", type=int, default=4)
-    p.add_argument("--lda-sampler", type=str, default="alias",
-                   choices=["exact", "alias"])
+    p.add_argument("--lda-sampler", type=str, default="alias_wave",
+                   choices=["exact", "alias", "alias_wave"])
     p.add_argument("--elastic", action="store_true",
                    help="runtime mode only: mid-run StopWorker + live "
                         "block migration of the last rank, then StartWorker"
@@ -397,7 +397,8 @@ def main():
                         "batch": args.mlr_batch},
                 "lda": {"vocab": args.lda_vocab, "topics": args.lda_topics,
                         "docs_per_batch": args.lda_docs_per_batch,
-                        "tokens_per_doc": args.lda_tokens_per_doc},
+                        "tokens_per_doc": args.lda_tokens_per_doc,
+                        "sampler": args.lda_sampler},
             },
         }
         print(json.dumps(out))

@@ -40,6 +40,7 @@ def defaults(job: JobConfig) -> dict:
     a = dict(num_docs=32768, num_vocabs=30000, num_topics=256,
              tokens_per_doc=64, alpha=0.1, beta=0.01, docs_per_batch=4096,
              sampler="exact",   # "exact" (dense Gibbs) | "alias" (MH, K7b)
+                                # | "alias_wave" (K7c wave-per-doc MH)
              alias_refresh=4)   # rebuild alias tables every N pulls
     a.update(job.app_args)
     return a
@@ -207,6 +208,9 @@ class LDATrainer(Trainer):
         self.word_topic = pulled[:-1]          # [n_uniq_words, K]
         self.topic_sum = pulled[-1]            # [K]
 
+    def _is_alias(self) -> bool:
+        return self.a["sampler"] in ("alias", "alias_wave")
+
     def _ensure_alias(self) -> None:
         # two-level alias tables over the word factor (K7b), rebuilt every
         # alias_refresh uses OF THIS BLOCK (tables index the block's local
@@ -240,16 +244,18 @@ class LDATrainer(Trainer):
             dt = self.doc_topic[b.doc_lo:b.doc_lo + b.num_examples]
         else:
             dt = self.doc_topic[b.doc_ids]      # gather copy
-        if self.a["sampler"] == "alias":
+        if self._is_alias():
             self._ensure_alias()
             prob, alias, tprob, talias, qv, _, invden = self._alias
+            fn = (ops.lda_mh_wave if self.a["sampler"] == "alias_wave"
+                  else ops.lda_mh)
             # NOTE: invden/qv index LOCAL word ids of the batch on which the
             # tables were built; with one static key set per block this is
             # consistent across refreshes of the same block
-            ops.lda_mh(dt, self.word_topic, invden, prob, alias, tprob,
-                       talias, qv, b.doc_offsets, b.word_local, z,
-                       self.a["alpha"], self.a["beta"],
-                       self._epoch_seed + self._step)
+            fn(dt, self.word_topic, invden, prob, alias, tprob,
+               talias, qv, b.doc_offsets, b.word_local, z,
+               self.a["alpha"], self.a["beta"],
+               self._epoch_seed + self._step)
         else:
             ops.lda_gibbs(dt, self.word_topic,
                           self.topic_sum, b.doc_offsets, b.word_local, z,

@@ -440,6 +440,31 @@ def lda_alias_build(word_topic: torch.Tensor, topic_sum: torch.Tensor,
     return prob, alias, top_prob, top_alias, qv, total, invden
 
 
+def lda_mh_wave(doc_topic: torch.Tensor, word_topic: torch.Tensor,
+                invden: torch.Tensor, prob: torch.Tensor,
+                alias: torch.Tensor, top_prob: torch.Tensor,
+                top_alias: torch.Tensor, qv: torch.Tensor,
+                doc_offsets: torch.Tensor, word_ids: torch.Tensor,
+                assignments: torch.Tensor, alpha: float, beta: float,
+                seed: int) -> torch.Tensor:
+    """Wave-per-doc MH sweep (K7c): 64 lanes cooperate on each doc with
+    the doc-topic row shared in LDS via atomics — within-doc token
+    updates are approximately parallel (acceptance may read slightly
+    stale counts; standard GPU-LDA relaxation, convergence validated by
+    scripts/lda_convergence.py). On CPU this falls back to the SERIAL
+    sampler: the wave kernel's interleaving is hardware-scheduling
+    dependent, so there is no bit-exact oracle — the two are compared
+    statistically, not elementwise."""
+    if _use_hip(word_topic):
+        return _hip.lda_mh_wave(doc_topic, word_topic, invden, prob, alias,
+                                top_prob, top_alias, qv, doc_offsets,
+                                word_ids, assignments, float(alpha),
+                                float(beta), int(seed))
+    return lda_mh(doc_topic, word_topic, invden, prob, alias, top_prob,
+                  top_alias, qv, doc_offsets, word_ids, assignments,
+                  alpha, beta, seed)
+
+
 def lda_mh(doc_topic: torch.Tensor, word_topic: torch.Tensor,
            invden: torch.Tensor, prob: torch.Tensor, alias: torch.Tensor,
            top_prob: torch.Tensor, top_alias: torch.Tensor,

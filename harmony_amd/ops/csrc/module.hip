@@ -41,6 +41,13 @@ void lda_apply_all(torch::Tensor shard, torch::Tensor word_rows,
 std::vector<torch::Tensor> lda_alias_build(torch::Tensor word_topic,
                                            torch::Tensor topic_sum,
                                            double beta, int64_t num_vocabs);
+torch::Tensor lda_mh_wave(torch::Tensor doc_topic, torch::Tensor word_topic,
+                          torch::Tensor invden, torch::Tensor prob,
+                          torch::Tensor alias, torch::Tensor top_prob,
+                          torch::Tensor top_alias, torch::Tensor qv,
+                          torch::Tensor doc_offsets,
+                          torch::Tensor word_ids, torch::Tensor assignments,
+                          double alpha, double beta, int64_t seed);
 torch::Tensor lda_mh(torch::Tensor doc_topic, torch::Tensor word_topic,
                      torch::Tensor invden, torch::Tensor prob,
                      torch::Tensor alias, torch::Tensor top_prob,
@@ -104,6 +111,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "per-word Vose alias tables over the stale word factor (K7b)");
   m.def("lda_mh", &lda_mh,
         "Metropolis-Hastings alias LDA sweep, thread-per-doc (K7b)");
+  m.def("lda_mh_wave", &lda_mh_wave,
+        "wave-per-doc MH sweep (K7c, approximate within-doc parallelism)");
   m.def("gbt_hist", &gbt_hist, "GBT level histogram build (K10)");
   m.def("lasso_cd", &lasso_cd, "Lasso persistent CD sweep (K11)");
   m.def("os_shard_alloc", &os_shard_alloc, "one-sided shard (hipMalloc)");

@@ -42,7 +42,7 @@ def run(sampler, epochs=8, K=256):
 
 
 if __name__ == "__main__":
-    for sampler in ("exact", "alias"):
+    for sampler in ("exact", "alias", "alias_wave"):
         print(f"== {sampler}")
         for ep, t, ll in run(sampler):
             print(f"epoch {ep}  cum_time {t:8.3f}s  ll {ll:10.3f}M nats")
