@@ -263,7 +263,13 @@ class OneSidedTable(Table):
 
         n = int(keys.shape[0])
         cap, vd = self._ring_cap, self.cfg.value_dim
-        assert n <= cap, f"single push of {n} items exceeds ring cap {cap}"
+        if n > cap:
+            # a push larger than the ring: send in capacity-sized chunks
+            # (each chunk applies backpressure; per-writer FIFO holds)
+            for off in range(0, n, cap):
+                self._ring_push_to(r, keys[off:off + cap],
+                                   deltas[off:off + cap])
+            return
         # exact host-side backpressure: we are the only writer of our slot
         deadline = time.monotonic() + 30.0
         while self._pushed[r] + n - self._head_cache[r] > cap:
@@ -274,6 +280,10 @@ class OneSidedTable(Table):
             self._head_cache[r] = int(self._scratch[0])
             if self._pushed[r] + n - self._head_cache[r] <= cap:
                 break
+            # drain OUR ring while waiting: two ranks pushing large
+            # batches at each other would otherwise deadlock in mutual
+            # backpressure before either reaches its own drain
+            self.drain()
             if time.monotonic() > deadline:
                 raise RuntimeError(
                     f"one-sided ring to rank {r} full for 30s "

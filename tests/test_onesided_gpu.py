@@ -403,3 +403,56 @@ def test_async_nmf_job_one_sided_rings():
     for n, sq in res:
         assert n == 8
         assert sq is not None
+
+
+def _ring_big_worker(rank, world):
+    """Pushes larger than ring capacity chunk + mutually backpressure
+    without deadlock (both ranks push 4x cap at each other)."""
+    import time
+
+    from harmony_amd.config import TableConfig
+    from harmony_amd.et.onesided import OneSidedTable
+
+    torch.cuda.set_device(0)
+    store = _store(rank, world)
+    cfg = TableConfig(table_id="os_big", num_keys=64, value_dim=4,
+                      num_blocks=8, update_fn="nmf_sgd",
+                      update_args={"step_size": 1.0, "max_val": 1e9},
+                      init_fn="zeros")
+    t = OneSidedTable(cfg, rank, world, torch.device("cuda"), store=store,
+                      ring_capacity=64)
+    _barrier(store, "alloc", rank, world)
+    t.connect()
+    _barrier(store, "conn", rank, world)
+    other = 1 - rank
+    all_keys = torch.arange(64, device="cuda")
+    owners = t._owner_of(t.part.block_of(all_keys))
+    rkeys = all_keys[owners == other]
+    # one oversized push: 4x capacity (8 repeats of 32 keys = 256 items)
+    big_k = rkeys.repeat(8)
+    big_d = torch.full((big_k.shape[0], 4), -1.0, device="cuda")
+    t.push(big_k, big_d)          # chunks at cap=64, mutual backpressure
+    t.fence()
+    store.add("pushed_done", 1)
+    while int(store.add("pushed_done", 0)) < world:
+        t.drain()
+        time.sleep(0.002)
+    empties = 0
+    while empties < 3:
+        empties = empties + 1 if t.drain() == 0 else 0
+        time.sleep(0.002)
+    t.fence()
+    _barrier(store, "ok", rank, world)
+    mine = all_keys[owners == rank]
+    got = t.pull(mine)
+    # 8 pushes of -1 with step 1 => value 8.0 everywhere
+    assert torch.allclose(got, torch.full_like(got, 8.0)), got[:2]
+    _barrier(store, "done", rank, world)
+    t.close()
+    _finish(store, rank, world)
+    return True
+
+
+def test_onesided_ring_oversized_push_no_deadlock():
+    assert all(run_dist(_ring_big_worker, world=2, timeout=300))
+
