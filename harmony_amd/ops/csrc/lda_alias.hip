@@ -443,7 +443,9 @@ torch::Tensor lda_mh_wave(torch::Tensor doc_topic, torch::Tensor word_topic,
   CHECK_IN(assignments);
   const int D = doc_topic.size(0), K = doc_topic.size(1);
   if (D == 0) return assignments;
-  const int waves_wg = 4;              // 256 threads, 4 docs per block
+  int waves_wg = 4;                    // 256 threads, 4 docs per block
+  const char* we = getenv("HARMONY_LDA_WAVE_WG");   // occupancy A/B
+  if (we) { int v = atoi(we); if (v >= 1 && v <= 4) waves_wg = v; }  // <=4: 256-thread launch bound
   dim3 blk(WAVE * waves_wg), grid((D + waves_wg - 1) / waves_wg);
   const size_t shmem = (size_t)waves_wg * K * 4;
   hipLaunchKernelGGL(lda_mh_wave_kernel, grid, blk, shmem, current_stream(),
