@@ -52,3 +52,26 @@ def test_bench_runtime_mode_contract():
     for job in ("nmf", "mlr", "lda"):
         pj = d["per_job"][job]
         assert pj["examples_per_sec"] > 0 and pj["timed_epochs"] >= 1
+
+
+def test_bench_elastic_contract():
+    """--mode runtime --elastic at world 2: the timeline shows the
+    stopped rank's zero-example window and recovery (BASELINE config #5
+    through the bench contract)."""
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29734", "bench.py", "--mode", "runtime",
+         "--elastic", "--gpus", "2", "--steps", "40", "--device", "cpu",
+         "--apps", "nmf,mlr", "--nmf-cols", "2048",
+         "--nmf-rows-per-batch", "256", "--nmf-nnz-per-row", "8",
+         "--mlr-features", "512", "--mlr-batch", "256"],
+        capture_output=True, text=True, timeout=600)
+    assert out.returncode == 0, out.stderr[-2000:]
+    line = [l for l in out.stdout.splitlines() if '"metric"' in l][-1]
+    d = json.loads(line)
+    tl = d["elastic_timeline"]
+    assert tl is not None and len(tl) == 2
+    ex1 = [row[3] for row in tl[1]]
+    assert 0 in ex1, ex1[:8]            # rank 1 truly stopped mid-run
+    assert ex1[0] > 0 and ex1[-1] > 0   # worked at both ends (restarted)
