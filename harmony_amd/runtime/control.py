@@ -212,17 +212,25 @@ class TaskUnitScheduler:
         with tus.net(job_id, phase_idx):
             ... issue collectives ...
 
-    Ticket allocation: the first rank to reach phase (job, idx) wins a
-    compare_set race, draws seq = incr("tu/seq"), and publishes
-    tu/seq_of/<job>/<idx> = seq and tu/job_of/<seq> = job. Tickets are
-    therefore gap-free and per-job monotone (a job's phase N+1 is only
-    requested after phase N's collective completed somewhere, which is after
-    phase N drew its ticket).
+    Ticket allocation (v2 — measured 4.5x cheaper than round 1 at 8
+    ranks x 3 jobs, profiles/r02_control_plane.md): the job's designated
+    DRAWER (job-local rank 0 via set_drawer; compare_set race as the
+    membership-free fallback) draws seq = incr("tu/seq", 1+lookahead) —
+    the worker's PULL draws the PUSH ticket too — appends seq=job to a
+    paged log (tu/log/<seq/256>) and publishes every drawn seq in ONE
+    multi_set; every other rank learns the whole batch from ONE blocking
+    get. Tickets are gap-free and per-job monotone (a job's phase N+1 is
+    only requested after phase N's collective completed somewhere, which
+    is after phase N drew its ticket); call sites must use identical
+    lookahead on every rank.
 
     Each rank runs phases in ticket order filtered to `my_jobs`: before
-    entering seq S it waits until every seq < S that belongs to one of its
-    jobs has locally completed. Single-job mode short-circuits (no store
-    traffic) — ordering is only needed when jobs co-locate.
+    entering seq S it waits (condition variable held across the check —
+    a missed notify costs a poll timeout otherwise) until every seq < S
+    that belongs to one of its jobs has locally completed; seq->job for
+    foreign seqs comes from the paged log. Single-job mode
+    short-circuits (no store traffic) — ordering is only needed when
+    jobs co-locate AND collectives exist (world > 1).
     """
 
     def __init__(self, cp: ControlPlane, my_jobs: Optional[Set[str]] = None,
