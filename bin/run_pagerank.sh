@@ -1,0 +1,6 @@
+#!/bin/bash
+# Standalone pagerank run without the job server (reference run_pagerank.sh /
+# ETDolphinLauncher mode). Multi-GPU: wrap with torchrun (see
+# harmony_amd/standalone.py).
+cd "$(dirname "$0")/.."
+exec python -m harmony_amd.standalone -app pagerank "$@"

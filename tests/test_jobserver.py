@@ -315,3 +315,20 @@ def test_jobserver_pregel_submit():
     assert r["per_rank"][0]["supersteps"] >= 5
     client.shutdown(port=port)
     t.join(timeout=30)
+
+
+def test_standalone_launcher_cli():
+    """run_*.sh / ETDolphinLauncher standalone mode (no jobserver):
+    reference dolphin/core/client/ETDolphinLauncher.java:111,219."""
+    import subprocess
+    import sys as _sys
+
+    out = subprocess.run(
+        [_sys.executable, "-m", "harmony_amd.standalone", "-app", "nmf",
+         "-device", "cpu", "-max_num_epochs", "2", "-num_mini_batches", "2",
+         "-num_cols", "64", "-rank", "8", "-nnz_per_row", "4",
+         "-rows_per_batch", "32"],
+        capture_output=True, text=True, timeout=300)
+    assert out.returncode == 0, out.stderr[-1500:]
+    s = json.loads(out.stdout.strip().splitlines()[-1])
+    assert s["num_batches"] == 4 and "sq_err" in s
