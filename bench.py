@@ -58,6 +58,11 @@ def parse_args():
     p.add_argument("--lda-alias-refresh", type=int, default=4)
     p.add_argument("--lda-sampler", type=str, default="alias_wave",
                    choices=["exact", "alias", "alias_wave"])
+    p.add_argument("--cu-partition", type=str, default="",
+                   help="direct mode: pin each job's stream to a disjoint "
+                        "CU range, e.g. 'lda=96,nmf=96,mlr=64' (multi-"
+                        "tenant partitioning via hipExtStreamCreateWith"
+                        "CUMask; empty = shared full chip)")
     p.add_argument("--elastic", action="store_true",
                    help="runtime mode only: mid-run StopWorker + live "
                         "block migration of the last rank, then StartWorker"
@@ -319,8 +324,21 @@ def main():
                             multi_job=multi)
     for j in jobs.values():
         tus.set_drawer(j.job_id, rank == 0)
-    benches = [JobBench(j, ctx, cp, tus, use_stream=dev_cuda)
-               for j in jobs.values()]
+    cu_streams = None
+    if args.cu_partition and dev_cuda:
+        from harmony_amd.utils.custreams import cu_partitioned_streams
+
+        shares = {}
+        for part in args.cu_partition.split(","):
+            name, n = part.split("=")
+            shares[name.strip()] = int(n)
+        cu_streams = cu_partitioned_streams(shares)
+    benches = []
+    for name, j in jobs.items():
+        b = JobBench(j, ctx, cp, tus, use_stream=dev_cuda)
+        if cu_streams is not None and name in cu_streams:
+            b.stream = cu_streams[name]
+        benches.append(b)
     for b in benches:
         b.initialize()
 

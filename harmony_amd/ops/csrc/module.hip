@@ -64,6 +64,8 @@ void os_ipc_close(int64_t ptr);
 torch::Tensor os_gather(int64_t ptr, torch::Tensor idx, int64_t k,
                         int64_t dtype_i32);
 void os_scatter_add(int64_t ptr, torch::Tensor idx, torch::Tensor delta);
+int64_t os_cu_masked_stream(torch::Tensor mask_words);
+void os_stream_destroy(int64_t stream);
 int64_t os_ring_bytes(int64_t W, int64_t cap, int64_t vd);
 void os_ring_reserve(int64_t base, int64_t W, int64_t cap, int64_t vd,
                      int64_t writer, int64_t n, torch::Tensor scratch);
@@ -122,6 +124,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("os_gather", &os_gather, "gather rows from a mapped shard (K12)");
   m.def("os_scatter_add", &os_scatter_add,
         "atomic scatter-add into a mapped shard (K12)");
+  m.def("os_cu_masked_stream", &os_cu_masked_stream,
+        "create a HIP stream pinned to a CU-mask (multi-tenant partitioning)");
+  m.def("os_stream_destroy", &os_stream_destroy, "destroy a raw HIP stream");
   m.def("os_ring_bytes", &os_ring_bytes, "ring buffer size (K12b)");
   m.def("os_ring_reserve", &os_ring_reserve, "reserve seq range in a peer ring");
   m.def("os_ring_read_head", &os_ring_read_head, "read peer ring head");

@@ -195,3 +195,26 @@ void os_scatter_add(int64_t ptr, torch::Tensor idx, torch::Tensor delta) {
                        (float*)(uintptr_t)ptr, idx.data_ptr<int64_t>(),
                        delta.data_ptr<float>(), n, k);
 }
+
+// ---- CU-masked streams (multi-tenant GPU partitioning) ----------------
+// MI355X has no MPS-style preemption; co-located jobs' full-chip kernels
+// timeshare CUs and thrash each other's per-XCD L2. hipExtStreamCreate-
+// WithCUMask pins a stream's kernels to a CU subset — the MI355X-native
+// analogue of the reference's per-executor resource arbitration
+// (LocalTaskUnitScheduler's CPU semaphores). Masks are bit-per-CU words.
+
+int64_t os_cu_masked_stream(torch::Tensor mask_words) {
+  TORCH_CHECK(mask_words.dtype() == torch::kInt32);
+  auto m = mask_words.contiguous();
+  hipStream_t s = nullptr;
+  TORCH_CHECK(hipExtStreamCreateWithCUMask(
+                  &s, (uint32_t)m.numel(),
+                  (const uint32_t*)m.data_ptr<int>()) == hipSuccess,
+              "hipExtStreamCreateWithCUMask failed");
+  return (int64_t)(uintptr_t)s;
+}
+
+void os_stream_destroy(int64_t stream) {
+  TORCH_CHECK(hipStreamDestroy((hipStream_t)(uintptr_t)stream)
+              == hipSuccess);
+}
