@@ -376,3 +376,22 @@ def test_mlr_step_mfma_matches_torch():
         assert rel < 1e-3, (rb, sf, sb, rel)
         assert abs(loss1.item() - loss0.item()) < 1e-2 * abs(loss0.item())
         assert int(corr1) == int(corr0)
+
+
+def test_cu_partitioned_streams():
+    """CU-masked streams (hipExtStreamCreateWithCUMask): disjoint
+    partitions each run kernels to completion with correct results, and
+    events/synchronization work across the external streams."""
+    from harmony_amd.utils.custreams import cu_partitioned_streams
+
+    streams = cu_partitioned_streams({"a": 64, "b": 192})
+    outs = {}
+    for name, s in streams.items():
+        with torch.cuda.stream(s):
+            x = torch.randn(1 << 20, device="cuda")
+            outs[name] = (x, (x * 2 + 1).sum())
+    for name, s in streams.items():
+        s.synchronize()
+        x, got = outs[name]
+        ref = (x * 2 + 1).sum()
+        assert torch.allclose(got, ref), name
