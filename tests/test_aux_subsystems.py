@@ -158,7 +158,7 @@ def test_hetero_milp_optimizer():
     # slower rank gets less data; block totals conserved
     assert d[0] > d[1]
     assert sum(m) == 64
-    opt = HeterogeneousOptimizer(benefit_threshold=0.01)
+    opt = HeterogeneousOptimizer(benefit_threshold=0.01, role_select=False)
     metrics = [RankMetrics(0, 0.30, 0.10, 0.02, 0.02, num_examples=8192),
                RankMetrics(1, 0.30, 0.30, 0.02, 0.02, num_examples=8192)]
     owners = {"t": [0] * 16 + [1] * 16}
@@ -166,6 +166,38 @@ def test_hetero_milp_optimizer():
     assert not plan.empty()
     kinds = {type(op).__name__ for op in plan.ops}
     assert "SetBatchShareOp" in kinds
+
+
+def test_hetero_milp_role_selection():
+    """The (w,s) dimension of the reference's ILPSolver: a much slower
+    machine is demoted to pure server (worker stopped, model blocks moved
+    onto it) with the stop->move->share DAG order."""
+    from harmony_amd.optimizer.hetero import HeterogeneousOptimizer, best_roles
+    from harmony_amd.optimizer.optimizers import RankMetrics
+    from harmony_amd.optimizer.plan import (MoveOp, SetBatchShareOp,
+                                            StopWorkerOp)
+
+    # direct solver: equal machines -> everyone works
+    d, m, w, s, T = best_roles([1e-6] * 4, kappa=1e-5,
+                               total_examples=40000, total_blocks=64)
+    assert all(w) and sum(m) == 64 and abs(sum(d) - 40000) < 1
+    opt = HeterogeneousOptimizer(benefit_threshold=0.01)
+    mk = lambda r, comp: RankMetrics(r, comp + 0.01, comp, 0.005, 0.005,
+                                     num_examples=10000)  # noqa: E731
+    metrics = [mk(0, 0.04), mk(1, 0.04), mk(2, 0.04), mk(3, 0.79)]
+    owners = {"t": [0] * 16 + [1] * 16 + [2] * 16 + [3] * 16}
+    plan = opt.optimize(metrics, owners, 4)
+    stops = [op.rank for op in plan.ops if isinstance(op, StopWorkerOp)]
+    assert stops == [3]
+    share = next(op for op in plan.ops if isinstance(op, SetBatchShareOp))
+    assert 3 not in {r for r, _ in share.shares}
+    mv = next(op for op in plan.ops if isinstance(op, MoveOp))
+    # demoted rank gains blocks (pure server), and order is stop->move->share
+    assert any(dst == 3 for _, dst in mv.moves)
+    si = plan.ops.index(next(o for o in plan.ops
+                             if isinstance(o, StopWorkerOp)))
+    mi, shi = plan.ops.index(mv), plan.ops.index(share)
+    assert (si, mi) in plan.deps and (mi, shi) in plan.deps
 
 
 def test_offline_eval_via_job_flags(tmp_path):
