@@ -51,6 +51,11 @@ def migrate(table, moves: Dict[int, int], rank: int, world_size: int,
              if table.ownership.owner_of_int(b) != d}
     if not moves:
         return
+    if hasattr(table, "migrate_blocks"):
+        # one-sided table: its own quiesce/read/remap protocol (the data
+        # move is a one-sided peer-HBM read, not RCCL p2p)
+        table.migrate_blocks(moves)
+        return
     sends, recvs = plan_transfers(table, moves, rank)
     # 1. ownership-first flip (every rank, identically)
     table.ownership.update_many(moves)

@@ -17,8 +17,13 @@ throughput path; THIS is its async analogue, the MI355X way:
 
 This makes SSP slack > 0 a REAL bounded-async mode: workers proceed at
 their own pace (uneven batch counts, no deadlock) with the SSP clock as
-the only cross-worker coupling. v1 scope: float32 dense tables, static
-ownership (no live migration of one-sided tables), `add` update fn.
+the only cross-worker coupling. v1 scope: float32 dense tables, `add`
+update fn. v2 adds owner-side apply-queue rings for arbitrary update
+fns. v3 adds LIVE MIGRATION (`migrate_blocks`): adopters read moving
+blocks straight out of the owner's HBM (one-sided, no owner
+participation in the data move), then every rank remaps at a
+generation-versioned quiesce point — the ownership-first protocol of
+MigrationExecutor.java:48 expressed as drain -> read -> remap.
 """
 
 from __future__ import annotations
@@ -85,6 +90,8 @@ class OneSidedTable(Table):
         # per-table op stats (reference RemoteAccessOpStat): pulled rows /
         # pushed rows / remote bytes moved over xGMI
         self.stats = {"pull_rows": 0, "push_rows": 0, "remote_bytes": 0}
+        self._gen = 0                # IPC-mapping generation (bumped by
+        #                              migrate_blocks' collective remap)
         # every rank's block->slot map is derivable from the static
         # round-robin ownership, so remote row indices need no exchange
         self._peer_slot = {}
@@ -120,11 +127,13 @@ class OneSidedTable(Table):
         self._block_slot = slot.to(self.device)
 
     def connect(self, store=None) -> None:
-        """Collective: export my shard (+ ring), map every peer's."""
+        """Collective: export my shard (+ ring), map every peer's. Keys are
+        generation-versioned so a re-connect after migration can never read
+        a peer's pre-migration (stale) handle."""
         store = store or self.store
         assert store is not None
         torch.cuda.synchronize()
-        key = f"os/{self.cfg.table_id}"
+        key = f"os/{self.cfg.table_id}/g{self._gen}"
         h = self._hip.os_ipc_handle(self.shard)
         store.set(f"{key}/{self.rank}", bytes(h.tolist()).hex())
         if self._ring_mode:
@@ -344,9 +353,106 @@ class OneSidedTable(Table):
         return self.pull_full()
 
     def drop_blocks(self, blocks) -> None:
-        raise RuntimeError("one-sided tables are statically owned (v1): "
-                           "peer IPC mappings pin the shard layout")
+        raise RuntimeError("one-sided shards move only through the "
+                           "collective migrate_blocks (IPC mappings pin "
+                           "the layout between remap generations)")
 
     def adopt_blocks(self, blocks) -> None:
-        raise RuntimeError("one-sided tables are statically owned (v1): "
-                           "peer IPC mappings pin the shard layout")
+        raise RuntimeError("one-sided shards move only through the "
+                           "collective migrate_blocks (IPC mappings pin "
+                           "the layout between remap generations)")
+
+    # ------------------------------------------------------- live migration
+
+    def _os_barrier(self, tag: str) -> None:
+        """Store-counter barrier (no collective plane on async jobs)."""
+        import time
+
+        key = f"os/{self.cfg.table_id}/bar/g{self._gen}/{tag}"
+        self.store.add(key, 1)
+        deadline = time.monotonic() + 120.0
+        while int(self.store.add(key, 0)) < self.world_size:
+            if time.monotonic() > deadline:
+                raise RuntimeError(f"one-sided barrier {tag} timed out")
+            time.sleep(0.0005)
+
+    def migrate_blocks(self, moves) -> None:
+        """Collective live migration (every rank, identical `moves` =
+        {block_id: dst_rank}, at a quiesced point — PlanExecutor calls it
+        between batches). Ownership-first protocol, one-sided data move:
+
+        1. quiesce — everyone fences issued pushes, owners drain rings;
+        2. adopters gather the moving blocks' rows DIRECTLY from the
+           current owner's HBM over xGMI (os_gather on the mapped peer
+           pointer; the owner does nothing);
+        3. remap — unmap all peers, flip ownership, rebuild the local
+           shard in a fresh IPC-exportable allocation (kept blocks copied
+           device-to-device, adopted blocks from step 2), reset rings,
+           re-export under the next generation and re-map.
+
+        Reference: MigrationExecutor.java:48 ownership-first migration;
+        the per-block access locks collapse to this phase-level quiesce."""
+        moves = {int(b): int(d) for b, d in dict(moves).items()
+                 if self.ownership.owner_of_int(int(b)) != int(d)}
+        if self.world_size == 1 or not moves:
+            self.ownership.update_many(moves)
+            return
+        assert self.store is not None, "migrate_blocks needs the store"
+        bs, vd = self.part.block_size, self.cfg.value_dim
+        isint = 1 if self.cfg.dtype == "int32" else 0
+        # 1. quiesce: all pushes issued + visible, all rings applied
+        self.fence()
+        self._os_barrier("q0")
+        self.drain()
+        torch.cuda.synchronize()
+        self._os_barrier("q1")
+        # 2. one-sided reads of adopted blocks from the (pre-flip) owner
+        adopted = {}
+        for b, dst in moves.items():
+            if dst != self.rank:
+                continue
+            src = self.ownership.owner_of_int(b)
+            rows = (int(self._peer_slot[src][b]) * bs
+                    + torch.arange(bs, device=self.device))
+            ptr = (self.shard.data_ptr() if src == self.rank
+                   else self._peer_ptr[src])
+            adopted[b] = self._hip.os_gather(ptr, rows, vd, isint)
+            self.stats["remote_bytes"] += bs * vd * 4
+        torch.cuda.synchronize()
+        self._os_barrier("read")         # owners keep shards until here
+        # 3. remap: close peer mappings, flip ownership, rebuild shard
+        self.close()
+        self._os_barrier("closed")
+        self.ownership.update_many(moves)
+        self._owner_dev = None
+        old_shard = self.shard
+        old_slot = {b: i for i, b in enumerate(self._local_blocks)}
+        new_owned = list(self.ownership.owned_blocks(self.rank))
+        new_shard = self._hip.os_shard_alloc(len(new_owned) * bs, vd, isint)
+        for i, b in enumerate(new_owned):
+            dstv = new_shard[i * bs:(i + 1) * bs]
+            if b in old_slot:
+                j = old_slot[b]
+                dstv.copy_(old_shard[j * bs:(j + 1) * bs])
+            else:
+                dstv.copy_(adopted[b])
+        self.shard = new_shard
+        self._local_blocks = new_owned
+        slot = torch.full((self.cfg.num_blocks,), -1, dtype=torch.int64)
+        for i, b in enumerate(new_owned):
+            slot[b] = i
+        self._block_slot = slot.to(self.device)
+        for r in range(self.world_size):
+            s = torch.full((self.cfg.num_blocks,), -1, dtype=torch.int64)
+            for i, b in enumerate(self.ownership.owned_blocks(r)):
+                s[b] = i
+            self._peer_slot[r] = s.to(self.device)
+        if self._ring_mode:
+            # rings are empty (drained at the quiesce); restart counters
+            self._ring_buf.zero_()
+            self._pushed = {r: 0 for r in range(self.world_size)}
+            self._head_cache = {r: 0 for r in range(self.world_size)}
+        torch.cuda.synchronize()
+        del old_shard, adopted           # freed only after "closed" barrier
+        self._gen += 1
+        self.connect()                   # re-export + re-map at gen+1

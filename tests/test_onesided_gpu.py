@@ -456,3 +456,100 @@ def _ring_big_worker(rank, world):
 def test_onesided_ring_oversized_push_no_deadlock():
     assert all(run_dist(_ring_big_worker, world=2, timeout=300))
 
+
+
+def _migrate_worker(rank, world):
+    """Live migration of a one-sided table (v3 migrate_blocks): values
+    survive the remap exactly, ownership flips on every rank, and pushes
+    keep conserving afterwards — for both the atomic (add) and the
+    apply-queue-ring (nmf_sgd) planes."""
+    import time
+
+    from harmony_amd.config import TableConfig
+    from harmony_amd.et.onesided import OneSidedTable
+
+    torch.cuda.set_device(0)
+    store = _store(rank, world)
+    dev = torch.device("cuda")
+
+    # ---- add-mode table
+    cfg = TableConfig(table_id="os_miga", num_keys=64, value_dim=4,
+                      num_blocks=8, update_fn="add", init_fn="zeros")
+    t = OneSidedTable(cfg, rank, world, dev, store=store)
+    _barrier(store, "a_alloc", rank, world)
+    t.connect()
+    _barrier(store, "a_conn", rank, world)
+    keys = torch.arange(64, device=dev)
+    ones = torch.ones(64, 4, device=dev)
+    t.push(keys, ones)
+    t.fence()
+    _barrier(store, "a_p1", rank, world)
+    moves = {0: 1, 1: 1, 7: 0}        # 0,1 leave rank 0; 7 leaves rank 1
+    t.migrate_blocks(moves)
+    assert t.ownership.owner_of_int(0) == 1
+    assert t.ownership.owner_of_int(7) == 0
+    # both ranks' pre-migration pushes survived the remap exactly
+    full = t.pull_full()
+    torch.cuda.synchronize()
+    assert bool((full == 2.0).all()), full.unique().tolist()
+    # post-migration pushes route by the NEW ownership and conserve
+    t.push(keys, ones)
+    t.fence()
+    _barrier(store, "a_p2", rank, world)
+    full = t.pull_full()
+    torch.cuda.synchronize()
+    assert bool((full == 4.0).all()), full.unique().tolist()
+    # moved blocks live in the adopter's local shard now
+    if rank == 1:
+        assert 0 in t.owned_blocks and 1 in t.owned_blocks
+    else:
+        assert 7 in t.owned_blocks and 0 not in t.owned_blocks
+    _barrier(store, "a_ok", rank, world)
+    t.close()
+
+    # ---- ring-mode table (arbitrary update fn through apply queues)
+    cfgr = TableConfig(table_id="os_migr", num_keys=32, value_dim=8,
+                       num_blocks=8, update_fn="nmf_sgd", init_fn="zeros",
+                       update_args={"step_size": 0.1, "max_val": 100.0})
+    tr = OneSidedTable(cfgr, rank, world, dev, store=store)
+    _barrier(store, "r_alloc", rank, world)
+    tr.connect()
+    _barrier(store, "r_conn", rank, world)
+    rkeys = torch.arange(32, device=dev)
+    d = torch.full((32, 8), -1.0, device=dev)   # value += 0.1 per push
+    tr.push(rkeys, d)                 # remote halves queue in rings
+    tr.fence()
+    store.add("r_pushed", 1)
+    while int(store.add("r_pushed", 0)) < world:
+        tr.drain()
+        time.sleep(0.002)
+    # migrate_blocks drains the queues itself at the quiesce point
+    tr.migrate_blocks({2: 1 - (2 % world), 5: 0})
+    full = tr.pull_full()
+    torch.cuda.synchronize()
+    assert torch.allclose(full, torch.full_like(full, 0.1 * world),
+                          atol=1e-5), full.unique().tolist()
+    # ring counters restarted: a fresh round of pushes still applies
+    tr.push(rkeys, d)
+    tr.fence()
+    store.add("r_pushed2", 1)
+    while int(store.add("r_pushed2", 0)) < world:
+        tr.drain()
+        time.sleep(0.002)
+    empties = 0
+    while empties < 3:
+        empties = empties + 1 if tr.drain() == 0 else 0
+        time.sleep(0.002)
+    _barrier(store, "r_done", rank, world)
+    full = tr.pull_full()
+    torch.cuda.synchronize()
+    assert torch.allclose(full, torch.full_like(full, 0.2 * world),
+                          atol=1e-5), full.unique().tolist()
+    _barrier(store, "r_ok", rank, world)
+    tr.close()
+    _finish(store, rank, world)
+    return True
+
+
+def test_onesided_live_migration():
+    assert all(run_dist(_migrate_worker, world=2, timeout=300))
