@@ -92,6 +92,13 @@ class OneSidedTable(Table):
         self.stats = {"pull_rows": 0, "push_rows": 0, "remote_bytes": 0}
         self._gen = 0                # IPC-mapping generation (bumped by
         #                              migrate_blocks' collective remap)
+        import threading
+
+        # serializes THIS process's async ops against migrate_blocks'
+        # unmap/remap window (e.g. CachedOneSidedAccessor's background
+        # refresh thread pulling while peers are closed). RLock: ring
+        # push's backpressure wait re-enters drain().
+        self._lk = threading.RLock()
         # every rank's block->slot map is derivable from the static
         # round-robin ownership, so remote row indices need no exchange
         self._peer_slot = {}
@@ -153,6 +160,10 @@ class OneSidedTable(Table):
                 self._ring_peer[r] = self._hip.os_ipc_open(hb)
 
     def close(self) -> None:
+        with self._lk:
+            self._close_locked()
+
+    def _close_locked(self) -> None:
         for p in self._peer_ptr.values():
             self._hip.os_ipc_close(p)
         self._peer_ptr.clear()
@@ -175,6 +186,10 @@ class OneSidedTable(Table):
 
     def pull(self, keys: torch.Tensor) -> torch.Tensor:
         """Async pull: gather rows straight out of each owner's HBM."""
+        with self._lk:
+            return self._pull_locked(keys)
+
+    def _pull_locked(self, keys: torch.Tensor) -> torch.Tensor:
         keys = keys.to(self.device, torch.int64)
         self.stats["pull_rows"] += keys.shape[0]
         owner = self._owner_of(self.part.block_of(keys))
@@ -202,6 +217,11 @@ class OneSidedTable(Table):
         """Async push. add-algebra fns: atomicAdd scatter into each
         owner's HBM (v1). Other fns: enqueue into each owner's apply-queue
         ring (v2) — the owner applies on its next drain()."""
+        with self._lk:
+            self._push_locked(keys, deltas)
+
+    def _push_locked(self, keys: torch.Tensor,
+                     deltas: torch.Tensor) -> None:
         keys = keys.to(self.device, torch.int64)
         self.stats["push_rows"] += keys.shape[0]
         deltas = deltas.to(self.device, self.dtype).contiguous()
@@ -312,6 +332,10 @@ class OneSidedTable(Table):
         batch boundary is the natural quiesce point). Returns items applied."""
         if not self._ring_mode:
             return 0
+        with self._lk:
+            return self._drain_locked(max_per)
+
+    def _drain_locked(self, max_per: int = 0) -> int:
         cap, vd = self._ring_cap, self.cfg.value_dim
         max_per = max_per or cap
         keys, deltas, counts = self._hip.os_ring_drain(
@@ -392,6 +416,10 @@ class OneSidedTable(Table):
 
         Reference: MigrationExecutor.java:48 ownership-first migration;
         the per-block access locks collapse to this phase-level quiesce."""
+        with self._lk:
+            self._migrate_locked(moves)
+
+    def _migrate_locked(self, moves) -> None:
         moves = {int(b): int(d) for b, d in dict(moves).items()
                  if self.ownership.owner_of_int(int(b)) != int(d)}
         if self.world_size == 1 or not moves:
