@@ -22,7 +22,11 @@ from harmony_amd.runtime.control import ControlPlane  # noqa: E402
 def build(app, ctx, cp, **app_args):
     job = JobConfig(job_id=f"qos_{app}", app=app, num_mini_batches=4,
                     num_worker_blocks=4, app_args=app_args)
-    return mlapps.get_app(app).build(job, ctx, cp)
+    tables, tr, pr = mlapps.get_app(app).build(job, ctx, cp)
+    if hasattr(tr, "initialize"):
+        tr.initialize()
+    torch.cuda.synchronize()
+    return tables, tr, pr
 
 
 def main():
@@ -44,14 +48,18 @@ def main():
 
     def run_mode(mlr_stream, lda_stream, label, secs=4.0):
         stop = threading.Event()
+        hammer_n = [0]
+        hammer_err = []
 
         def hammer():
-            i = 0
-            with torch.cuda.stream(lda_stream):
-                while not stop.is_set():
-                    step(lda_tr, lda_pr, i)
-                    i += 1
-                torch.cuda.synchronize()
+            try:
+                with torch.cuda.stream(lda_stream):
+                    while not stop.is_set():
+                        step(lda_tr, lda_pr, hammer_n[0])
+                        hammer_n[0] += 1
+                    torch.cuda.synchronize()
+            except Exception as e:              # noqa: BLE001
+                hammer_err.append(repr(e))
 
         lat = []
         t = threading.Thread(target=hammer)
@@ -69,18 +77,33 @@ def main():
         stop.set()
         t.join()
         torch.cuda.synchronize()
+        assert not hammer_err, f"co-tenant crashed: {hammer_err[0]}"
+        assert hammer_n[0] > 100, f"co-tenant barely ran: {hammer_n[0]}"
         lat.sort()
         n = len(lat)
         p = lambda q: lat[min(n - 1, int(q * n))] * 1e3  # noqa: E731
-        print(f"{label}: n={n} p50={p(.5):.3f} p95={p(.95):.3f} "
-              f"p99={p(.99):.3f} max={lat[-1]*1e3:.3f} ms")
+        print(f"{label}: n={n} lda_steps={hammer_n[0]} p50={p(.5):.3f} "
+              f"p95={p(.95):.3f} p99={p(.99):.3f} max={lat[-1]*1e3:.3f} ms")
 
-    s_sh_m, s_sh_l = torch.cuda.Stream(), torch.cuda.Stream()
-    run_mode(s_sh_m, s_sh_l, "shared   ")
+    # warm EVERYTHING untimed (Tensile kernel selection, alias builds,
+    # SCLK ramp) so mode ordering cannot masquerade as isolation
+    tw = time.monotonic()
+    i = 0
+    while time.monotonic() - tw < 2.0:
+        step(lda_tr, lda_pr, i)
+        step(mlr_tr, mlr_pr, i)
+        i += 1
+    torch.cuda.synchronize()
+
     from harmony_amd.utils.custreams import cu_partitioned_streams
 
     st = cu_partitioned_streams({"mlr": 64, "lda": 192})
+    s_sh_m, s_sh_l = torch.cuda.Stream(), torch.cuda.Stream()
+    # interleave the modes to cancel any residual drift
+    run_mode(s_sh_m, s_sh_l, "shared   ")
     run_mode(st["mlr"], st["lda"], "partition")
+    run_mode(s_sh_m, s_sh_l, "shared2  ")
+    run_mode(st["mlr"], st["lda"], "partition2")
 
 
 if __name__ == "__main__":
