@@ -23,7 +23,7 @@ from __future__ import annotations
 
 import threading
 import time
-from typing import Dict, Optional, Set
+from typing import Dict, Optional, Set, Tuple
 
 from harmony_amd.utils import sanitize
 
@@ -368,6 +368,10 @@ class TaskUnitScheduler:
             with self._cv:
                 while self._watermark in self._done:
                     self._done.discard(self._watermark)
+                    # seqs below the watermark are never consulted again:
+                    # prune the seq->job cache so week-long runs don't
+                    # accumulate one entry per global phase forever
+                    self._job_cache.pop(self._watermark, None)
                     self._watermark += 1
                 blockers = False
                 for s in range(self._watermark, seq):
