@@ -177,7 +177,12 @@ class PlanExecutor:
                     migrate(table, dict(op.moves), self.rank,
                             self.world_size, group=self.group)
             elif isinstance(op, SetBatchShareOp):
-                self.batch_shares = dict(op.shares)
+                # MERGE (not replace): a plan can stop rank r and set the
+                # remaining workers' shares — replacing here would silently
+                # wipe the StopWorkerOp's share-0 applied moments earlier
+                if self.batch_shares is None:
+                    self.batch_shares = {}
+                self.batch_shares.update(dict(op.shares))
             elif isinstance(op, StopWorkerOp):
                 if self.batch_shares is None:
                     self.batch_shares = {}

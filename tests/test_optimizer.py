@@ -258,3 +258,52 @@ def test_stop_start_worker_sheds_work():
     # stopped in the middle, working at both ends
     assert ex1[0] == 64 and ex1[-1] > 0, ex1
     assert len(ex0) == len(ex1)                    # same batch COUNT (collective)
+
+
+def _hetero_demote_worker(rank, world):
+    """HeterogeneousOptimizer with role selection drives the REAL runtime:
+    an artificially slow rank 1 is demoted to pure server (StopWorkerOp)
+    by the orchestrator mid-job, and the job completes."""
+    import time
+
+    from harmony_amd import mlapps
+    from harmony_amd.config import JobConfig, RuntimeConfig
+    from harmony_amd.dolphin.worker import WorkerTasklet
+    from harmony_amd.optimizer.hetero import HeterogeneousOptimizer
+    from harmony_amd.optimizer.orchestrator import OptimizationOrchestrator
+    from harmony_amd.runtime.bootstrap import init_executor
+    from harmony_amd.runtime.control import ControlPlane, TaskUnitScheduler
+
+    ctx = init_executor(RuntimeConfig(device="cpu"))
+    cp = ControlPlane(ctx.store, ctx.rank, ctx.world_size)
+    job = JobConfig(job_id="het_demote", app="addvector", max_num_epochs=4,
+                    num_mini_batches=4,
+                    app_args={"num_keys": 32, "vector_dim": 4})
+    tus = TaskUnitScheduler(cp, {job.job_id})
+    app = mlapps.get_app("addvector")
+    tables, trainer, provider = app.build(job, ctx, cp)
+    if rank == 1:                       # make rank 1 ~100x slower
+        orig = trainer.local_compute
+
+        def slow():
+            time.sleep(0.05)
+            orig()
+
+        trainer.local_compute = slow
+    orch = OptimizationOrchestrator(
+        cp, job.job_id, ctx.rank, ctx.world_size, tables,
+        optimizer=HeterogeneousOptimizer(benefit_threshold=0.05),
+        check_period=3)
+    t = WorkerTasklet(job, trainer, provider, cp, tus, ctx.rank,
+                      ctx.world_size, orchestrator=orch)
+    t.run()
+    shares = orch.executor.batch_shares or {}
+    return (orch.applied_plans, shares.get(1))
+
+
+def test_hetero_role_selection_demotes_slow_rank():
+    res = run_dist(_hetero_demote_worker, world=2, timeout=240)
+    assert res[0] == res[1]              # collective agreement
+    applied, share1 = res[0]
+    assert applied > 0
+    assert share1 == 0                   # slow rank stopped (pure server)
