@@ -265,11 +265,13 @@ class DataPlane:
         else:
             # piggyback varies per call -> bypass the static-counts cache
             W = self.world_size
-            ext = torch.cat([send_counts.to(torch.int64),
-                             piggyback.to(torch.int64)])
             dev = (self.device if self.backend == "nccl"
                    else torch.device("cpu"))
-            ext = ext.to(dev)
+            # counts ride on the keys' device, the piggyback on the
+            # caller's (often CPU): unify BEFORE cat — mixed-device cat
+            # only crashed on the RCCL path (tests/test_comm_nccl1.py)
+            ext = torch.cat([send_counts.to(dev, torch.int64),
+                             piggyback.to(dev, torch.int64)])
             bufs = [torch.empty_like(ext) for _ in range(W)]
             dist.all_gather(bufs, ext, group=self.group)
             mat = torch.stack(bufs).cpu()
