@@ -99,6 +99,21 @@ class OneSidedTable(Table):
         # refresh thread pulling while peers are closed). RLock: ring
         # push's backpressure wait re-enters drain().
         self._lk = threading.RLock()
+        # jobserver fail-fast flag (runtime/control.py ControlPlane): ring
+        # backpressure and migration barriers poll it so a dead peer
+        # unwinds this rank instead of wedging it for the full timeout
+        self._failed_key = "js/failed"
+
+    def _check_failed(self) -> None:
+        from harmony_amd.runtime.control import JobCancelled
+
+        try:
+            failed = self.store is not None \
+                and self.store.check([self._failed_key])
+        except Exception:            # noqa: BLE001 — store gone == failed
+            failed = True
+        if failed:
+            raise JobCancelled("one-sided wait: jobserver failed fast")
         # every rank's block->slot map is derivable from the static
         # round-robin ownership, so remote row indices need no exchange
         self._peer_slot = {}
@@ -301,7 +316,11 @@ class OneSidedTable(Table):
             return
         # exact host-side backpressure: we are the only writer of our slot
         deadline = time.monotonic() + 30.0
+        it = 0
         while self._pushed[r] + n - self._head_cache[r] > cap:
+            it += 1
+            if it % 100 == 0:
+                self._check_failed()
             # refresh the owner's head (remote read), bounded wait
             self._hip.os_ring_read_head(self._ring_peer[r], self.world_size,
                                         cap, vd, self.rank, self._scratch)
@@ -395,7 +414,11 @@ class OneSidedTable(Table):
         key = f"os/{self.cfg.table_id}/bar/g{self._gen}/{tag}"
         self.store.add(key, 1)
         deadline = time.monotonic() + 120.0
+        it = 0
         while int(self.store.add(key, 0)) < self.world_size:
+            it += 1
+            if it % 200 == 0:
+                self._check_failed()
             if time.monotonic() > deadline:
                 raise RuntimeError(f"one-sided barrier {tag} timed out")
             time.sleep(0.0005)
