@@ -103,6 +103,14 @@ class OneSidedTable(Table):
         # backpressure and migration barriers poll it so a dead peer
         # unwinds this rank instead of wedging it for the full timeout
         self._failed_key = "js/failed"
+        # every rank's block->slot map is derivable from the static
+        # round-robin ownership, so remote row indices need no exchange
+        self._peer_slot = {}
+        for r in range(world_size):
+            slot = torch.full((cfg.num_blocks,), -1, dtype=torch.int64)
+            for i, b in enumerate(self.ownership.owned_blocks(r)):
+                slot[b] = i
+            self._peer_slot[r] = slot.to(device)
 
     def _check_failed(self) -> None:
         from harmony_amd.runtime.control import JobCancelled
@@ -114,14 +122,6 @@ class OneSidedTable(Table):
             failed = True
         if failed:
             raise JobCancelled("one-sided wait: jobserver failed fast")
-        # every rank's block->slot map is derivable from the static
-        # round-robin ownership, so remote row indices need no exchange
-        self._peer_slot = {}
-        for r in range(world_size):
-            slot = torch.full((cfg.num_blocks,), -1, dtype=torch.int64)
-            for i, b in enumerate(self.ownership.owned_blocks(r)):
-                slot[b] = i
-            self._peer_slot[r] = slot.to(device)
 
     # ------------------------------------------------------------- lifecycle
 
