@@ -107,10 +107,14 @@ def run_job(job: JobConfig, ctx: ExecutorContext,
     if optimizer is not None:
         from harmony_amd.optimizer.orchestrator import OptimizationOrchestrator
 
+        from harmony_amd.dolphin.model_accessor import OneSidedAccessor
+
         orch = OptimizationOrchestrator(
             cp, job.job_id, ctx.rank, ctx.world_size, tables,
             optimizer=optimizer, check_period=job.optimizer_period,
-            group=getattr(ctx, "group", None))
+            group=getattr(ctx, "group", None),
+            async_plane=isinstance(getattr(trainer, "accessor", None),
+                                   OneSidedAccessor))
     tracer = None
     if job.trace_path:
         from harmony_amd.utils.tracing import Tracer

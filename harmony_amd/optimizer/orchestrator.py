@@ -32,7 +32,7 @@ class OptimizationOrchestrator:
                  world_size: int, tables: Dict[str, object],
                  optimizer: Optional[Optimizer] = None,
                  check_period: int = 8, min_metrics: int = 1,
-                 ema: float = 0.5, group=None):
+                 ema: float = 0.5, group=None, async_plane: bool = False):
         self.cp = cp
         self.job_id = job_id
         self.rank = rank
@@ -42,6 +42,15 @@ class OptimizationOrchestrator:
         self.period = max(1, check_period)
         self.min_metrics = min_metrics
         self.ema_w = ema
+        # async_plane (one-sided jobs): ranks are NOT lockstepped, so a
+        # batch-index-aligned apply point can be passed by a fast rank
+        # before rank 0 publishes (observed: one rank alone in the
+        # migration barrier). Plans go through a sequential queue instead:
+        # every rank applies plan #n at its NEXT boundary after seeing it —
+        # same plans, same order, possibly different boundaries (safe: the
+        # async plane has no collectives; migration syncs via the store).
+        self.async_plane = async_plane
+        self._seq_applied = 0
         self.executor = PlanExecutor(tables, rank, world_size, group=group)
         self._ema: Dict[int, RankMetrics] = {}
         self._batches = 0
@@ -101,19 +110,53 @@ class OptimizationOrchestrator:
         if self._batches % self.period != 0:
             return None
         at = self._batches
-        if self.rank == 0 and self.optimizer is not None and at > self.period:
+        if self.async_plane:
+            # async ranks skew freely: "at - period" may not be reported
+            # by a lagging rank yet (observed: every window skipped).
+            # Decide on the LATEST window every rank has fully reported.
+            if self.rank == 0 and self.optimizer is not None:
+                t = self._latest_complete_window(at)
+                if t is not None:
+                    self._dec_win = t
+                    self._decide(t, plan_key_base="")
+        elif self.rank == 0 and self.optimizer is not None \
+                and at > self.period:
             # decide on the PREVIOUS window's metrics: co-located ranks are
             # within one batch of each other, so the current window's
             # reports race rank 0's check (observed: perpetual skips when
             # jobs co-run); the previous window is guaranteed complete
             self._decide(at - self.period,
                          plan_key_base=f"opt/{self.job_id}/plan")
+        if self.async_plane:
+            # sequential queue: apply the next unseen plan, if any
+            n_pub = self.cp.read(f"opt/{self.job_id}/planseq_n")
+            if n_pub > self._seq_applied:
+                key = f"opt/{self.job_id}/planseq/{self._seq_applied}"
+                plan = Plan.from_json(self.cp.store.get(key).decode())
+                self._seq_applied += 1
+                if not plan.empty():
+                    return plan
+            return None
         my_key = f"opt/{self.job_id}/plan/{at}"
         if self.cp.flag_set(my_key):
             plan = Plan.from_json(self.cp.store.get(my_key).decode())
             if not plan.empty():
                 return plan
         return None
+
+    def _latest_complete_window(self, upto: int):
+        """Highest period-multiple <= upto for which EVERY rank's metric
+        window exists (async plane: ranks report at their own pace)."""
+        best = None
+        t = getattr(self, "_dec_win", 0) + self.period
+        while t <= upto:
+            if all(self.cp.flag_set(f"opt/{self.job_id}/m/{r}/{t}")
+                   for r in range(self.world_size)):
+                best = t
+                t += self.period
+            else:
+                break
+        return best
 
     def apply(self, plan: Plan) -> None:
         """Collective: execute the plan (inside the caller's NET ticket)."""
@@ -131,6 +174,15 @@ class OptimizationOrchestrator:
                   for t in self.tables.values() if hasattr(t, "cfg")}
         plan = self.optimizer.optimize(metrics, owners, self.world_size)
         if plan is not None and not plan.empty():
+            if self.async_plane:
+                # sequential queue: value first, THEN the count (readers
+                # poll the count and fetch by index — same store client,
+                # server applies in order)
+                n = self.cp.read(f"opt/{self.job_id}/planseq_n")
+                self.cp.store.set(f"opt/{self.job_id}/planseq/{n}",
+                                  plan.to_json())
+                self.cp.incr(f"opt/{self.job_id}/planseq_n", 1)
+                return
             # applies two windows after `at` (one after the caller's
             # current boundary — see boundary_plan's at-period shift)
             self.cp.store.set(f"{plan_key_base}/{at + 2 * self.period}",

@@ -553,3 +553,34 @@ def _migrate_worker(rank, world):
 
 def test_onesided_live_migration():
     assert all(run_dist(_migrate_worker, world=2, timeout=300))
+
+
+def _async_migrate_job_worker(rank, world):
+    """Full-stack v3: an orchestrator plan (MoveOp) migrates a ONE-SIDED
+    table live inside a real async job — PlanExecutor -> et/migration
+    delegation -> OneSidedTable.migrate_blocks (quiesce, one-sided block
+    reads, generation remap) — and training converges through the remap."""
+    from harmony_amd.config import JobConfig, RuntimeConfig
+    from harmony_amd.dolphin.master import run_job
+    from harmony_amd.optimizer.optimizers import SampleOptimizers
+    from harmony_amd.runtime.bootstrap import init_executor
+
+    ctx = init_executor(RuntimeConfig(device="cuda", backend="gloo"))
+    job = JobConfig(job_id="os_mlr_mig", app="mlr", max_num_epochs=6,
+                    num_mini_batches=2, optimizer_period=2,
+                    app_args={"num_classes": 4, "num_features": 32,
+                              "num_parts_per_class": 4, "batch_size": 512,
+                              "step_size": 0.3, "one_sided": True})
+    opt = SampleOptimizers.rotate_blocks("os_mlr_mig/mlr_model", stride=2)
+    m = run_job(job, ctx, optimizer=opt)
+    s = m.summary()
+    return (s["num_batches"], s.get("accuracy"),
+            getattr(m, "_applied_plans", 0))
+
+
+def test_async_job_live_migration():
+    res = run_dist(_async_migrate_job_worker, world=2, timeout=300)
+    for n, acc, plans in res:
+        assert n == 12, res
+        assert acc is not None and acc > 0.5, res
+        assert plans >= 1, res
