@@ -183,6 +183,13 @@ class WorkerTasklet:
                         "data_processing_rate":
                             ep_examples / ep_dt if ep_dt else 0.0,
                         "epoch_time_sec": ep_dt}, t=time.time())
+            # async plan queue: drain plans published while this rank was
+            # finishing (a straggler plan applied by only SOME ranks would
+            # leave them alone in the migration barrier — observed flake)
+            if self.orch is not None:
+                for plan in self.orch.finalize_async():
+                    self.orch.apply(plan)
+                    self._consume_shares()
             # RUN -> CLEANUP barrier
             self.cp.barrier(f"{jid}/cleanup", self.world_size)
             self.trainer.cleanup()

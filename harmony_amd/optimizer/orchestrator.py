@@ -144,6 +144,35 @@ class OptimizationOrchestrator:
                 return plan
         return None
 
+    def finalize_async(self):
+        """End-of-job plan-queue drain (async plane; called by every rank
+        BEFORE the cleanup barrier). Rank 0 freezes the final plan count;
+        every rank then applies any published-but-unapplied plans so the
+        whole world has executed the identical plan sequence. Returns the
+        remaining plans to apply, in order. No-op on the collective plane."""
+        if not self.async_plane:
+            return []
+        import time
+
+        fin_key = f"opt/{self.job_id}/planseq_final"
+        if self.rank == 0:
+            # freeze: no _decide runs after this (rank 0 is past its last
+            # boundary when it gets here)
+            self.cp.store.set(fin_key,
+                              str(self.cp.read(f"opt/{self.job_id}/planseq_n")))
+        while not self.cp.flag_set(fin_key):
+            self.cp.check_failed()
+            time.sleep(0.0005)
+        n_final = int(self.cp.store.get(fin_key))
+        out = []
+        while self._seq_applied < n_final:
+            key = f"opt/{self.job_id}/planseq/{self._seq_applied}"
+            plan = Plan.from_json(self.cp.store.get(key).decode())
+            self._seq_applied += 1
+            if not plan.empty():
+                out.append(plan)
+        return out
+
     def _latest_complete_window(self, upto: int):
         """Highest period-multiple <= upto for which EVERY rank's metric
         window exists (async plane: ranks report at their own pace)."""
