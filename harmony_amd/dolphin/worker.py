@@ -185,11 +185,26 @@ class WorkerTasklet:
                         "epoch_time_sec": ep_dt}, t=time.time())
             # async plan queue: drain plans published while this rank was
             # finishing (a straggler plan applied by only SOME ranks would
-            # leave them alone in the migration barrier — observed flake)
-            if self.orch is not None:
-                for plan in self.orch.finalize_async():
-                    self.orch.apply(plan)
-                    self._consume_shares()
+            # leave them alone in the migration barrier — observed flake).
+            # Drain INCREMENTALLY while polling the freeze key: a peer may
+            # be blocked inside plan k's migration barrier at its own
+            # boundary, so plan k must be applied here BEFORE any
+            # wait-for-freeze (waiting first deadlocked — observed).
+            if self.orch is not None and self.orch.async_plane:
+                self.orch.finalize_mark()
+                it = 0
+                while True:
+                    plan = self.orch.next_pending_plan()
+                    if plan is not None:
+                        self.orch.apply(plan)
+                        self._consume_shares()
+                        continue
+                    if self.orch.async_drained():
+                        break
+                    it += 1
+                    if it % 200 == 0:
+                        self.cp.check_failed()
+                    time.sleep(0.0005)
             # RUN -> CLEANUP barrier
             self.cp.barrier(f"{jid}/cleanup", self.world_size)
             self.trainer.cleanup()

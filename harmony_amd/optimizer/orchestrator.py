@@ -144,34 +144,33 @@ class OptimizationOrchestrator:
                 return plan
         return None
 
-    def finalize_async(self):
-        """End-of-job plan-queue drain (async plane; called by every rank
-        BEFORE the cleanup barrier). Rank 0 freezes the final plan count;
-        every rank then applies any published-but-unapplied plans so the
-        whole world has executed the identical plan sequence. Returns the
-        remaining plans to apply, in order. No-op on the collective plane."""
-        if not self.async_plane:
-            return []
-        import time
-
-        fin_key = f"opt/{self.job_id}/planseq_final"
-        if self.rank == 0:
-            # freeze: no _decide runs after this (rank 0 is past its last
-            # boundary when it gets here)
-            self.cp.store.set(fin_key,
-                              str(self.cp.read(f"opt/{self.job_id}/planseq_n")))
-        while not self.cp.flag_set(fin_key):
-            self.cp.check_failed()
-            time.sleep(0.0005)
-        n_final = int(self.cp.store.get(fin_key))
-        out = []
-        while self._seq_applied < n_final:
+    def next_pending_plan(self):
+        """Async end-of-job drain step: the next published-but-unapplied
+        plan, or None. Must be callable while a PEER is blocked inside a
+        plan's migration barrier — so it never waits on the freeze key."""
+        n_pub = self.cp.read(f"opt/{self.job_id}/planseq_n")
+        while self._seq_applied < n_pub:
             key = f"opt/{self.job_id}/planseq/{self._seq_applied}"
             plan = Plan.from_json(self.cp.store.get(key).decode())
             self._seq_applied += 1
             if not plan.empty():
-                out.append(plan)
-        return out
+                return plan
+        return None
+
+    def finalize_mark(self) -> None:
+        """Rank 0, on entering the end-of-job drain (past its last
+        boundary, so no further decisions): freeze the final plan count."""
+        if self.async_plane and self.rank == 0:
+            self.cp.store.set(f"opt/{self.job_id}/planseq_final",
+                              str(self.cp.read(f"opt/{self.job_id}/planseq_n")))
+
+    def async_drained(self) -> bool:
+        """True once the freeze is published and every frozen plan has
+        been applied by THIS rank."""
+        fin_key = f"opt/{self.job_id}/planseq_final"
+        if not self.cp.flag_set(fin_key):
+            return False
+        return self._seq_applied >= int(self.cp.store.get(fin_key))
 
     def _latest_complete_window(self, upto: int):
         """Highest period-multiple <= upto for which EVERY rank's metric
