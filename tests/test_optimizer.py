@@ -307,3 +307,63 @@ def test_hetero_role_selection_demotes_slow_rank():
     applied, share1 = res[0]
     assert applied > 0
     assert share1 == 0                   # slow rank stopped (pure server)
+
+
+def test_async_plan_queue_skew_and_drain():
+    """Async-plane plan delivery (sequential store queue): ranks at
+    unbounded skew apply the IDENTICAL plan sequence, including plans
+    published after a fast rank passed its last boundary (end-of-job
+    drain protocol: finalize_mark + next_pending_plan + async_drained)."""
+    import torch
+
+    from torch.distributed import TCPStore
+
+    from harmony_amd.config import TableConfig
+    from harmony_amd.et.table import Table
+    from harmony_amd.optimizer.optimizers import SampleOptimizers
+    from harmony_amd.optimizer.orchestrator import OptimizationOrchestrator
+    from harmony_amd.runtime.control import ControlPlane
+
+    store = TCPStore("127.0.0.1", 29781, 2, is_master=True,
+                     wait_for_workers=False)
+
+    def mk(rank):
+        cfg = TableConfig(table_id="qt", num_keys=32, value_dim=4,
+                          num_blocks=8, update_fn="add", init_fn="zeros")
+        t = Table(cfg, rank, 2, torch.device("cpu"))
+        cp = ControlPlane(store, rank, 2)
+        return OptimizationOrchestrator(
+            cp, "qdrain", rank, 2, {"qt": t},
+            optimizer=SampleOptimizers.rotate_blocks("qt", 2),
+            check_period=2, async_plane=True)
+
+    o0, o1 = mk(0), mk(1)
+    applied = [0, 0]
+
+    def boundary(o, i):
+        o.report_batch(0.01, 0.005, 0.001, 0.001, 512)
+        p = o.boundary_plan()
+        if p is not None:
+            applied[i] += 1
+
+    # rank 1 races through ALL 12 batches before rank 0 starts (unbounded
+    # skew: the failure mode of batch-index-aligned application)
+    for _ in range(12):
+        boundary(o1, 1)
+    for _ in range(12):
+        boundary(o0, 0)
+    # end-of-job drain: rank 0 (decider) marks; both drain
+    o0.finalize_mark()
+    for i, o in ((0, o0), (1, o1)):
+        while True:
+            p = o.next_pending_plan()
+            if p is not None:
+                applied[i] += 1
+                continue
+            if o.async_drained():
+                break
+    # both ranks applied every published plan exactly once, same order
+    n_pub = int(store.add("opt/qdrain/planseq_n", 0))
+    assert n_pub >= 1
+    assert applied[0] == applied[1] == n_pub, (applied, n_pub)
+    assert o0._seq_applied == o1._seq_applied == n_pub
