@@ -319,7 +319,10 @@ def main():
     if args.mode == "runtime":
         run_runtime_mode(args, ctx, cp, jobs)
         return
-    multi = len(jobs) > 1
+    # tickets order CROSS-RANK collectives; at world 1 there are none, so
+    # a torchrun-launched N=1 (TCPStore present) must not pay ~0.1 ms/phase
+    # of store round-trips for nothing
+    multi = len(jobs) > 1 and world > 1
     tus = TaskUnitScheduler(cp, {j.job_id for j in jobs.values()},
                             multi_job=multi)
     for j in jobs.values():
