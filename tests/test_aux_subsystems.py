@@ -579,3 +579,35 @@ def test_cached_one_sided_accessor_background_refresh():
         assert float(acc.pull_all()[0, 0]) >= 6.0
     finally:
         acc.close()
+
+
+def test_onesided_check_failed_unit():
+    """OneSidedTable._check_failed raises JobCancelled when the jobserver
+    failed-fast flag is set or the store is gone (ring backpressure and
+    migration barriers poll it so a dead peer can't wedge a writer)."""
+    import pytest
+
+    from harmony_amd.et.onesided import OneSidedTable
+    from harmony_amd.runtime.control import JobCancelled
+
+    class FakeStore:
+        def __init__(self):
+            self.flag = False
+            self.dead = False
+
+        def check(self, keys):
+            if self.dead:
+                raise RuntimeError("connection reset")
+            return self.flag
+
+    t = OneSidedTable.__new__(OneSidedTable)   # method under test only
+    t.store = FakeStore()
+    t._failed_key = "js/failed"
+    t._check_failed()                           # healthy: no raise
+    t.store.flag = True
+    with pytest.raises(JobCancelled):
+        t._check_failed()
+    t.store.flag = False
+    t.store.dead = True                         # store gone == failed
+    with pytest.raises(JobCancelled):
+        t._check_failed()
