@@ -96,3 +96,34 @@ def test_gbt_codec_roundtrip(depth, n, seed):
         assert t0.threshold == t1.threshold
         assert torch.allclose(torch.tensor(t0.leaf_value),
                               torch.tensor(t1.leaf_value))
+
+
+@settings(max_examples=40, deadline=None)
+@given(world=st.integers(2, 8), blocks=st.integers(2, 64),
+       seed=st.integers(0, 2**31 - 1))
+def test_rebalance_moves_reach_targets(world, blocks, seed):
+    """rebalance_moves (the reference's priority-queue TransferStep
+    pairing): applying the produced moves lands exactly on the target
+    counts, and only genuinely-moving blocks appear."""
+    import torch
+
+    from harmony_amd.config import TableConfig
+    from harmony_amd.et.migration import rebalance_moves
+    from harmony_amd.et.table import Table
+
+    g = torch.Generator().manual_seed(seed)
+    cfg = TableConfig(table_id="rb", num_keys=blocks * 4, value_dim=2,
+                      num_blocks=blocks, update_fn="add", init_fn="zeros")
+    t = Table(cfg, 0, world, torch.device("cpu"))
+    # random-but-valid target: a permutation of the current counts
+    cur = [len(t.ownership.owned_blocks(r)) for r in range(world)]
+    perm = torch.randperm(world, generator=g).tolist()
+    target = [cur[p] for p in perm]
+    moves = rebalance_moves(t, target)
+    after = list(cur)
+    for b, dst in moves.items():
+        src = t.ownership.owner_of_int(b)
+        assert src != dst            # no vacuous moves
+        after[src] -= 1
+        after[dst] += 1
+    assert after == target, (cur, target, moves)
