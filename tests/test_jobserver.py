@@ -332,3 +332,20 @@ def test_standalone_launcher_cli():
     assert out.returncode == 0, out.stderr[-1500:]
     s = json.loads(out.stdout.strip().splitlines()[-1])
     assert s["num_batches"] == 4 and "sq_err" in s
+
+
+def test_standalone_launcher_pregel_cli():
+    """Standalone mode also runs Pregel apps (reference graphapps mains):
+    -app pagerank goes through run_pregel_job with a single-rank view."""
+    import subprocess
+    import sys as _sys
+
+    out = subprocess.run(
+        [_sys.executable, "-m", "harmony_amd.standalone", "-app", "pagerank",
+         "-device", "cpu", "-num_vertices", "200", "-out_degree", "4",
+         "-num_iters", "5"],
+        capture_output=True, text=True, timeout=300)
+    assert out.returncode == 0, out.stderr[-1500:]
+    s = json.loads(out.stdout.strip().splitlines()[-1])
+    assert s.get("supersteps", 0) >= 1
+    assert s.get("num_local_vertices") == 200
