@@ -99,6 +99,21 @@ class WorkerTasklet:
         stream_ctx = (torch.cuda.stream(self.stream) if self.stream is not None
                       else contextlib.nullcontext())
         with stream_ctx:
+            # per-epoch batch COUNTS must be identical across ranks (each
+            # batch is a set of collectives on the collective plane, and
+            # the SSP done-counters assume uniform counts on both planes).
+            # An uneven file split can produce unequal block counts — fail
+            # LOUDLY on every rank instead of hanging in a collective.
+            if self.world_size > 1 and not self.is_async:
+                nb = self.provider.num_batches
+                mx = self.cp.agree_max(f"{jid}/nbmax", nb, self.world_size)
+                mn = -self.cp.agree_max(f"{jid}/nbmin", -nb, self.world_size)
+                if mx != mn:
+                    raise RuntimeError(
+                        f"rank {self.rank}: per-epoch batch counts differ "
+                        f"across ranks ({mn}..{mx}; mine {nb}) — the input "
+                        "split produced unequal block counts; rebalance the "
+                        "input or lower num_worker_blocks")
             # initialize() may issue collectives (e.g. LDA's initial count
             # push) — serialize it like any NET phase.
             with self._net(jid):

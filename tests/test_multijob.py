@@ -125,3 +125,36 @@ def test_three_concurrent_jobs_three_ranks_staggered():
     res = run_dist(_three_jobs_worker, world=3, timeout=300)
     for r in res:
         assert r == {"s3_mlr": 9, "s3_nmf": 8, "s3_lda": 4}
+
+
+def _uneven_blocks_worker(rank, world):
+    """Unequal per-rank batch counts (uneven input split) must fail
+    LOUDLY on every rank instead of hanging in a collective."""
+    from harmony_amd import mlapps
+    from harmony_amd.config import JobConfig, RuntimeConfig
+    from harmony_amd.dolphin.worker import WorkerTasklet
+    from harmony_amd.runtime.bootstrap import init_executor
+    from harmony_amd.runtime.control import ControlPlane, TaskUnitScheduler
+
+    ctx = init_executor(RuntimeConfig(device="cpu"))
+    cp = ControlPlane(ctx.store, ctx.rank, ctx.world_size)
+    job = JobConfig(job_id="uneven", app="addvector", max_num_epochs=2,
+                    num_mini_batches=4,
+                    app_args={"num_keys": 32, "vector_dim": 4})
+    tables, trainer, provider = mlapps.get_app("addvector").build(
+        job, ctx, cp)
+    if rank == 1:
+        provider.blocks = provider.blocks[:-1]     # simulate uneven split
+    t = WorkerTasklet(job, trainer, provider, cp,
+                      TaskUnitScheduler(cp, {job.job_id}), ctx.rank,
+                      ctx.world_size)
+    try:
+        t.run()
+        return "no-error"
+    except RuntimeError as e:
+        return "batch counts differ" in str(e)
+
+
+def test_uneven_batch_counts_fail_loudly():
+    assert all(r is True for r in
+               run_dist(_uneven_blocks_worker, world=2, timeout=120))
