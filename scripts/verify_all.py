@@ -6,7 +6,11 @@ GPU box (gpurun):   python scripts/verify_all.py --gpu
 
 Runs, in order: build (hipcc cross-compile + import), the CPU test suite,
 and with --gpu additionally the GPU suite, smoke(), and a short bench with
-a JSON-contract check. Exits nonzero on the first failure."""
+a JSON-contract check. Exits nonzero on the first failure.
+
+Budget note: each step is a separate python process, and the FIRST torch
+import on a fresh box can take 1-2 min while the image pages in — budget
+>=10 min of box time for --gpu (a 3.5-min clamp killed the first run)."""
 import argparse
 import json
 import subprocess
