@@ -127,3 +127,38 @@ def test_rebalance_moves_reach_targets(world, blocks, seed):
         after[src] -= 1
         after[dst] += 1
     assert after == target, (cur, target, moves)
+
+
+@settings(max_examples=40, deadline=None)
+@given(blocks=st.integers(1, 128), world=st.integers(1, 9),
+       seed=st.integers(0, 2**31 - 1))
+def test_ownership_invariants(blocks, world, seed):
+    """Ownership map invariants (reference BlockManager even split +
+    collective update): initial partition is contiguous/even within 1,
+    every block always has exactly one owner, counts are conserved, the
+    version advances on every mutation, and slot maps are dense."""
+    import torch
+
+    from harmony_amd.et.ownership import Ownership
+
+    o = Ownership(blocks, world)
+    cnt = o.counts()
+    assert sum(cnt) == blocks
+    assert max(cnt) - min(cnt) <= 1           # even within one block
+    # contiguous: owner tensor is non-decreasing
+    assert bool((o.owner[1:] >= o.owner[:-1]).all())
+    g = torch.Generator().manual_seed(seed)
+    v0 = o.version
+    moves = {int(torch.randint(0, blocks, (1,), generator=g)):
+             int(torch.randint(0, world, (1,), generator=g))
+             for _ in range(min(8, blocks))}
+    o.update_many(moves)
+    assert o.version == v0 + 1
+    for b, r in moves.items():
+        assert o.owner_of_int(b) == r          # last write wins per key
+    assert sum(o.counts()) == blocks           # conservation
+    for r in range(world):
+        slots = o.slot_of(r)
+        ob = o.owned_blocks(r)
+        assert sorted(slots.keys()) == ob      # dense, sorted slot map
+        assert sorted(slots.values()) == list(range(len(ob)))
