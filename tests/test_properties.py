@@ -162,3 +162,41 @@ def test_ownership_invariants(blocks, world, seed):
         ob = o.owned_blocks(r)
         assert sorted(slots.keys()) == ob      # dense, sorted slot map
         assert sorted(slots.values()) == list(range(len(ob)))
+
+
+@settings(max_examples=40, deadline=None)
+@given(seed=st.integers(0, 2**31 - 1), nops=st.integers(0, 8))
+def test_plan_json_roundtrip(seed, nops):
+    """Plans are published through the control store as JSON (collective
+    and async-queue delivery): serialization must be lossless for every
+    op type, deps, and the benefit estimate."""
+    import random
+
+    from harmony_amd.optimizer.plan import (DropTableOp, MoveOp, Plan,
+                                            SetBatchShareOp, StartWorkerOp,
+                                            StopWorkerOp)
+
+    rng = random.Random(seed)
+    ops = []
+    for _ in range(nops):
+        kind = rng.randrange(5)
+        if kind == 0:
+            ops.append(MoveOp(f"t{rng.randrange(3)}",
+                              tuple((rng.randrange(64), rng.randrange(8))
+                                    for _ in range(rng.randrange(4)))))
+        elif kind == 1:
+            ops.append(SetBatchShareOp(tuple((r, rng.randrange(1, 100))
+                                             for r in range(rng.randrange(1, 4)))))
+        elif kind == 2:
+            ops.append(StopWorkerOp(rng.randrange(8)))
+        elif kind == 3:
+            ops.append(StartWorkerOp(rng.randrange(8), rng.randrange(1, 50)))
+        else:
+            ops.append(DropTableOp(f"t{rng.randrange(3)}"))
+    deps = [(i, j) for i in range(len(ops)) for j in range(i + 1, len(ops))
+            if rng.random() < 0.3]
+    p = Plan(ops=ops, deps=deps, estimated_benefit=rng.random())
+    q = Plan.from_json(p.to_json())
+    assert q.ops == p.ops
+    assert [tuple(d) for d in q.deps] == [tuple(d) for d in p.deps]
+    assert abs(q.estimated_benefit - p.estimated_benefit) < 1e-12
