@@ -367,3 +367,13 @@ def test_async_plan_queue_skew_and_drain():
     assert n_pub >= 1
     assert applied[0] == applied[1] == n_pub, (applied, n_pub)
     assert o0._seq_applied == o1._seq_applied == n_pub
+
+
+def test_hetero_role_selection_world4():
+    """Role selection at world 4: demote exactly the slow rank, keep 3
+    workers (exercises the candidate-count enumeration beyond 2 ranks)."""
+    res = run_dist(_hetero_demote_worker, world=4, timeout=300)
+    assert len(set(res)) == 1                    # collective agreement
+    applied, share1 = res[0]
+    assert applied > 0
+    assert share1 == 0
