@@ -31,10 +31,17 @@ def _rank_main(rank, world, dist_port, js_port, q):
                              hb_period=0.2)   # fail-fast budget = 2 s
 
     if rank == 1:
-        # the victim: die abruptly 3 s in (several epochs + checkpoints
-        # will have completed)
+        # the victim: die abruptly — but only once at least one epoch
+        # checkpoint EXISTS on disk (a wall-clock delay flaked on loaded
+        # machines: imports + first epochs could exceed it, leaving no
+        # checkpoint to restore from). epoch1's dir appearing means
+        # epoch0's snapshot completed.
         def die():
-            time.sleep(3.0)
+            deadline = time.monotonic() + 60
+            marker = os.path.join(CHKP, "chaos1", "epoch1")
+            while time.monotonic() < deadline and not os.path.isdir(marker):
+                time.sleep(0.1)
+            time.sleep(0.5)       # let a couple more epochs through
             os._exit(1)
 
         threading.Thread(target=die, daemon=True).start()
