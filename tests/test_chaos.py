@@ -85,7 +85,8 @@ def test_chaos_kill_rank_then_restore(tmp_path):
     # rank 1 died at ~3 s; heartbeat budget is 10 * 0.2 = 2 s; the server
     # must have failed fast and shut down well within the test window
     assert res["failed_flag"], res
-    assert total < 60, total
+    assert total < 110, total      # bounded well below the queue timeout
+    #                                (fail-fast, not the 600 s clean join)
     # checkpoints from completed epochs exist (both ranks' block files)
     epochs = sorted(os.listdir(os.path.join(CHKP, "chaos1")))
     assert epochs, "no epoch checkpoints written before the kill"
