@@ -72,7 +72,21 @@ def _coerce(raw: str) -> Any:
 
 
 def _send(msg: dict, port: int, timeout: float = 3600.0) -> dict:
-    with socket.create_connection(("127.0.0.1", port), timeout=timeout) as s:
+    # bounded connect retry: a submit issued right after start_jobserver
+    # races the listener's bind (the reference CommandSender has the same
+    # localhost race); refused connections back off briefly instead of
+    # failing the command
+    deadline = time.monotonic() + 10.0
+    while True:
+        try:
+            conn = socket.create_connection(("127.0.0.1", port),
+                                            timeout=timeout)
+            break
+        except ConnectionRefusedError:
+            if time.monotonic() > deadline:
+                raise
+            time.sleep(0.1)
+    with conn as s:
         s.sendall((json.dumps(msg) + "\n").encode())
         return json.loads(s.makefile().readline())
 
