@@ -177,8 +177,10 @@ class SSPClock:
         deadlock against the global NET-ticket order — rank0.jobA blocked in
         SSP needs rank1.jobA's clock, rank1.jobA is queued behind a jobB seq
         whose local thread is blocked in ITS SSP needing rank0.jobB, which
-        is queued behind rank0.jobA (cycle; reproduced by
-        scripts/control_overhead.py --deadlock-demo). Real bounded-async
+        is queued behind rank0.jobA (cycle — realizable once ranks skew
+        by >=1 step; scripts/control_overhead.py --deadlock-demo runs the
+        hazardous configuration, though with empty step bodies the hang
+        is timing-dependent). Real bounded-async
         staleness lives in the one-sided plane (et/onesided.py), whose jobs
         take no tickets — so their SSP waits cannot enter such a cycle."""
         mine = self._clock.get(rank, 0) + 1

@@ -68,7 +68,7 @@ class V1SSPClock(SSPClock):
 
 
 def run_rank(rank, world, jobs, steps, version, port, out_q,
-             ssp_wait=False):
+             ssp_wait=False, slack=4):
     import faulthandler, os, sys
     if os.environ.get("CO_DEBUG"):
         sys.stderr = open(f"/tmp/co_r{rank}.err", "w", buffering=1)
@@ -85,7 +85,7 @@ def run_rank(rank, world, jobs, steps, version, port, out_q,
     for j in job_ids:
         tus.set_drawer(j, rank == 0)
     ssp_cls = SSPClock if version == "v2" else V1SSPClock
-    clocks = {j: ssp_cls(cp, j, world, slack=4) for j in job_ids}
+    clocks = {j: ssp_cls(cp, j, world, slack=slack) for j in job_ids}
 
     def job_thread(jid, res):
         import traceback
@@ -99,7 +99,8 @@ def run_rank(rank, world, jobs, steps, version, port, out_q,
         for _ in range(steps):
             # collective-plane jobs tick without blocking (wait=False); a
             # blocking SSP here can deadlock against the ticket order
-            # (--deadlock-demo reproduces it; see SSPClock.tick_and_wait)
+            # given >=1 step of cross-rank skew (--deadlock-demo runs the
+            # hazardous config; see SSPClock.tick_and_wait)
             clocks[jid].tick_and_wait(rank, wait=ssp_wait)
             with tus.net(jid, nxt(), lookahead=1):
                 pass                        # PULL (control only)
@@ -148,11 +149,11 @@ def run_rank(rank, world, jobs, steps, version, port, out_q,
     out_q.put((rank, wall, res))
 
 
-def measure(version, world, jobs, steps, port, ssp_wait=False):
+def measure(version, world, jobs, steps, port, ssp_wait=False, slack=4):
     q = mp.Queue()
     ps = [mp.Process(target=run_rank,
                      args=(r, world, jobs, steps, version, port, q,
-                           ssp_wait))
+                           ssp_wait, slack))
           for r in range(world)]
     for p in ps:
         p.start()
@@ -171,13 +172,17 @@ def main():
     ap.add_argument("--versions", default="v1,v2")
     ap.add_argument("--deadlock-demo", action="store_true",
                     help="run with blocking SSP waits inside ticketed jobs "
-                         "(the round-1 bug): HANGS at world>=2, jobs>=2")
+                         "at slack=0 — the hazardous configuration. The "
+                         "SSP/ticket wait cycle needs >=1 step of cross-"
+                         "rank skew, so with these empty step bodies the "
+                         "hang is TIMING-DEPENDENT (it was observed in "
+                         "development, not deterministic here)")
     args = ap.parse_args()
     if args.deadlock_demo:
-        print("deadlock demo: blocking SSP + tickets (expect a hang; "
-              "Ctrl-C or timeout to stop)")
+        print("deadlock demo: blocking SSP + tickets, slack=0 "
+              "(timing-dependent hang; run under a timeout)")
         measure("v2", args.world, args.jobs, args.steps, 29660,
-                ssp_wait=True)
+                ssp_wait=True, slack=0)
         return
     port0 = 29650
     out = {}
