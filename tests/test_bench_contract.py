@@ -3,6 +3,9 @@
 import json
 import subprocess
 import sys
+from pathlib import Path
+
+REPO = str(Path(__file__).resolve().parents[1])
 
 
 def test_bench_json_contract():
@@ -75,3 +78,22 @@ def test_bench_elastic_contract():
     ex1 = [row[3] for row in tl[1]]
     assert 0 in ex1, ex1[:8]            # rank 1 truly stopped mid-run
     assert ex1[0] > 0 and ex1[-1] > 0   # worked at both ends (restarted)
+
+
+def test_bench_direct_world2_torchrun():
+    """The driver's SCALE invocation shape: direct mode under torchrun at
+    world 2 (tickets + collectives + clock-warm loop all engaged)."""
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29734", "bench.py", "--gpus", "2",
+         "--device", "cpu", "--steps", "3", "--warmup", "1",
+         "--nmf-rows-per-batch", "64", "--nmf-cols", "256",
+         "--mlr-batch", "64", "--mlr-features", "64",
+         "--lda-docs-per-batch", "32", "--lda-vocab", "500",
+         "--lda-tokens-per-doc", "8"],
+        capture_output=True, text=True, timeout=300, cwd=REPO)
+    assert out.returncode == 0, out.stderr[-2000:]
+    d = json.loads(out.stdout.strip().splitlines()[-1])
+    assert d["metric"] == "aggregate_examples_per_sec_3job"
+    assert d["n_gpus"] == 2 and d["value"] > 0
